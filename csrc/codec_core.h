@@ -1,0 +1,826 @@
+// Shared host/device protobuf codec core for tf.Example / tf.SequenceExample.
+//
+// This is a from-scratch, schema-driven implementation of exactly the proto
+// subset the TFRecord formats use (SURVEY.md §1 "On-disk format"):
+//   Example           { Features features = 1; }
+//   Features          { map<string, Feature> feature = 1; }
+//   Feature           { oneof: BytesList=1 | FloatList=2 | Int64List=3 }
+//   BytesList         { repeated bytes value = 1; }
+//   FloatList         { repeated float value = 1 [packed]; }
+//   Int64List         { repeated int64 value = 1 [packed]; }
+//   SequenceExample   { Features context = 1; FeatureLists feature_lists = 2; }
+//   FeatureLists      { map<string, FeatureList> feature_list = 1; }
+//   FeatureList       { repeated Feature feature = 1; }
+//
+// All functions are __host__ __device__ so the identical logic backs the CPU
+// path (csrc/host_codec.cpp) and the gfx950 kernels (csrc/hip/*.hip): records
+// are the parallel axis on the GPU, bytes within a record are sequential here.
+//
+// Columnar ("wire-form") layout shared with Python (spark_tfrecord_amd/columnar.py):
+//   non-seq field : presence u8[R]; row_off i64[R+1] (values per row, cumulative)
+//   seq field     : + list_off i64[R+1] (sub-lists per row), sub_off i64[L+1]
+//                   (values per sub-list); row_off then covers values per ROW
+//   bytes kind    : values = flat u8; elem_off i64[E+1] (bytes per string)
+//   int64 / float : values = i64[V] / f32[V]
+#pragma once
+
+#include <cstdint>
+#include <cstring>
+
+#include "crc32c.h"
+
+namespace tfrec {
+
+using i64 = int64_t;
+using u64 = uint64_t;
+using u32 = uint32_t;
+using u8 = uint8_t;
+
+// Feature oneof field numbers double as kind codes (schema.py KIND_*).
+enum Kind : int32_t { KIND_BYTES = 1, KIND_FLOAT = 2, KIND_INT64 = 3 };
+
+enum RecordFormat : int32_t { FMT_EXAMPLE = 0, FMT_SEQUENCE = 1, FMT_BYTE_ARRAY = 2 };
+
+// Error codes (negative) returned by parse/size routines.
+enum CodecErr : int32_t {
+  ERR_OK = 0,
+  ERR_TRUNCATED = -1,
+  ERR_BAD_VARINT = -2,
+  ERR_KIND_MISMATCH = -3,
+  ERR_BAD_WIRETYPE = -4,
+  ERR_OVERFLOW = -5,
+};
+
+// ---------------------------------------------------------------------------
+// varint
+// ---------------------------------------------------------------------------
+
+// Reads a base-128 varint; returns new cursor or nullptr on malformed/overrun.
+TFR_HOSTDEV inline const u8* read_varint(const u8* p, const u8* end, u64* out) {
+  u64 v = 0;
+  int shift = 0;
+  while (p < end) {
+    u8 b = *p++;
+    v |= static_cast<u64>(b & 0x7F) << shift;
+    if (!(b & 0x80)) {
+      *out = v;
+      return p;
+    }
+    shift += 7;
+    if (shift >= 64) return nullptr;
+  }
+  return nullptr;
+}
+
+TFR_HOSTDEV inline int varint_size(u64 v) {
+  int n = 1;
+  while (v >= 0x80) {
+    v >>= 7;
+    ++n;
+  }
+  return n;
+}
+
+TFR_HOSTDEV inline u8* write_varint(u8* p, u64 v) {
+  while (v >= 0x80) {
+    *p++ = static_cast<u8>(v) | 0x80;
+    v >>= 7;
+  }
+  *p++ = static_cast<u8>(v);
+  return p;
+}
+
+// ---------------------------------------------------------------------------
+// Schema blob: [i32 nfields][FieldDescRaw x n][name bytes]
+// Built in Python (columnar.py), identical bytes shipped to host calls and
+// device constant buffers.
+// ---------------------------------------------------------------------------
+
+struct FieldDescRaw {
+  int32_t kind;      // Kind
+  int32_t is_seq;    // 1 => SequenceExample feature_lists entry (2-D ragged)
+  int32_t name_off;  // offset into the names section
+  int32_t name_len;
+};
+
+struct SchemaView {
+  int32_t nfields;
+  const FieldDescRaw* fields;
+  const u8* names;
+
+  TFR_HOSTDEV const u8* name(int f) const { return names + fields[f].name_off; }
+  TFR_HOSTDEV int name_len(int f) const { return fields[f].name_len; }
+};
+
+TFR_HOSTDEV inline SchemaView schema_view(const u8* blob) {
+  SchemaView v;
+  int32_t n;
+  __builtin_memcpy(&n, blob, 4);
+  v.nfields = n;
+  v.fields = reinterpret_cast<const FieldDescRaw*>(blob + 4);
+  v.names = blob + 4 + n * static_cast<int>(sizeof(FieldDescRaw));
+  return v;
+}
+
+TFR_HOSTDEV inline bool name_eq(const u8* a, const u8* b, int n) {
+  for (int i = 0; i < n; ++i)
+    if (a[i] != b[i]) return false;
+  return true;
+}
+
+// Linear schema lookup by feature name. Schemas are tens of fields; records
+// are the parallel axis, so this inner scan is cheap and branch-uniform.
+TFR_HOSTDEV inline int schema_find(const SchemaView& s, const u8* name, int len,
+                                   int want_seq) {
+  for (int f = 0; f < s.nfields; ++f) {
+    if (s.fields[f].is_seq == want_seq && s.fields[f].name_len == len &&
+        name_eq(s.name(f), name, len))
+      return f;
+  }
+  return -1;
+}
+
+// ---------------------------------------------------------------------------
+// Decode pass A: per-record structure scan.
+// For each schema field, locate the Feature (or FeatureList) body and count
+// elements. Mirrors what the reference's deserializer dispatch does per row
+// (TFRecordDeserializer.scala:21-61) but split into stats so the GPU can
+// prefix-sum offsets between passes.
+// ---------------------------------------------------------------------------
+
+struct FieldStat {
+  i64 pos;         // absolute byte offset of the body within the data buffer; -1 absent
+  i64 len;         // body length in bytes
+  i64 nvals;       // total scalar elements (ints / floats / strings)
+  i64 nbytes;      // total string payload bytes (bytes kind only)
+  i64 nlists;      // sub-list count (seq fields only)
+  int32_t kind_found;  // observed oneof kind, 0 if absent/empty Feature
+  int32_t err;
+};
+
+TFR_HOSTDEV inline void field_stat_clear(FieldStat* st) {
+  st->pos = -1;
+  st->len = 0;
+  st->nvals = 0;
+  st->nbytes = 0;
+  st->nlists = 0;
+  st->kind_found = 0;
+  st->err = 0;
+}
+
+// Skips one field value of the given wire type. Returns new cursor or nullptr.
+TFR_HOSTDEV inline const u8* skip_field(const u8* p, const u8* end, u32 wiretype) {
+  u64 tmp;
+  switch (wiretype) {
+    case 0:  // varint
+      return read_varint(p, end, &tmp);
+    case 1:  // fixed64
+      return (end - p >= 8) ? p + 8 : nullptr;
+    case 2:  // length-delimited
+      p = read_varint(p, end, &tmp);
+      if (!p || static_cast<u64>(end - p) < tmp) return nullptr;
+      return p + tmp;
+    case 5:  // fixed32
+      return (end - p >= 4) ? p + 4 : nullptr;
+    default:
+      return nullptr;
+  }
+}
+
+// Scans one list submessage body (BytesList/FloatList/Int64List content) and
+// counts elements. Accepts both packed and unpacked encodings for the numeric
+// kinds (parsers must; TF itself writes packed).
+TFR_HOSTDEV inline int32_t count_list_body(const u8* p, const u8* end, int32_t kind,
+                                           i64* nvals, i64* nbytes) {
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    if (fieldno != 1) {  // unknown field inside the list message: skip
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+      continue;
+    }
+    if (kind == KIND_BYTES) {
+      if (wt != 2) return ERR_BAD_WIRETYPE;
+      u64 len;
+      p = read_varint(p, end, &len);
+      if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+      *nvals += 1;
+      *nbytes += static_cast<i64>(len);
+      p += len;
+    } else if (kind == KIND_FLOAT) {
+      if (wt == 2) {  // packed
+        u64 len;
+        p = read_varint(p, end, &len);
+        if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+        *nvals += static_cast<i64>(len / 4);
+        p += len;
+      } else if (wt == 5) {
+        if (end - p < 4) return ERR_TRUNCATED;
+        *nvals += 1;
+        p += 4;
+      } else {
+        return ERR_BAD_WIRETYPE;
+      }
+    } else {  // KIND_INT64
+      if (wt == 2) {  // packed varints
+        u64 len;
+        p = read_varint(p, end, &len);
+        if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+        const u8* q = p;
+        const u8* qe = p + len;
+        while (q < qe) {
+          u64 v;
+          q = read_varint(q, qe, &v);
+          if (!q) return ERR_BAD_VARINT;
+          *nvals += 1;
+        }
+        p = qe;
+      } else if (wt == 0) {
+        u64 v;
+        p = read_varint(p, end, &v);
+        if (!p) return ERR_BAD_VARINT;
+        *nvals += 1;
+      } else {
+        return ERR_BAD_WIRETYPE;
+      }
+    }
+  }
+  return ERR_OK;
+}
+
+// Scans one Feature message body: identifies the oneof kind and counts
+// elements. `expect_kind` < 0 means "report what you find" (inference);
+// otherwise a different kind is ERR_KIND_MISMATCH (reference behavior:
+// TFRecordDeserializer.scala:177-221 kind checks).
+TFR_HOSTDEV inline int32_t scan_feature_body(const u8* p, const u8* end,
+                                             int32_t expect_kind, int32_t* kind_found,
+                                             i64* nvals, i64* nbytes) {
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    if (fieldno >= 1 && fieldno <= 3 && wt == 2) {
+      u64 len;
+      p = read_varint(p, end, &len);
+      if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+      *kind_found = static_cast<int32_t>(fieldno);
+      if (expect_kind >= 0 && static_cast<int32_t>(fieldno) != expect_kind)
+        return ERR_KIND_MISMATCH;
+      int32_t rc = count_list_body(p, p + len, static_cast<int32_t>(fieldno), nvals,
+                                   nbytes);
+      if (rc != ERR_OK) return rc;
+      p += len;
+    } else {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+    }
+  }
+  return ERR_OK;
+}
+
+// Scans a Features message body (map<string, Feature>) and fills stats for
+// schema-matched entries. `base` is the absolute offset of `p` within the
+// file buffer so stats can store absolute positions.
+TFR_HOSTDEV inline int32_t scan_features_body(const u8* p, const u8* end, i64 base,
+                                              const SchemaView& schema, int want_seq,
+                                              FieldStat* stats) {
+  const u8* body_start = p;
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    if (fieldno != 1 || wt != 2) {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+      continue;
+    }
+    u64 entry_len;
+    p = read_varint(p, end, &entry_len);
+    if (!p || static_cast<u64>(end - p) < entry_len) return ERR_TRUNCATED;
+    const u8* ep = p;
+    const u8* ee = p + entry_len;
+    p = ee;
+    // Map entry: key = 1 (string), value = 2 (Feature / FeatureList message).
+    const u8* key = nullptr;
+    u64 key_len = 0;
+    const u8* val = nullptr;
+    u64 val_len = 0;
+    while (ep < ee) {
+      u64 etag;
+      ep = read_varint(ep, ee, &etag);
+      if (!ep) return ERR_BAD_VARINT;
+      u32 efn = static_cast<u32>(etag >> 3);
+      u32 ewt = static_cast<u32>(etag & 7);
+      if (efn == 1 && ewt == 2) {
+        ep = read_varint(ep, ee, &key_len);
+        if (!ep || static_cast<u64>(ee - ep) < key_len) return ERR_TRUNCATED;
+        key = ep;
+        ep += key_len;
+      } else if (efn == 2 && ewt == 2) {
+        ep = read_varint(ep, ee, &val_len);
+        if (!ep || static_cast<u64>(ee - ep) < val_len) return ERR_TRUNCATED;
+        val = ep;
+        ep += val_len;
+      } else {
+        ep = skip_field(ep, ee, ewt);
+        if (!ep) return ERR_TRUNCATED;
+      }
+    }
+    if (!key) continue;
+    int f = schema_find(schema, key, static_cast<int>(key_len), want_seq);
+    if (f < 0) continue;  // unknown feature: ignored, like the reference
+    FieldStat* st = &stats[f];
+    st->pos = base + (val ? (val - body_start) : 0);
+    st->len = static_cast<i64>(val_len);
+    if (!want_seq) {
+      int32_t rc = scan_feature_body(val, val + val_len, schema.fields[f].kind,
+                                     &st->kind_found, &st->nvals, &st->nbytes);
+      if (rc != ERR_OK) st->err = rc;
+    } else {
+      // FeatureList body: repeated Feature (field 1).
+      const u8* lp = val;
+      const u8* le = val + val_len;
+      while (lp < le) {
+        u64 ltag;
+        lp = read_varint(lp, le, &ltag);
+        if (!lp) {
+          st->err = ERR_BAD_VARINT;
+          break;
+        }
+        u32 lfn = static_cast<u32>(ltag >> 3);
+        u32 lwt = static_cast<u32>(ltag & 7);
+        if (lfn == 1 && lwt == 2) {
+          u64 flen;
+          lp = read_varint(lp, le, &flen);
+          if (!lp || static_cast<u64>(le - lp) < flen) {
+            st->err = ERR_TRUNCATED;
+            break;
+          }
+          st->nlists += 1;
+          int32_t kf = 0;
+          int32_t rc = scan_feature_body(lp, lp + flen, schema.fields[f].kind, &kf,
+                                         &st->nvals, &st->nbytes);
+          if (rc != ERR_OK) {
+            st->err = rc;
+            break;
+          }
+          if (kf) st->kind_found = kf;
+          lp += flen;
+        } else {
+          lp = skip_field(lp, le, lwt);
+          if (!lp) {
+            st->err = ERR_TRUNCATED;
+            break;
+          }
+        }
+      }
+    }
+  }
+  return ERR_OK;
+}
+
+// Pass A entry point: scan one record payload. `fmt` selects Example vs
+// SequenceExample framing of the top-level message. `abs_off` is the record's
+// absolute offset in the data buffer. stats[f] must be pre-cleared.
+TFR_HOSTDEV inline int32_t scan_record(const u8* data, i64 abs_off, i64 rec_len,
+                                       int32_t fmt, const SchemaView& schema,
+                                       FieldStat* stats) {
+  const u8* p = data + abs_off;
+  const u8* end = p + rec_len;
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    bool is_features =
+        (fmt == FMT_EXAMPLE && fieldno == 1) || (fmt == FMT_SEQUENCE && fieldno == 1);
+    bool is_feature_lists = (fmt == FMT_SEQUENCE && fieldno == 2);
+    if ((is_features || is_feature_lists) && wt == 2) {
+      u64 len;
+      p = read_varint(p, end, &len);
+      if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+      i64 base = (p - data);
+      if (is_features) {
+        int32_t rc = scan_features_body(p, p + len, base, schema, 0, stats);
+        if (rc != ERR_OK) return rc;
+      } else {
+        int32_t rc = scan_features_body(p, p + len, base, schema, 1, stats);
+        if (rc != ERR_OK) return rc;
+      }
+      p += len;
+    } else {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+    }
+  }
+  return ERR_OK;
+}
+
+// ---------------------------------------------------------------------------
+// Decode pass B: value extraction for one (record, field), given the body
+// extent from pass A and destination offsets from the prefix sums.
+// ---------------------------------------------------------------------------
+
+struct DecodeDst {
+  i64* i64_vals;
+  float* f32_vals;
+  u8* bytes_data;   // flat string bytes
+  i64* elem_len;    // per-string byte length, later prefix-summed in Python
+  i64* sub_count;   // per-sub-list value count (seq), later prefix-summed
+};
+
+// Extracts one list body's values. Cursors are advanced versions of the
+// pass-A counters: *vi = next value slot, *bi = next byte offset.
+TFR_HOSTDEV inline int32_t extract_list_body(const u8* p, const u8* end, int32_t kind,
+                                             const DecodeDst& dst, i64* vi, i64* bi) {
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    if (fieldno != 1) {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+      continue;
+    }
+    if (kind == KIND_BYTES) {
+      u64 len;
+      p = read_varint(p, end, &len);
+      if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+      for (u64 i = 0; i < len; ++i) dst.bytes_data[*bi + static_cast<i64>(i)] = p[i];
+      dst.elem_len[*vi] = static_cast<i64>(len);
+      *bi += static_cast<i64>(len);
+      *vi += 1;
+      p += len;
+    } else if (kind == KIND_FLOAT) {
+      if (wt == 2) {
+        u64 len;
+        p = read_varint(p, end, &len);
+        if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+        u64 n = len / 4;
+        for (u64 i = 0; i < n; ++i) {
+          float v;
+          __builtin_memcpy(&v, p + 4 * i, 4);
+          dst.f32_vals[*vi + static_cast<i64>(i)] = v;
+        }
+        *vi += static_cast<i64>(n);
+        p += len;
+      } else {  // fixed32
+        if (end - p < 4) return ERR_TRUNCATED;
+        float v;
+        __builtin_memcpy(&v, p, 4);
+        dst.f32_vals[(*vi)++] = v;
+        p += 4;
+      }
+    } else {  // INT64
+      if (wt == 2) {
+        u64 len;
+        p = read_varint(p, end, &len);
+        if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+        const u8* q = p;
+        const u8* qe = p + len;
+        while (q < qe) {
+          u64 v;
+          q = read_varint(q, qe, &v);
+          if (!q) return ERR_BAD_VARINT;
+          dst.i64_vals[(*vi)++] = static_cast<i64>(v);
+        }
+        p = qe;
+      } else {
+        u64 v;
+        p = read_varint(p, end, &v);
+        if (!p) return ERR_BAD_VARINT;
+        dst.i64_vals[(*vi)++] = static_cast<i64>(v);
+      }
+    }
+  }
+  return ERR_OK;
+}
+
+TFR_HOSTDEV inline int32_t extract_feature_body(const u8* p, const u8* end,
+                                                int32_t kind, const DecodeDst& dst,
+                                                i64* vi, i64* bi) {
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    if (fieldno >= 1 && fieldno <= 3 && wt == 2) {
+      u64 len;
+      p = read_varint(p, end, &len);
+      if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
+      int32_t rc = extract_list_body(p, p + len, kind, dst, vi, bi);
+      if (rc != ERR_OK) return rc;
+      p += len;
+    } else {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+    }
+  }
+  return ERR_OK;
+}
+
+// Pass B for one (record, field): body extent [pos, pos+len) from pass A.
+// val_base / byte_base / list_base are this row's starting slots from the
+// prefix sums. For seq fields, also emits per-sub-list counts.
+TFR_HOSTDEV inline int32_t extract_field(const u8* data, i64 pos, i64 len,
+                                         int32_t kind, int32_t is_seq,
+                                         const DecodeDst& dst, i64 val_base,
+                                         i64 byte_base, i64 list_base) {
+  if (pos < 0) return ERR_OK;
+  const u8* p = data + pos;
+  const u8* end = p + len;
+  i64 vi = val_base;
+  i64 bi = byte_base;
+  if (!is_seq) return extract_feature_body(p, end, kind, dst, &vi, &bi);
+  // FeatureList: repeated Feature
+  i64 li = list_base;
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    if (fieldno == 1 && wt == 2) {
+      u64 flen;
+      p = read_varint(p, end, &flen);
+      if (!p || static_cast<u64>(end - p) < flen) return ERR_TRUNCATED;
+      i64 v_before = vi;
+      int32_t rc = extract_feature_body(p, p + flen, kind, dst, &vi, &bi);
+      if (rc != ERR_OK) return rc;
+      dst.sub_count[li++] = vi - v_before;
+      p += flen;
+    } else {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+    }
+  }
+  return ERR_OK;
+}
+
+// ---------------------------------------------------------------------------
+// Encode: columnar wire-form -> serialized Example/SequenceExample payloads.
+// Two passes (size, then emit) so the GPU can prefix-sum record offsets in
+// between; the per-record logic is identical for both and shared with host.
+// Mirrors the reference serializer's feature construction
+// (TFRecordSerializer.scala:20-60, builders :182-207).
+// ---------------------------------------------------------------------------
+
+struct FieldColumn {
+  int32_t kind;
+  int32_t is_seq;
+  const u8* presence;    // u8[R]; 0 => feature omitted (nullable null)
+  const i64* row_off;    // i64[R+1] cumulative VALUES per row
+  const i64* list_off;   // i64[R+1] cumulative sub-lists per row (seq only)
+  const i64* sub_off;    // i64[L+1] cumulative values per sub-list (seq only)
+  const i64* elem_off;   // i64[E+1] cumulative bytes per string (bytes only)
+  const u8* bytes_data;
+  const i64* i64_vals;
+  const float* f32_vals;
+};
+
+// Size of one list payload (the BytesList/FloatList/Int64List *body*), for
+// values [v0, v1) of a column.
+TFR_HOSTDEV inline i64 list_body_size(const FieldColumn& c, i64 v0, i64 v1) {
+  if (c.kind == KIND_FLOAT) {
+    i64 n = v1 - v0;
+    return n ? (1 + varint_size(static_cast<u64>(4 * n)) + 4 * n) : 0;
+  }
+  if (c.kind == KIND_INT64) {
+    i64 packed = 0;
+    for (i64 v = v0; v < v1; ++v)
+      packed += varint_size(static_cast<u64>(c.i64_vals[v]));
+    return packed ? (1 + varint_size(static_cast<u64>(packed)) + packed) : 0;
+  }
+  // bytes: repeated strings, each tag+len+data
+  i64 sz = 0;
+  for (i64 v = v0; v < v1; ++v) {
+    i64 blen = c.elem_off[v + 1] - c.elem_off[v];
+    sz += 1 + varint_size(static_cast<u64>(blen)) + blen;
+  }
+  return sz;
+}
+
+// Feature message body size for values [v0, v1): kind tag + len + list body.
+TFR_HOSTDEV inline i64 feature_body_size(const FieldColumn& c, i64 v0, i64 v1) {
+  i64 body = list_body_size(c, v0, v1);
+  return 1 + varint_size(static_cast<u64>(body)) + body;
+}
+
+TFR_HOSTDEV inline u8* emit_list_body(u8* o, const FieldColumn& c, i64 v0, i64 v1) {
+  if (c.kind == KIND_FLOAT) {
+    i64 n = v1 - v0;
+    if (n) {
+      *o++ = 0x0A;  // field 1, wiretype 2 (packed)
+      o = write_varint(o, static_cast<u64>(4 * n));
+      __builtin_memcpy(o, c.f32_vals + v0, static_cast<size_t>(4 * n));
+      o += 4 * n;
+    }
+  } else if (c.kind == KIND_INT64) {
+    i64 packed = 0;
+    for (i64 v = v0; v < v1; ++v)
+      packed += varint_size(static_cast<u64>(c.i64_vals[v]));
+    if (packed) {
+      *o++ = 0x0A;
+      o = write_varint(o, static_cast<u64>(packed));
+      for (i64 v = v0; v < v1; ++v)
+        o = write_varint(o, static_cast<u64>(c.i64_vals[v]));
+    }
+  } else {
+    for (i64 v = v0; v < v1; ++v) {
+      i64 b0 = c.elem_off[v];
+      i64 blen = c.elem_off[v + 1] - b0;
+      *o++ = 0x0A;
+      o = write_varint(o, static_cast<u64>(blen));
+      for (i64 i = 0; i < blen; ++i) o[i] = c.bytes_data[b0 + i];
+      o += blen;
+    }
+  }
+  return o;
+}
+
+TFR_HOSTDEV inline u8* emit_feature_body(u8* o, const FieldColumn& c, i64 v0, i64 v1) {
+  i64 body = list_body_size(c, v0, v1);
+  *o++ = static_cast<u8>((c.kind << 3) | 2);  // bytes_list=0x0A float=0x12 int64=0x1A
+  o = write_varint(o, static_cast<u64>(body));
+  return emit_list_body(o, c, v0, v1);
+}
+
+// map-entry body (key + Feature value) size for a non-seq field of row r.
+TFR_HOSTDEV inline i64 features_entry_size(const FieldColumn& c, const SchemaView& s,
+                                           int f, i64 v0, i64 v1) {
+  i64 fb = feature_body_size(c, v0, v1);
+  i64 klen = s.name_len(f);
+  return (1 + varint_size(static_cast<u64>(klen)) + klen) +
+         (1 + varint_size(static_cast<u64>(fb)) + fb);
+}
+
+TFR_HOSTDEV inline u8* emit_features_entry(u8* o, const FieldColumn& c,
+                                           const SchemaView& s, int f, i64 v0, i64 v1) {
+  i64 fb = feature_body_size(c, v0, v1);
+  i64 klen = s.name_len(f);
+  *o++ = 0x0A;  // key
+  o = write_varint(o, static_cast<u64>(klen));
+  const u8* nm = s.name(f);
+  for (i64 i = 0; i < klen; ++i) o[i] = nm[i];
+  o += klen;
+  *o++ = 0x12;  // value (Feature)
+  o = write_varint(o, static_cast<u64>(fb));
+  return emit_feature_body(o, c, v0, v1);
+}
+
+// FeatureList value body (repeated Feature) for row r of a seq field.
+TFR_HOSTDEV inline i64 feature_list_body_size(const FieldColumn& c, i64 r) {
+  i64 sz = 0;
+  for (i64 j = c.list_off[r]; j < c.list_off[r + 1]; ++j) {
+    i64 fb = feature_body_size(c, c.sub_off[j], c.sub_off[j + 1]);
+    sz += 1 + varint_size(static_cast<u64>(fb)) + fb;
+  }
+  return sz;
+}
+
+TFR_HOSTDEV inline u8* emit_feature_list_body(u8* o, const FieldColumn& c, i64 r) {
+  for (i64 j = c.list_off[r]; j < c.list_off[r + 1]; ++j) {
+    i64 fb = feature_body_size(c, c.sub_off[j], c.sub_off[j + 1]);
+    *o++ = 0x0A;  // FeatureList.feature
+    o = write_varint(o, static_cast<u64>(fb));
+    o = emit_feature_body(o, c, c.sub_off[j], c.sub_off[j + 1]);
+  }
+  return o;
+}
+
+// Seq map-entry (key + FeatureList) size for row r.
+TFR_HOSTDEV inline i64 feature_lists_entry_size(const FieldColumn& c,
+                                                const SchemaView& s, int f, i64 r) {
+  i64 flb = feature_list_body_size(c, r);
+  i64 klen = s.name_len(f);
+  return (1 + varint_size(static_cast<u64>(klen)) + klen) +
+         (1 + varint_size(static_cast<u64>(flb)) + flb);
+}
+
+TFR_HOSTDEV inline u8* emit_feature_lists_entry(u8* o, const FieldColumn& c,
+                                                const SchemaView& s, int f, i64 r) {
+  i64 flb = feature_list_body_size(c, r);
+  i64 klen = s.name_len(f);
+  *o++ = 0x0A;
+  o = write_varint(o, static_cast<u64>(klen));
+  const u8* nm = s.name(f);
+  for (i64 i = 0; i < klen; ++i) o[i] = nm[i];
+  o += klen;
+  *o++ = 0x12;
+  o = write_varint(o, static_cast<u64>(flb));
+  return emit_feature_list_body(o, c, r);
+}
+
+// Full record payload size for row r (Example or SequenceExample).
+TFR_HOSTDEV inline i64 record_payload_size(const FieldColumn* cols,
+                                           const SchemaView& s, int32_t fmt, i64 r) {
+  i64 ctx_body = 0;  // Features body (Example.features / SequenceExample.context)
+  i64 fl_body = 0;   // FeatureLists body
+  for (int f = 0; f < s.nfields; ++f) {
+    const FieldColumn& c = cols[f];
+    if (!c.presence[r]) continue;
+    if (!c.is_seq) {
+      i64 e = features_entry_size(c, s, f, c.row_off[r], c.row_off[r + 1]);
+      ctx_body += 1 + varint_size(static_cast<u64>(e)) + e;
+    } else {
+      i64 e = feature_lists_entry_size(c, s, f, r);
+      fl_body += 1 + varint_size(static_cast<u64>(e)) + e;
+    }
+  }
+  i64 total = 0;
+  if (fmt == FMT_EXAMPLE) {
+    // Always emit the features submessage (tag+len), even when empty — matches
+    // a builder with features always set.
+    total += 1 + varint_size(static_cast<u64>(ctx_body)) + ctx_body;
+  } else {
+    if (ctx_body) total += 1 + varint_size(static_cast<u64>(ctx_body)) + ctx_body;
+    total += 1 + varint_size(static_cast<u64>(fl_body)) + fl_body;
+  }
+  return total;
+}
+
+TFR_HOSTDEV inline u8* emit_record_payload(u8* o, const FieldColumn* cols,
+                                           const SchemaView& s, int32_t fmt, i64 r) {
+  i64 ctx_body = 0, fl_body = 0;
+  for (int f = 0; f < s.nfields; ++f) {
+    const FieldColumn& c = cols[f];
+    if (!c.presence[r]) continue;
+    if (!c.is_seq) {
+      i64 e = features_entry_size(c, s, f, c.row_off[r], c.row_off[r + 1]);
+      ctx_body += 1 + varint_size(static_cast<u64>(e)) + e;
+    } else {
+      i64 e = feature_lists_entry_size(c, s, f, r);
+      fl_body += 1 + varint_size(static_cast<u64>(e)) + e;
+    }
+  }
+  auto emit_ctx = [&](u8* oo) {
+    for (int f = 0; f < s.nfields; ++f) {
+      const FieldColumn& c = cols[f];
+      if (!c.presence[r] || c.is_seq) continue;
+      i64 e = features_entry_size(c, s, f, c.row_off[r], c.row_off[r + 1]);
+      *oo++ = 0x0A;  // Features.feature map entry
+      oo = write_varint(oo, static_cast<u64>(e));
+      oo = emit_features_entry(oo, c, s, f, c.row_off[r], c.row_off[r + 1]);
+    }
+    return oo;
+  };
+  auto emit_fl = [&](u8* oo) {
+    for (int f = 0; f < s.nfields; ++f) {
+      const FieldColumn& c = cols[f];
+      if (!c.presence[r] || !c.is_seq) continue;
+      i64 e = feature_lists_entry_size(c, s, f, r);
+      *oo++ = 0x0A;  // FeatureLists.feature_list map entry
+      oo = write_varint(oo, static_cast<u64>(e));
+      oo = emit_feature_lists_entry(oo, c, s, f, r);
+    }
+    return oo;
+  };
+  if (fmt == FMT_EXAMPLE) {
+    *o++ = 0x0A;  // Example.features
+    o = write_varint(o, static_cast<u64>(ctx_body));
+    o = emit_ctx(o);
+  } else {
+    if (ctx_body) {
+      *o++ = 0x0A;  // SequenceExample.context
+      o = write_varint(o, static_cast<u64>(ctx_body));
+      o = emit_ctx(o);
+    }
+    *o++ = 0x12;  // SequenceExample.feature_lists
+    o = write_varint(o, static_cast<u64>(fl_body));
+    o = emit_fl(o);
+  }
+  return o;
+}
+
+// ---------------------------------------------------------------------------
+// Frame emit: payload -> [u64 len][masked crc(len)][payload][masked crc(payload)]
+// at an absolute offset. Payload must already be in place at frame_off + 12.
+// ---------------------------------------------------------------------------
+
+constexpr i64 kFrameOverhead = 16;  // 8 len + 4 crc + 4 crc
+
+TFR_HOSTDEV inline void write_frame_header_footer(u8* file, i64 frame_off,
+                                                  i64 payload_len,
+                                                  const u32 (*tab)[256]) {
+  u8* h = file + frame_off;
+  u64 len_le = static_cast<u64>(payload_len);
+  __builtin_memcpy(h, &len_le, 8);  // little-endian host & device
+  u32 lc = mask_crc(crc32c_sw(h, 8, 0, tab));
+  __builtin_memcpy(h + 8, &lc, 4);
+  u32 dc = mask_crc(crc32c_sw(h + 12, static_cast<size_t>(payload_len), 0, tab));
+  __builtin_memcpy(h + 12 + payload_len, &dc, 4);
+}
+
+}  // namespace tfrec

@@ -1,0 +1,120 @@
+// CRC32C (Castagnoli) + TFRecord masking, shared host/device.
+//
+// The TFRecord frame (SURVEY.md §1 "On-disk format"; behavior of the
+// reference's external TFRecordWriter/TFRecordFileInputFormat, call sites
+// TFRecordOutputWriter.scala:21,37 and TFRecordFileReader.scala:32,51):
+//   uint64 length (LE) | uint32 masked_crc32c(length_bytes) |
+//   byte data[length]  | uint32 masked_crc32c(data)
+//   masked = ((crc >> 15) | (crc << 17)) + 0xa282ead8
+//
+// Implementation: slicing-by-8 with compile-time-generated tables so the same
+// code runs on the x86 host and inside gfx950 kernels (tables land in .rodata
+// on host and constant memory on device). The host additionally has an
+// SSE4.2 hardware-CRC path.
+#pragma once
+
+#include <cstddef>
+#include <cstdint>
+
+#if defined(__HIPCC__)
+#define TFR_HOSTDEV __host__ __device__
+#else
+#define TFR_HOSTDEV
+#endif
+
+namespace tfrec {
+
+constexpr uint32_t kCrc32cPoly = 0x82F63B78u;  // reversed Castagnoli
+constexpr uint32_t kMaskDelta = 0xA282EAD8u;
+
+struct Crc32cTables {
+  uint32_t t[8][256];
+};
+
+constexpr Crc32cTables make_crc32c_tables() {
+  Crc32cTables tb{};
+  for (uint32_t i = 0; i < 256; ++i) {
+    uint32_t c = i;
+    for (int k = 0; k < 8; ++k) c = (c & 1) ? (kCrc32cPoly ^ (c >> 1)) : (c >> 1);
+    tb.t[0][i] = c;
+  }
+  for (uint32_t i = 0; i < 256; ++i) {
+    uint32_t c = tb.t[0][i];
+    for (int s = 1; s < 8; ++s) {
+      c = tb.t[0][c & 0xFF] ^ (c >> 8);
+      tb.t[s][i] = c;
+    }
+  }
+  return tb;
+}
+
+inline constexpr Crc32cTables kCrcTables = make_crc32c_tables();
+
+// Table-driven slicing-by-8. `tab` lets device code pass an LDS-staged copy.
+TFR_HOSTDEV inline uint32_t crc32c_sw(const uint8_t* p, size_t n,
+                                      uint32_t crc = 0,
+                                      const uint32_t (*tab)[256] = kCrcTables.t) {
+  crc = ~crc;
+  // Align to 8 bytes.
+  while (n && (reinterpret_cast<uintptr_t>(p) & 7u)) {
+    crc = tab[0][(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+    --n;
+  }
+  while (n >= 8) {
+    uint64_t w;
+    __builtin_memcpy(&w, p, 8);
+    w ^= crc;  // little-endian
+    crc = tab[7][w & 0xFF] ^ tab[6][(w >> 8) & 0xFF] ^ tab[5][(w >> 16) & 0xFF] ^
+          tab[4][(w >> 24) & 0xFF] ^ tab[3][(w >> 32) & 0xFF] ^
+          tab[2][(w >> 40) & 0xFF] ^ tab[1][(w >> 48) & 0xFF] ^
+          tab[0][(w >> 56) & 0xFF];
+    p += 8;
+    n -= 8;
+  }
+  while (n--) crc = tab[0][(crc ^ *p++) & 0xFF] ^ (crc >> 8);
+  return ~crc;
+}
+
+#if defined(__SSE4_2__) && !defined(__HIP_DEVICE_COMPILE__)
+inline uint32_t crc32c_hw(const uint8_t* p, size_t n, uint32_t crc = 0) {
+  crc = ~crc;
+  while (n && (reinterpret_cast<uintptr_t>(p) & 7u)) {
+    crc = __builtin_ia32_crc32qi(crc, *p++);
+    --n;
+  }
+  uint64_t c64 = crc;
+  while (n >= 8) {
+    uint64_t w;
+    __builtin_memcpy(&w, p, 8);
+    c64 = __builtin_ia32_crc32di(c64, w);
+    p += 8;
+    n -= 8;
+  }
+  crc = static_cast<uint32_t>(c64);
+  while (n--) crc = __builtin_ia32_crc32qi(crc, *p++);
+  return ~crc;
+}
+#endif
+
+TFR_HOSTDEV inline uint32_t crc32c(const uint8_t* p, size_t n, uint32_t crc = 0) {
+#if defined(__SSE4_2__) && !defined(__HIP_DEVICE_COMPILE__)
+  return crc32c_hw(p, n, crc);
+#else
+  return crc32c_sw(p, n, crc);
+#endif
+}
+
+TFR_HOSTDEV inline uint32_t mask_crc(uint32_t crc) {
+  return ((crc >> 15) | (crc << 17)) + kMaskDelta;
+}
+
+TFR_HOSTDEV inline uint32_t unmask_crc(uint32_t masked) {
+  uint32_t rot = masked - kMaskDelta;
+  return (rot << 15) | (rot >> 17);
+}
+
+TFR_HOSTDEV inline uint32_t masked_crc32c(const uint8_t* p, size_t n) {
+  return mask_crc(crc32c(p, n));
+}
+
+}  // namespace tfrec

@@ -1,0 +1,516 @@
+// Python bindings + host-side (CPU) orchestration for the TFRecord codec.
+//
+// The CPU path here serves three roles:
+//  - the no-GPU plumbing configuration (BASELINE.json config 1),
+//  - the golden reference the HIP kernels are tested against,
+//  - small-file metadata work (frame header scan) that stays on host even in
+//    the GPU pipeline.
+//
+// GPU entry points live in csrc/hip/kernels.hip and are registered by
+// register_gpu() when compiled in.
+
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstring>
+#include <map>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "codec_core.h"
+
+namespace py = pybind11;
+using namespace tfrec;
+
+namespace {
+
+struct BufView {
+  const u8* data;
+  i64 size;
+};
+
+BufView as_bytes(const py::buffer& b, py::buffer_info& info) {
+  info = b.request();
+  if (info.itemsize != 1) throw std::invalid_argument("expected a byte buffer");
+  return BufView{static_cast<const u8*>(info.ptr), static_cast<i64>(info.size)};
+}
+
+// ---------------------------------------------------------------------------
+// Frame scan (host): sequential walk of [len][crc][payload][crc] frames.
+// Mirrors the read-side behavior of the reference's record reader
+// (TFRecordFileReader.scala:51 via tensorflow-hadoop). The headers are ~16
+// bytes per record; this stays on host even for the GPU path, feeding the
+// record-offset array that the kernels parallelize over (SURVEY.md §7
+// "hard parts": TFRecord has no sync markers).
+// ---------------------------------------------------------------------------
+
+std::pair<py::array_t<i64>, py::array_t<i64>> scan_frames(py::buffer data,
+                                                          bool verify_crc) {
+  py::buffer_info info;
+  BufView buf = as_bytes(data, info);
+  std::vector<i64> offs, lens;
+  i64 p = 0;
+  while (p < buf.size) {
+    if (buf.size - p < 12)
+      throw std::runtime_error("corrupt TFRecord: truncated frame header at offset " +
+                               std::to_string(p));
+    u64 len;
+    std::memcpy(&len, buf.data + p, 8);
+    u32 len_crc;
+    std::memcpy(&len_crc, buf.data + p + 8, 4);
+    if (verify_crc && mask_crc(crc32c(buf.data + p, 8)) != len_crc)
+      throw std::runtime_error("corrupt TFRecord: bad length CRC at offset " +
+                               std::to_string(p));
+    if (static_cast<u64>(buf.size - p - 12) < len + 4)
+      throw std::runtime_error("corrupt TFRecord: truncated payload at offset " +
+                               std::to_string(p));
+    if (verify_crc) {
+      u32 data_crc;
+      std::memcpy(&data_crc, buf.data + p + 12 + len, 4);
+      if (mask_crc(crc32c(buf.data + p + 12, len)) != data_crc)
+        throw std::runtime_error("corrupt TFRecord: bad data CRC at offset " +
+                                 std::to_string(p));
+    }
+    offs.push_back(p + 12);
+    lens.push_back(static_cast<i64>(len));
+    p += 12 + static_cast<i64>(len) + 4;
+  }
+  auto off_arr = py::array_t<i64>(offs.size());
+  auto len_arr = py::array_t<i64>(lens.size());
+  std::memcpy(off_arr.mutable_data(), offs.data(), offs.size() * 8);
+  std::memcpy(len_arr.mutable_data(), lens.data(), lens.size() * 8);
+  return {off_arr, len_arr};
+}
+
+// Fast header-only scan used by the GPU read path: offsets out, CRC checking
+// deferred to the device kernel.
+std::pair<py::array_t<i64>, py::array_t<i64>> scan_frame_headers(py::buffer data) {
+  return scan_frames(data, false);
+}
+
+// ---------------------------------------------------------------------------
+// Decode (host)
+// ---------------------------------------------------------------------------
+
+const char* err_name(int32_t e) {
+  switch (e) {
+    case ERR_TRUNCATED: return "truncated message";
+    case ERR_BAD_VARINT: return "malformed varint";
+    case ERR_KIND_MISMATCH: return "feature kind does not match schema";
+    case ERR_BAD_WIRETYPE: return "unexpected wire type";
+    default: return "codec error";
+  }
+}
+
+py::list decode_records(py::buffer data, py::array_t<i64> rec_off,
+                        py::array_t<i64> rec_len, py::bytes schema_blob,
+                        int32_t fmt) {
+  py::buffer_info info;
+  BufView buf = as_bytes(data, info);
+  std::string blob = schema_blob;
+  SchemaView schema = schema_view(reinterpret_cast<const u8*>(blob.data()));
+  const i64 R = rec_off.size();
+  const int F = schema.nfields;
+  auto off = rec_off.unchecked<1>();
+  auto len = rec_len.unchecked<1>();
+
+  // Pass A: structure scan -> per (record, field) stats.
+  std::vector<FieldStat> stats(static_cast<size_t>(R) * F);
+  for (i64 r = 0; r < R; ++r) {
+    FieldStat* st = stats.data() + r * F;
+    for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+    int32_t rc = scan_record(buf.data, off(r), len(r), fmt, schema, st);
+    if (rc != ERR_OK)
+      throw std::runtime_error(std::string("TFRecord decode failed in record ") +
+                               std::to_string(r) + ": " + err_name(rc));
+    for (int f = 0; f < F; ++f)
+      if (st[f].err == ERR_KIND_MISMATCH) {
+        std::string nm(reinterpret_cast<const char*>(schema.name(f)),
+                       schema.name_len(f));
+        throw std::runtime_error("Feature '" + nm +
+                                 "' kind does not match requested data type (record " +
+                                 std::to_string(r) + ")");
+      } else if (st[f].err != ERR_OK) {
+        throw std::runtime_error(std::string("TFRecord decode failed in record ") +
+                                 std::to_string(r) + ": " + err_name(st[f].err));
+      }
+  }
+
+  // Prefix sums per field + allocation + pass B.
+  py::list out;
+  for (int f = 0; f < F; ++f) {
+    const FieldDescRaw& fd = schema.fields[f];
+    i64 total_vals = 0, total_bytes = 0, total_lists = 0;
+    auto presence = py::array_t<u8>(R);
+    auto row_off_arr = py::array_t<i64>(R + 1);
+    u8* pres = presence.mutable_data();
+    i64* row_off_p = row_off_arr.mutable_data();
+    row_off_p[0] = 0;
+    py::array_t<i64> list_off_arr(fd.is_seq ? R + 1 : 1);
+    i64* list_off_p = list_off_arr.mutable_data();
+    list_off_p[0] = 0;
+    for (i64 r = 0; r < R; ++r) {
+      const FieldStat& st = stats[r * F + f];
+      pres[r] = st.pos >= 0 ? 1 : 0;
+      total_vals += st.nvals;
+      total_bytes += st.nbytes;
+      total_lists += st.nlists;
+      row_off_p[r + 1] = total_vals;
+      if (fd.is_seq) list_off_p[r + 1] = total_lists;
+    }
+
+    DecodeDst dst{};
+    py::array_t<i64> i64_vals(0);
+    py::array_t<float> f32_vals(0);
+    py::array_t<u8> bytes_data(0);
+    py::array_t<i64> elem_len(0);
+    py::array_t<i64> sub_count(0);
+    if (fd.kind == KIND_INT64) {
+      i64_vals = py::array_t<i64>(total_vals);
+      dst.i64_vals = i64_vals.mutable_data();
+    } else if (fd.kind == KIND_FLOAT) {
+      f32_vals = py::array_t<float>(total_vals);
+      dst.f32_vals = f32_vals.mutable_data();
+    } else {
+      bytes_data = py::array_t<u8>(total_bytes);
+      elem_len = py::array_t<i64>(total_vals);
+      dst.bytes_data = bytes_data.mutable_data();
+      dst.elem_len = elem_len.mutable_data();
+    }
+    if (fd.is_seq) {
+      sub_count = py::array_t<i64>(total_lists);
+      dst.sub_count = sub_count.mutable_data();
+    }
+
+    i64 val_base = 0, byte_base = 0, list_base = 0;
+    for (i64 r = 0; r < R; ++r) {
+      const FieldStat& st = stats[r * F + f];
+      int32_t rc = extract_field(buf.data, st.pos, st.len, fd.kind, fd.is_seq, dst,
+                                 val_base, byte_base, list_base);
+      if (rc != ERR_OK)
+        throw std::runtime_error(std::string("TFRecord decode failed in record ") +
+                                 std::to_string(r) + ": " + err_name(rc));
+      val_base += st.nvals;
+      byte_base += st.nbytes;
+      list_base += st.nlists;
+    }
+
+    py::dict d;
+    d["presence"] = presence;
+    d["row_off"] = row_off_arr;
+    if (fd.kind == KIND_INT64) d["values"] = i64_vals;
+    else if (fd.kind == KIND_FLOAT) d["values"] = f32_vals;
+    else {
+      d["values"] = bytes_data;
+      d["elem_len"] = elem_len;  // Python cumsums to elem_off
+    }
+    if (fd.is_seq) {
+      d["list_off"] = list_off_arr;
+      d["sub_count"] = sub_count;  // Python cumsums to sub_off
+    }
+    out.append(d);
+  }
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// Encode (host): columnar wire-form -> full framed file image.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+const T* opt_ptr(const py::dict& d, const char* key) {
+  if (!d.contains(key)) return nullptr;
+  auto arr = d[key].cast<py::array_t<T>>();
+  return arr.data();
+}
+
+// Keeps the cast py::array objects alive while we hold raw pointers.
+struct ColKeepAlive {
+  std::vector<py::object> refs;
+  template <typename T>
+  const T* get(const py::dict& d, const char* key) {
+    if (!d.contains(key)) return nullptr;
+    py::array_t<T, py::array::c_style | py::array::forcecast> arr =
+        py::cast<py::array_t<T, py::array::c_style | py::array::forcecast>>(d[key]);
+    refs.push_back(arr);
+    return arr.data();
+  }
+};
+
+py::bytes encode_records(py::bytes schema_blob, int32_t fmt, py::list col_dicts,
+                         i64 R) {
+  std::string blob = schema_blob;
+  SchemaView schema = schema_view(reinterpret_cast<const u8*>(blob.data()));
+  const int F = schema.nfields;
+  if (static_cast<int>(col_dicts.size()) != F)
+    throw std::invalid_argument("column count does not match schema");
+
+  ColKeepAlive keep;
+  std::vector<FieldColumn> cols(F);
+  for (int f = 0; f < F; ++f) {
+    py::dict d = col_dicts[f].cast<py::dict>();
+    FieldColumn& c = cols[f];
+    c.kind = schema.fields[f].kind;
+    c.is_seq = schema.fields[f].is_seq;
+    c.presence = keep.get<u8>(d, "presence");
+    c.row_off = keep.get<i64>(d, "row_off");
+    c.list_off = keep.get<i64>(d, "list_off");
+    c.sub_off = keep.get<i64>(d, "sub_off");
+    c.elem_off = keep.get<i64>(d, "elem_off");
+    c.bytes_data = keep.get<u8>(d, "values_bytes");
+    c.i64_vals = keep.get<i64>(d, "values_i64");
+    c.f32_vals = keep.get<float>(d, "values_f32");
+    if (!c.presence || !c.row_off)
+      throw std::invalid_argument("column missing presence/row_off");
+  }
+
+  // Pass A: payload sizes -> frame offsets.
+  std::vector<i64> frame_off(R + 1);
+  std::vector<i64> psize(R);
+  frame_off[0] = 0;
+  for (i64 r = 0; r < R; ++r) {
+    psize[r] = record_payload_size(cols.data(), schema, fmt, r);
+    frame_off[r + 1] = frame_off[r] + psize[r] + kFrameOverhead;
+  }
+  i64 total = frame_off[R];
+
+  // Pass B: emit payloads + frames.
+  std::string out(static_cast<size_t>(total), '\0');
+  u8* file = reinterpret_cast<u8*>(out.data());
+  for (i64 r = 0; r < R; ++r) {
+    u8* o = file + frame_off[r] + 12;
+    u8* oe = emit_record_payload(o, cols.data(), schema, fmt, r);
+    if (oe - o != psize[r])
+      throw std::runtime_error("internal error: emit size mismatch (record " +
+                               std::to_string(r) + ")");
+    write_frame_header_footer(file, frame_off[r], psize[r], kCrcTables.t);
+  }
+  return py::bytes(out);
+}
+
+// ByteArray write path: frame raw binary payloads directly
+// (reference: TFRecordSerializer.scala:16-18 + TFRecordOutputWriter.scala:28-29).
+py::bytes frame_byte_arrays(py::buffer data, py::array_t<i64> elem_off) {
+  py::buffer_info info;
+  BufView buf = as_bytes(data, info);
+  auto off = elem_off.unchecked<1>();
+  i64 R = elem_off.size() - 1;
+  i64 total = 0;
+  for (i64 r = 0; r < R; ++r) total += (off(r + 1) - off(r)) + kFrameOverhead;
+  std::string out(static_cast<size_t>(total), '\0');
+  u8* file = reinterpret_cast<u8*>(out.data());
+  i64 fo = 0;
+  for (i64 r = 0; r < R; ++r) {
+    i64 n = off(r + 1) - off(r);
+    std::memcpy(file + fo + 12, buf.data + off(r), static_cast<size_t>(n));
+    write_frame_header_footer(file, fo, n, kCrcTables.t);
+    fo += n + kFrameOverhead;
+  }
+  return py::bytes(out);
+}
+
+// ---------------------------------------------------------------------------
+// Schema inference (host): per-record lattice codes merged with max.
+// Mirrors TensorFlowInferSchema.scala:75-118 + :132-188:
+//   per row: n==0 -> null, n==1 -> scalar, n>1 -> array; FeatureLists always
+//   infer 2-D; bytes infer as String; merge = lattice max.
+// Returns {name: code} with codes from schema.py lattice_code (seq fields
+// offset by their 2-D position in the lattice).
+// ---------------------------------------------------------------------------
+
+// kind (1=bytes,2=float,3=int64) -> rank 3/2/1; code = rank, +3 if multi, +6 if seq
+inline int lattice_code_for(int32_t kind, bool multi, bool seq) {
+  int rank = 4 - kind;  // int64->1 float->2 bytes->3
+  if (seq) return 6 + rank;
+  if (multi) return 3 + rank;
+  return rank;
+}
+
+void infer_features_body(const u8* p, const u8* end, bool seq,
+                         std::map<std::string, int>& codes) {
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) throw std::runtime_error("malformed record during schema inference");
+    u32 fieldno = static_cast<u32>(tag >> 3);
+    u32 wt = static_cast<u32>(tag & 7);
+    if (fieldno != 1 || wt != 2) {
+      p = skip_field(p, end, wt);
+      if (!p) throw std::runtime_error("malformed record during schema inference");
+      continue;
+    }
+    u64 entry_len;
+    p = read_varint(p, end, &entry_len);
+    if (!p || static_cast<u64>(end - p) < entry_len)
+      throw std::runtime_error("malformed record during schema inference");
+    const u8* ep = p;
+    const u8* ee = p + entry_len;
+    p = ee;
+    std::string key;
+    const u8* val = nullptr;
+    u64 val_len = 0;
+    while (ep < ee) {
+      u64 etag;
+      ep = read_varint(ep, ee, &etag);
+      if (!ep) throw std::runtime_error("malformed map entry during schema inference");
+      u32 efn = static_cast<u32>(etag >> 3);
+      u32 ewt = static_cast<u32>(etag & 7);
+      if (efn == 1 && ewt == 2) {
+        u64 klen;
+        ep = read_varint(ep, ee, &klen);
+        if (!ep || static_cast<u64>(ee - ep) < klen)
+          throw std::runtime_error("malformed map entry during schema inference");
+        key.assign(reinterpret_cast<const char*>(ep), klen);
+        ep += klen;
+      } else if (efn == 2 && ewt == 2) {
+        ep = read_varint(ep, ee, &val_len);
+        if (!ep || static_cast<u64>(ee - ep) < val_len)
+          throw std::runtime_error("malformed map entry during schema inference");
+        val = ep;
+        ep += val_len;
+      } else {
+        ep = skip_field(ep, ee, ewt);
+        if (!ep) throw std::runtime_error("malformed map entry during schema inference");
+      }
+    }
+    if (key.empty() && !val) continue;
+    int code = 0;
+    if (val) {
+      if (!seq) {
+        int32_t kf = 0;
+        i64 nvals = 0, nbytes = 0;
+        int32_t rc = scan_feature_body(val, val + val_len, -1, &kf, &nvals, &nbytes);
+        if (rc != ERR_OK)
+          throw std::runtime_error("malformed Feature during schema inference");
+        if (kf != 0 && nvals > 0) code = lattice_code_for(kf, nvals > 1, false);
+        // n==0 or no kind set -> null (code 0): TensorFlowInferSchema.scala:147-188
+      } else {
+        // FeatureList: any content infers as Array(Array(T)) regardless of
+        // lengths (TensorFlowInferSchema.scala:98-118).
+        const u8* lp = val;
+        const u8* le = val + val_len;
+        int32_t kind_seen = 0;
+        bool any_vals = false;
+        while (lp < le) {
+          u64 ltag;
+          lp = read_varint(lp, le, &ltag);
+          if (!lp) throw std::runtime_error("malformed FeatureList during inference");
+          u32 lfn = static_cast<u32>(ltag >> 3);
+          u32 lwt = static_cast<u32>(ltag & 7);
+          if (lfn == 1 && lwt == 2) {
+            u64 flen;
+            lp = read_varint(lp, le, &flen);
+            if (!lp || static_cast<u64>(le - lp) < flen)
+              throw std::runtime_error("malformed FeatureList during inference");
+            int32_t kf = 0;
+            i64 nvals = 0, nbytes = 0;
+            int32_t rc = scan_feature_body(lp, lp + flen, -1, &kf, &nvals, &nbytes);
+            if (rc != ERR_OK)
+              throw std::runtime_error("malformed Feature during schema inference");
+            if (kf) {
+              kind_seen = std::max(kind_seen, lattice_code_for(kf, false, false));
+              if (nvals >= 0) any_vals = true;
+            }
+            lp += flen;
+          } else {
+            lp = skip_field(lp, le, lwt);
+            if (!lp) throw std::runtime_error("malformed FeatureList during inference");
+          }
+        }
+        (void)any_vals;
+        if (kind_seen) code = 6 + kind_seen;  // rank embedded in kind_seen
+      }
+    }
+    auto it = codes.find(key);
+    if (it == codes.end()) codes[key] = code;
+    else it->second = std::max(it->second, code);
+  }
+}
+
+py::dict infer_schema_codes(py::buffer data, py::array_t<i64> rec_off,
+                            py::array_t<i64> rec_len, int32_t fmt) {
+  py::buffer_info info;
+  BufView buf = as_bytes(data, info);
+  auto off = rec_off.unchecked<1>();
+  auto len = rec_len.unchecked<1>();
+  std::map<std::string, int> ctx_codes, seq_codes;
+  for (i64 r = 0; r < rec_off.size(); ++r) {
+    const u8* p = buf.data + off(r);
+    const u8* end = p + len(r);
+    while (p < end) {
+      u64 tag;
+      p = read_varint(p, end, &tag);
+      if (!p) throw std::runtime_error("malformed record during schema inference");
+      u32 fieldno = static_cast<u32>(tag >> 3);
+      u32 wt = static_cast<u32>(tag & 7);
+      bool is_features = fieldno == 1;
+      bool is_fl = (fmt == FMT_SEQUENCE && fieldno == 2);
+      if ((is_features || is_fl) && wt == 2) {
+        u64 blen;
+        p = read_varint(p, end, &blen);
+        if (!p || static_cast<u64>(end - p) < blen)
+          throw std::runtime_error("malformed record during schema inference");
+        infer_features_body(p, p + blen, is_fl, is_fl ? seq_codes : ctx_codes);
+        p += blen;
+      } else {
+        p = skip_field(p, end, wt);
+        if (!p) throw std::runtime_error("malformed record during schema inference");
+      }
+    }
+  }
+  py::dict out;
+  for (auto& kv : ctx_codes) out[py::str(kv.first)] = kv.second;
+  for (auto& kv : seq_codes) {
+    // A name can only be context or sequence within one record type read.
+    out[py::str(kv.first)] = kv.second;
+  }
+  return out;
+}
+
+u32 crc32c_py(py::buffer data) {
+  py::buffer_info info;
+  BufView buf = as_bytes(data, info);
+  return crc32c(buf.data, static_cast<size_t>(buf.size));
+}
+
+u32 masked_crc32c_py(py::buffer data) {
+  py::buffer_info info;
+  BufView buf = as_bytes(data, info);
+  return masked_crc32c(buf.data, static_cast<size_t>(buf.size));
+}
+
+}  // namespace
+
+void register_gpu(py::module_& m);  // defined in csrc/hip/kernels.hip
+
+PYBIND11_MODULE(_native, m) {
+  m.doc() = "MI355X-native TFRecord codec (host + gfx950 kernels)";
+  m.def("crc32c", &crc32c_py, "CRC32C (Castagnoli) of a byte buffer");
+  m.def("masked_crc32c", &masked_crc32c_py, "TFRecord-masked CRC32C");
+  m.def("scan_frames", &scan_frames, py::arg("data"), py::arg("verify_crc") = true,
+        "Scan TFRecord frames -> (payload offsets, payload lengths)");
+  m.def("scan_frame_headers", &scan_frame_headers, py::arg("data"),
+        "Header-only frame scan (CRC verification deferred to the GPU)");
+  m.def("decode_records", &decode_records, py::arg("data"), py::arg("rec_off"),
+        py::arg("rec_len"), py::arg("schema_blob"), py::arg("fmt"),
+        "Decode Example/SequenceExample payloads into columnar wire-form");
+  m.def("encode_records", &encode_records, py::arg("schema_blob"), py::arg("fmt"),
+        py::arg("columns"), py::arg("num_rows"),
+        "Encode columnar wire-form into a framed TFRecord file image");
+  m.def("frame_byte_arrays", &frame_byte_arrays, py::arg("data"), py::arg("elem_off"),
+        "Frame raw byte payloads (ByteArray record type)");
+  m.def("infer_schema_codes", &infer_schema_codes, py::arg("data"), py::arg("rec_off"),
+        py::arg("rec_len"), py::arg("fmt"),
+        "Per-feature type-lattice codes for schema inference");
+  m.attr("FMT_EXAMPLE") = static_cast<int>(FMT_EXAMPLE);
+  m.attr("FMT_SEQUENCE") = static_cast<int>(FMT_SEQUENCE);
+  m.attr("FMT_BYTE_ARRAY") = static_cast<int>(FMT_BYTE_ARRAY);
+#ifdef TFREC_WITH_HIP
+  register_gpu(m);
+  m.attr("HAS_GPU_KERNELS") = true;
+#else
+  m.attr("HAS_GPU_KERNELS") = false;
+#endif
+}

@@ -1,0 +1,355 @@
+// gfx950 (CDNA4) kernels for the TFRecord codec + pybind11 launch wrappers.
+//
+// Design (SURVEY.md §2b native inventory, §7 step 3):
+//  - Records are the parallel axis: one record per lane, grid-stride loops
+//    sized so a full file saturates 256 CUs (wave = 64 lanes; blocks of 256).
+//  - Frame boundaries come from a cheap sequential host header scan
+//    (csrc/ext.cpp scan_frame_headers); the GPU does everything per-record:
+//    masked-CRC32C verification, protobuf structure scan, value extraction,
+//    size/emit encode, and frame construction.
+//  - CRC32C slicing-by-8 tables (8 KiB) are staged in LDS per workgroup;
+//    byte-level parse state stays in registers.
+//  - Prefix sums between passes run as torch.cumsum on the same stream
+//    (Python orchestration in spark_tfrecord_amd/engine/gpu.py).
+//
+// All parsing/emit logic is shared with the host via csrc/codec_core.h — the
+// kernels only add the parallel decomposition and memory staging.
+
+#include <hip/hip_runtime.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "../codec_core.h"
+
+namespace py = pybind11;
+using namespace tfrec;
+
+#define HIP_CHECK(expr)                                                          \
+  do {                                                                           \
+    hipError_t _e = (expr);                                                      \
+    if (_e != hipSuccess)                                                        \
+      throw std::runtime_error(std::string("HIP error: ") +                      \
+                               hipGetErrorString(_e) + " at " __FILE__ ":" +     \
+                               std::to_string(__LINE__));                        \
+  } while (0)
+
+namespace {
+
+constexpr int kBlock = 256;
+
+__device__ inline void stage_crc_tables(uint32_t (*lds)[256]) {
+  const uint32_t* src = &kCrcTables.t[0][0];
+  uint32_t* dst = &lds[0][0];
+  for (int i = threadIdx.x; i < 8 * 256; i += blockDim.x) dst[i] = src[i];
+  __syncthreads();
+}
+
+// ---------------------------------------------------------------------------
+// CRC verification: one record per lane. err_out[0] stays 0 on success, else
+// holds 1 + first-bad-record index (atomicMin keeps the earliest).
+// ---------------------------------------------------------------------------
+
+__global__ void crc_verify_kernel(const u8* __restrict__ data,
+                                  const i64* __restrict__ off,
+                                  const i64* __restrict__ len, i64 R,
+                                  unsigned long long* err_out) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    const u8* h = data + off[r] - 12;  // frame header precedes payload
+    u32 len_crc, data_crc;
+    __builtin_memcpy(&len_crc, h + 8, 4);
+    __builtin_memcpy(&data_crc, h + 12 + len[r], 4);
+    bool ok = mask_crc(crc32c_sw(h, 8, 0, tab)) == len_crc &&
+              mask_crc(crc32c_sw(h + 12, (size_t)len[r], 0, tab)) == data_crc;
+    if (!ok) atomicMin(err_out, (unsigned long long)(r + 1));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Decode pass A: structure scan. Stats land directly in a global [R][F]
+// FieldStat buffer (48 B = 6 int64 words each) — no per-thread scratch array;
+// Python views the buffer as an int64 [R,F,6] tensor for torch prefix sums.
+// ---------------------------------------------------------------------------
+
+static_assert(sizeof(FieldStat) == 48, "FieldStat layout shared with Python");
+
+__global__ void scan_records_kernel(const u8* __restrict__ data,
+                                    const i64* __restrict__ off,
+                                    const i64* __restrict__ len, i64 R, int32_t fmt,
+                                    const u8* __restrict__ schema_blob, int F,
+                                    FieldStat* __restrict__ stats,
+                                    int32_t* __restrict__ err) {
+  SchemaView schema = schema_view(schema_blob);
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    FieldStat* st = stats + r * F;
+    for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+    int32_t rc = scan_record(data, off[r], len[r], fmt, schema, st);
+    if (rc != ERR_OK) err[0] = rc;
+    for (int f = 0; f < F; ++f)
+      if (st[f].err != ERR_OK) err[0] = st[f].err;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Decode pass B: value extraction, one (record, field) pair per lane.
+// Row bases come from the exclusive prefix sums ([F][R+1], row r's base at
+// index r, i.e. cumsum shifted by one).
+// ---------------------------------------------------------------------------
+
+struct DevFieldDst {
+  int32_t kind;
+  int32_t is_seq;
+  i64* i64_vals;
+  float* f32_vals;
+  u8* bytes_data;
+  i64* elem_len;
+  i64* sub_count;
+  const i64* val_base;   // [R+1] exclusive scan of nvals
+  const i64* byte_base;  // [R+1] exclusive scan of nbytes
+  const i64* list_base;  // [R+1] exclusive scan of nlists
+};
+
+__global__ void extract_fields_kernel(const u8* __restrict__ data, i64 R, int F,
+                                      const FieldStat* __restrict__ stats,
+                                      const DevFieldDst* __restrict__ metas,
+                                      int32_t* __restrict__ err) {
+  i64 total = R * F;
+  for (i64 idx = blockIdx.x * (i64)blockDim.x + threadIdx.x; idx < total;
+       idx += (i64)gridDim.x * blockDim.x) {
+    i64 r = idx / F;
+    int f = (int)(idx - r * F);
+    const DevFieldDst& m = metas[f];
+    const FieldStat& st = stats[r * F + f];
+    DecodeDst dst{m.i64_vals, m.f32_vals, m.bytes_data, m.elem_len, m.sub_count};
+    int32_t rc = extract_field(data, st.pos, st.len, m.kind, m.is_seq, dst,
+                               m.val_base ? m.val_base[r] : 0,
+                               m.byte_base ? m.byte_base[r] : 0,
+                               m.list_base ? m.list_base[r] : 0);
+    if (rc != ERR_OK) err[0] = rc;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Encode: size pass (thread per record), emit pass, frame+CRC pass.
+// ---------------------------------------------------------------------------
+
+constexpr int kMaxFieldsEnc = 64;
+
+struct DevCols {
+  int32_t nfields;
+  FieldColumn c[kMaxFieldsEnc];
+};
+
+__global__ void size_records_kernel(const DevCols* __restrict__ cols,
+                                    const u8* __restrict__ schema_blob, int32_t fmt,
+                                    i64 R, i64* __restrict__ psize) {
+  SchemaView schema = schema_view(schema_blob);
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    psize[r] = record_payload_size(cols->c, schema, fmt, r) + kFrameOverhead;
+  }
+}
+
+// frame_off is the exclusive scan of psize (includes frame overhead).
+__global__ void emit_records_kernel(const DevCols* __restrict__ cols,
+                                    const u8* __restrict__ schema_blob, int32_t fmt,
+                                    i64 R, const i64* __restrict__ frame_off,
+                                    u8* __restrict__ file,
+                                    int32_t* __restrict__ err) {
+  SchemaView schema = schema_view(schema_blob);
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    i64 payload = (frame_off[r + 1] - frame_off[r]) - kFrameOverhead;
+    u8* o = file + frame_off[r] + 12;
+    u8* oe = emit_record_payload(o, cols->c, schema, fmt, r);
+    if (oe - o != payload) err[0] = ERR_OVERFLOW;
+  }
+}
+
+__global__ void frame_crc_kernel(u8* __restrict__ file,
+                                 const i64* __restrict__ frame_off, i64 R) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    i64 payload = (frame_off[r + 1] - frame_off[r]) - kFrameOverhead;
+    write_frame_header_footer(file, frame_off[r], payload, tab);
+  }
+}
+
+// ByteArray framing: payload r occupies [elem_off[r], elem_off[r+1]) of src.
+__global__ void frame_bytes_kernel(const u8* __restrict__ src,
+                                   const i64* __restrict__ elem_off,
+                                   const i64* __restrict__ frame_off, i64 R,
+                                   u8* __restrict__ file) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    i64 n = elem_off[r + 1] - elem_off[r];
+    const u8* s = src + elem_off[r];
+    u8* d = file + frame_off[r] + 12;
+    for (i64 i = 0; i < n; ++i) d[i] = s[i];
+    write_frame_header_footer(file, frame_off[r], n, tab);
+  }
+}
+
+// Raw-payload gather for the ByteArray read path (payload extents -> packed).
+__global__ void gather_payloads_kernel(const u8* __restrict__ data,
+                                       const i64* __restrict__ off,
+                                       const i64* __restrict__ len,
+                                       const i64* __restrict__ dst_off, i64 R,
+                                       u8* __restrict__ out) {
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    const u8* s = data + off[r];
+    u8* d = out + dst_off[r];
+    for (i64 i = 0; i < len[r]; ++i) d[i] = s[i];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Launch wrappers (pointers are caller-owned torch tensors; stream is the
+// torch current stream so the kernels interleave with cumsum/casts).
+// ---------------------------------------------------------------------------
+
+inline int grid_for(i64 n) {
+  // ≫256 workgroups to fill 8 XCDs × 32 CUs; cap so grid-stride amortizes.
+  i64 blocks = (n + kBlock - 1) / kBlock;
+  if (blocks > 8192) blocks = 8192;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+void gpu_crc_verify(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
+                    uintptr_t err_out, uintptr_t stream) {
+  hipLaunchKernelGGL(crc_verify_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)data, (const i64*)off,
+                     (const i64*)len, R, (unsigned long long*)err_out);
+  HIP_CHECK(hipGetLastError());
+}
+
+void gpu_scan_records(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
+                      int32_t fmt, uintptr_t schema_blob, int F, uintptr_t stats,
+                      uintptr_t err, uintptr_t stream) {
+  hipLaunchKernelGGL(scan_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)data, (const i64*)off,
+                     (const i64*)len, R, fmt, (const u8*)schema_blob, F,
+                     (FieldStat*)stats, (int32_t*)err);
+  HIP_CHECK(hipGetLastError());
+}
+
+void gpu_extract_fields(uintptr_t data, i64 R, int F, uintptr_t stats,
+                        py::list metas, uintptr_t meta_dev, uintptr_t err,
+                        uintptr_t stream) {
+  std::vector<DevFieldDst> host_metas(F);
+  for (int f = 0; f < F; ++f) {
+    py::dict d = metas[f].cast<py::dict>();
+    DevFieldDst& m = host_metas[f];
+    m.kind = d["kind"].cast<int32_t>();
+    m.is_seq = d["is_seq"].cast<int32_t>();
+    m.i64_vals = (i64*)d["i64_vals"].cast<uintptr_t>();
+    m.f32_vals = (float*)d["f32_vals"].cast<uintptr_t>();
+    m.bytes_data = (u8*)d["bytes_data"].cast<uintptr_t>();
+    m.elem_len = (i64*)d["elem_len"].cast<uintptr_t>();
+    m.sub_count = (i64*)d["sub_count"].cast<uintptr_t>();
+    m.val_base = (const i64*)d["val_base"].cast<uintptr_t>();
+    m.byte_base = (const i64*)d["byte_base"].cast<uintptr_t>();
+    m.list_base = (const i64*)d["list_base"].cast<uintptr_t>();
+  }
+  HIP_CHECK(hipMemcpyAsync((void*)meta_dev, host_metas.data(),
+                           sizeof(DevFieldDst) * F, hipMemcpyHostToDevice,
+                           (hipStream_t)stream));
+  hipLaunchKernelGGL(extract_fields_kernel, dim3(grid_for(R * F)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)data, R, F,
+                     (const FieldStat*)stats, (const DevFieldDst*)meta_dev,
+                     (int32_t*)err);
+  HIP_CHECK(hipGetLastError());
+}
+
+size_t gpu_devcols_bytes() { return sizeof(DevCols); }
+
+void fill_devcols(py::list cols, DevCols* out) {
+  int F = (int)cols.size();
+  if (F > kMaxFieldsEnc)
+    throw std::invalid_argument("GPU encode supports at most 64 fields");
+  out->nfields = F;
+  for (int f = 0; f < F; ++f) {
+    py::dict d = cols[f].cast<py::dict>();
+    FieldColumn& c = out->c[f];
+    c.kind = d["kind"].cast<int32_t>();
+    c.is_seq = d["is_seq"].cast<int32_t>();
+    c.presence = (const u8*)d["presence"].cast<uintptr_t>();
+    c.row_off = (const i64*)d["row_off"].cast<uintptr_t>();
+    c.list_off = (const i64*)d["list_off"].cast<uintptr_t>();
+    c.sub_off = (const i64*)d["sub_off"].cast<uintptr_t>();
+    c.elem_off = (const i64*)d["elem_off"].cast<uintptr_t>();
+    c.bytes_data = (const u8*)d["values_bytes"].cast<uintptr_t>();
+    c.i64_vals = (const i64*)d["values_i64"].cast<uintptr_t>();
+    c.f32_vals = (const float*)d["values_f32"].cast<uintptr_t>();
+  }
+}
+
+void gpu_size_records(py::list cols, uintptr_t cols_dev, uintptr_t schema_blob,
+                      int32_t fmt, i64 R, uintptr_t psize, uintptr_t stream) {
+  DevCols host_cols{};
+  fill_devcols(cols, &host_cols);
+  HIP_CHECK(hipMemcpyAsync((void*)cols_dev, &host_cols, sizeof(DevCols),
+                           hipMemcpyHostToDevice, (hipStream_t)stream));
+  hipLaunchKernelGGL(size_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const DevCols*)cols_dev,
+                     (const u8*)schema_blob, fmt, R, (i64*)psize);
+  HIP_CHECK(hipGetLastError());
+}
+
+void gpu_emit_records(uintptr_t cols_dev, uintptr_t schema_blob, int32_t fmt, i64 R,
+                      uintptr_t frame_off, uintptr_t file, uintptr_t err,
+                      uintptr_t stream) {
+  hipLaunchKernelGGL(emit_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const DevCols*)cols_dev,
+                     (const u8*)schema_blob, fmt, R, (const i64*)frame_off,
+                     (u8*)file, (int32_t*)err);
+  HIP_CHECK(hipGetLastError());
+  hipLaunchKernelGGL(frame_crc_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (u8*)file, (const i64*)frame_off, R);
+  HIP_CHECK(hipGetLastError());
+}
+
+void gpu_frame_bytes(uintptr_t src, uintptr_t elem_off, uintptr_t frame_off, i64 R,
+                     uintptr_t file, uintptr_t stream) {
+  hipLaunchKernelGGL(frame_bytes_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)src, (const i64*)elem_off,
+                     (const i64*)frame_off, R, (u8*)file);
+  HIP_CHECK(hipGetLastError());
+}
+
+void gpu_gather_payloads(uintptr_t data, uintptr_t off, uintptr_t len,
+                         uintptr_t dst_off, i64 R, uintptr_t out, uintptr_t stream) {
+  hipLaunchKernelGGL(gather_payloads_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)data, (const i64*)off,
+                     (const i64*)len, (const i64*)dst_off, R, (u8*)out);
+  HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace
+
+void register_gpu(py::module_& m) {
+  m.def("gpu_crc_verify", &gpu_crc_verify);
+  m.def("gpu_scan_records", &gpu_scan_records);
+  m.def("gpu_extract_fields", &gpu_extract_fields);
+  m.def("gpu_size_records", &gpu_size_records);
+  m.def("gpu_emit_records", &gpu_emit_records);
+  m.def("gpu_frame_bytes", &gpu_frame_bytes);
+  m.def("gpu_gather_payloads", &gpu_gather_payloads);
+  m.def("gpu_devcols_bytes", &gpu_devcols_bytes);
+  m.def("gpu_devmeta_bytes", []() { return sizeof(DevFieldDst); });
+  m.def("gpu_fieldstat_words", []() { return sizeof(FieldStat) / 8; });
+}

@@ -1,0 +1,37 @@
+"""spark_tfrecord_amd — an MI355X-native TFRecord I/O engine.
+
+A from-scratch re-design of linkedin/spark-tfrecord's capabilities
+(format("tfrecord"), recordType = Example | SequenceExample | ByteArray,
+schema inference with the reference's type-promotion lattice, partitionBy
+writes, save modes, gzip codecs, identical on-disk bytes) with the hot path
+in C++/HIP for CDNA4 (gfx950) and multi-GPU scaling over RCCL/xGMI instead
+of a JVM + Spark. See SURVEY.md for the structural map of the reference.
+"""
+
+from .api import DataFrame, DataFrameReader, DataFrameWriter, TFRecordSession, session
+from .io.reader import read_tfrecord
+from .io.writer import write_tfrecord
+from .schema import (
+    ArrayType,
+    BinaryType,
+    DataType,
+    DecimalType,
+    DoubleType,
+    FloatType,
+    IntegerType,
+    LongType,
+    NullType,
+    StringType,
+    StructField,
+    StructType,
+)
+
+__version__ = "0.1.0"
+
+__all__ = [
+    "session", "DataFrame", "DataFrameReader", "DataFrameWriter",
+    "TFRecordSession", "read_tfrecord", "write_tfrecord",
+    "DataType", "NullType", "IntegerType", "LongType", "FloatType",
+    "DoubleType", "DecimalType", "StringType", "BinaryType", "ArrayType",
+    "StructField", "StructType",
+]

@@ -1,0 +1,340 @@
+"""pyarrow <-> wire-form conversions.
+
+The wire-form (columnar.py) is already Arrow-shaped — ragged offsets + flat
+value buffers — so reads materialize as pyarrow arrays without per-row Python
+work wherever the layout allows, and writes consume arrow buffers directly.
+pandas round-trips ride on arrow.
+
+Type mapping (logical -> arrow):
+    Long -> int64      Integer -> int32     Float -> float32
+    Double -> float64  Decimal -> float64 (wire is float32; lossy by design,
+                                           reference TFRecordSerializer.scala:88-90)
+    String -> large_utf8   Binary -> large_binary
+    Array(T) -> large_list(T)    Array(Array(T)) -> large_list(large_list(T))
+    Null -> null
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import numpy as np
+import pyarrow as pa
+
+from .columnar import RecordBatch, WireColumn, column_from_values, column_to_pylist, wire_fields
+from .schema import (
+    ArrayType,
+    BinaryType,
+    DataType,
+    DecimalType,
+    DoubleType,
+    FloatType,
+    IntegerType,
+    KIND_BYTES,
+    KIND_FLOAT,
+    KIND_INT64,
+    LongType,
+    NullType,
+    StringType,
+    StructField,
+    StructType,
+    is_sequence_field,
+    wire_kind_of,
+)
+
+__all__ = ["arrow_type_for", "datatype_from_arrow", "schema_to_arrow",
+           "schema_from_arrow", "batch_to_table", "table_to_batch"]
+
+
+def arrow_type_for(dt: DataType) -> pa.DataType:
+    if isinstance(dt, NullType):
+        return pa.null()
+    if isinstance(dt, IntegerType):
+        return pa.int32()
+    if isinstance(dt, LongType):
+        return pa.int64()
+    if isinstance(dt, FloatType):
+        return pa.float32()
+    if isinstance(dt, (DoubleType, DecimalType)):
+        return pa.float64()
+    if isinstance(dt, StringType):
+        return pa.large_utf8()
+    if isinstance(dt, BinaryType):
+        return pa.large_binary()
+    if isinstance(dt, ArrayType):
+        return pa.large_list(arrow_type_for(dt.elementType))
+    raise TypeError(f"No arrow mapping for {dt!r}")
+
+
+def datatype_from_arrow(t: pa.DataType) -> DataType:
+    if pa.types.is_null(t):
+        return NullType()
+    if pa.types.is_int8(t) or pa.types.is_int16(t) or pa.types.is_int32(t) or \
+            pa.types.is_uint8(t) or pa.types.is_uint16(t) or pa.types.is_uint32(t):
+        return IntegerType()
+    if pa.types.is_int64(t) or pa.types.is_uint64(t):
+        return LongType()
+    if pa.types.is_float32(t) or pa.types.is_float16(t):
+        return FloatType()
+    if pa.types.is_float64(t):
+        return DoubleType()
+    if pa.types.is_decimal(t):
+        return DecimalType(t.precision, t.scale)
+    if pa.types.is_string(t) or pa.types.is_large_string(t):
+        return StringType()
+    if pa.types.is_binary(t) or pa.types.is_large_binary(t) or \
+            pa.types.is_fixed_size_binary(t):
+        return BinaryType()
+    if pa.types.is_list(t) or pa.types.is_large_list(t) or \
+            pa.types.is_fixed_size_list(t):
+        return ArrayType(datatype_from_arrow(t.value_type))
+    raise TypeError(f"Unsupported arrow type for TFRecord: {t}")
+
+
+def schema_to_arrow(schema: StructType) -> pa.Schema:
+    return pa.schema([pa.field(f.name, arrow_type_for(f.dataType), f.nullable)
+                      for f in schema.fields])
+
+
+def schema_from_arrow(sch: pa.Schema) -> StructType:
+    return StructType([StructField(f.name, datatype_from_arrow(f.type), f.nullable)
+                       for f in sch])
+
+
+def _ragged_gather_idx(starts: np.ndarray, lengths: np.ndarray) -> np.ndarray:
+    """Element indices for gathering ragged extents [starts[i], starts[i]+len[i])."""
+    total = int(lengths.sum())
+    if total == 0:
+        return np.zeros(0, np.int64)
+    dst_off = np.zeros(len(lengths), np.int64)
+    np.cumsum(lengths[:-1], out=dst_off[1:])
+    return np.repeat(starts, lengths) + (np.arange(total, dtype=np.int64)
+                                         - np.repeat(dst_off, lengths))
+
+
+def _validity(presence: np.ndarray):
+    mask = presence.astype(bool)
+    if mask.all():
+        return None
+    return np.packbits(mask, bitorder="little")
+
+
+# ---------------------------------------------------------------------------
+# wire -> arrow
+# ---------------------------------------------------------------------------
+
+def _numeric_np(col: WireColumn, dt: DataType, idx=None) -> np.ndarray:
+    vals = np.asarray(col.values)
+    if idx is not None:
+        vals = vals[idx]
+    if isinstance(dt, IntegerType):
+        return vals.astype(np.int32)
+    if isinstance(dt, LongType):
+        return vals.astype(np.int64, copy=False)
+    if isinstance(dt, FloatType):
+        return vals.astype(np.float32, copy=False)
+    if isinstance(dt, (DoubleType, DecimalType)):
+        return vals.astype(np.float64)
+    raise TypeError(dt)
+
+
+def _bytes_arrow(col: WireColumn, dt: DataType, elem_idx: np.ndarray) -> pa.Array:
+    """Build a large_utf8/large_binary array for the given string elements."""
+    elem_off = np.asarray(col.elem_off)
+    data = np.asarray(col.values)
+    starts = elem_off[elem_idx]
+    lengths = elem_off[elem_idx + 1] - starts
+    gathered = data[_ragged_gather_idx(starts, lengths)]
+    new_off = np.zeros(len(elem_idx) + 1, np.int64)
+    np.cumsum(lengths, out=new_off[1:])
+    t = pa.large_utf8() if isinstance(dt, StringType) else pa.large_binary()
+    return pa.Array.from_buffers(
+        t, len(elem_idx),
+        [None, pa.py_buffer(new_off), pa.py_buffer(gathered)])
+
+
+def wire_to_arrow(col: WireColumn, dt: DataType, nullable: bool, name: str,
+                  num_rows: int) -> pa.Array:
+    if isinstance(dt, NullType):
+        return pa.nulls(num_rows)
+    expect = wire_kind_of(dt)
+    if col.kind != expect:
+        raise RuntimeError(
+            f"Feature '{name}' kind does not match requested type {dt.simple_string()}")
+    seq = is_sequence_field(dt)
+    if seq != col.is_seq:
+        raise RuntimeError(f"Feature '{name}' dimensionality does not match schema")
+    presence = np.asarray(col.presence)
+    mask = presence.astype(bool)
+    if not nullable and not mask.all():
+        r = int(np.argmin(mask))
+        raise ValueError(f"Feature '{name}' is required but missing (row {r})")
+    row_off = np.asarray(col.row_off).astype(np.int64, copy=False)
+    row_len = row_off[1:] - row_off[:-1]
+    valid_buf = _validity(presence)
+
+    if seq:
+        # 2-D ragged: rows -> lists (sub-lists) -> values
+        list_off = np.asarray(col.list_off).astype(np.int64, copy=False)
+        sub_off = np.asarray(col.sub_off).astype(np.int64, copy=False)
+        inner_dt = dt.elementType.elementType
+        if col.kind == KIND_BYTES:
+            n_elems = len(np.asarray(col.elem_off)) - 1
+            inner_vals = _bytes_arrow(col, inner_dt, np.arange(n_elems, dtype=np.int64))
+        else:
+            inner_vals = pa.array(_numeric_np(col, inner_dt))
+        inner = pa.LargeListArray.from_arrays(sub_off, inner_vals)
+        buffers = [None if valid_buf is None else pa.py_buffer(valid_buf),
+                   pa.py_buffer(list_off)]
+        return pa.Array.from_buffers(pa.large_list(inner.type), num_rows, buffers,
+                                     children=[inner])
+
+    if isinstance(dt, ArrayType):
+        inner_dt = dt.elementType
+        if col.kind == KIND_BYTES:
+            n_elems = len(np.asarray(col.elem_off)) - 1
+            inner_vals = _bytes_arrow(col, inner_dt, np.arange(n_elems, dtype=np.int64))
+        else:
+            inner_vals = pa.array(_numeric_np(col, inner_dt))
+        buffers = [None if valid_buf is None else pa.py_buffer(valid_buf),
+                   pa.py_buffer(row_off)]
+        return pa.Array.from_buffers(pa.large_list(inner_vals.type), num_rows, buffers,
+                                     children=[inner_vals])
+
+    # Scalar: head element per present row (reference head semantics).
+    if bool((mask & (row_len == 0)).any()):
+        r = int(np.argmax(mask & (row_len == 0)))
+        raise ValueError(
+            f"Feature '{name}' is present but empty; cannot read scalar (row {r})")
+    heads = row_off[:-1][mask]
+    if col.kind == KIND_BYTES:
+        present_arr = _bytes_arrow(col, dt, heads)
+        if valid_buf is None:
+            return present_arr
+        # Scatter present strings into a full-length array with nulls.
+        full_off = np.zeros(num_rows + 1, np.int64)
+        lens = np.zeros(num_rows, np.int64)
+        elem_off = np.asarray(col.elem_off)
+        lens[mask] = elem_off[heads + 1] - elem_off[heads]
+        np.cumsum(lens, out=full_off[1:])
+        data = np.asarray(col.values)[
+            _ragged_gather_idx(elem_off[heads], lens[mask])]
+        t = pa.large_utf8() if isinstance(dt, StringType) else pa.large_binary()
+        return pa.Array.from_buffers(
+            t, num_rows, [pa.py_buffer(valid_buf), pa.py_buffer(full_off),
+                          pa.py_buffer(data)])
+    head_vals = _numeric_np(col, dt, idx=heads)
+    if valid_buf is None:
+        return pa.array(head_vals)
+    full = np.zeros(num_rows, head_vals.dtype)
+    full[mask] = head_vals
+    return pa.array(full, mask=~mask)
+
+
+def batch_to_table(batch: RecordBatch) -> pa.Table:
+    arrays = []
+    names = []
+    cols = {f.name: c for f, c in zip(wire_fields(batch.schema), batch.columns)}
+    for f in batch.schema.fields:
+        names.append(f.name)
+        if isinstance(f.dataType, NullType):
+            arrays.append(pa.nulls(batch.num_rows))
+        else:
+            arrays.append(wire_to_arrow(cols[f.name], f.dataType, f.nullable, f.name,
+                                        batch.num_rows))
+    return pa.table(arrays, names=names)
+
+
+# ---------------------------------------------------------------------------
+# arrow -> wire
+# ---------------------------------------------------------------------------
+
+def _np_offsets(arr) -> np.ndarray:
+    """List-array offsets as int64 numpy (handles list vs large_list)."""
+    off_buf = arr.buffers()[1]
+    width = 8 if pa.types.is_large_list(arr.type) else 4
+    dt = np.int64 if width == 8 else np.int32
+    off = np.frombuffer(off_buf, dtype=dt, count=len(arr) + 1 + arr.offset)
+    return off[arr.offset:].astype(np.int64, copy=False)
+
+
+def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> WireColumn:
+    if isinstance(arr, pa.ChunkedArray):
+        arr = arr.combine_chunks()
+    kind = wire_kind_of(dt)
+    seq = is_sequence_field(dt)
+    R = len(arr)
+    null_count = arr.null_count
+    if null_count and not nullable:
+        raise ValueError(f"null value in non-nullable field '{name}'")
+
+    mask = np.ones(R, bool) if null_count == 0 else ~np.asarray(arr.is_null())
+    presence = mask.astype(np.uint8)
+
+    def numeric_cast(np_vals):
+        if kind == KIND_INT64:
+            return np_vals.astype(np.int64, copy=False)
+        return np_vals.astype(np.float32)  # Double/Decimal -> float32 downcast
+
+    # Fast path: numeric scalar
+    if not seq and not isinstance(dt, ArrayType) and kind != KIND_BYTES and \
+            (pa.types.is_integer(arr.type) or pa.types.is_floating(arr.type)):
+        if null_count == 0:
+            vals = numeric_cast(arr.to_numpy(zero_copy_only=False))
+            return WireColumn(kind, False, presence,
+                              np.arange(R + 1, dtype=np.int64), vals)
+        vals = numeric_cast(arr.drop_null().to_numpy(zero_copy_only=False))
+        row_off = np.zeros(R + 1, np.int64)
+        np.cumsum(presence.astype(np.int64), out=row_off[1:])
+        return WireColumn(kind, False, presence, row_off, vals)
+
+    # Fast path: string/binary scalar
+    if not seq and not isinstance(dt, ArrayType) and kind == KIND_BYTES and \
+            (pa.types.is_string(arr.type) or pa.types.is_large_string(arr.type) or
+             pa.types.is_binary(arr.type) or pa.types.is_large_binary(arr.type)):
+        a = arr.cast(pa.large_binary())
+        off = np.frombuffer(a.buffers()[1], np.int64, len(a) + 1 + a.offset)[a.offset:]
+        data_buf = a.buffers()[2]
+        data = np.frombuffer(data_buf, np.uint8, len(data_buf)) if data_buf else np.zeros(0, np.uint8)
+        lens = (off[1:] - off[:-1]).astype(np.int64)
+        lens_present = lens[mask]
+        starts = off[:-1][mask].astype(np.int64)
+        gathered = data[_ragged_gather_idx(starts, lens_present)]
+        elem_off = np.zeros(int(mask.sum()) + 1, np.int64)
+        np.cumsum(lens_present, out=elem_off[1:])
+        row_off = np.zeros(R + 1, np.int64)
+        np.cumsum(presence.astype(np.int64), out=row_off[1:])
+        return WireColumn(kind, False, presence, row_off, gathered, elem_off)
+
+    # Fast path: 1-D numeric list
+    if not seq and isinstance(dt, ArrayType) and kind != KIND_BYTES and \
+            (pa.types.is_list(arr.type) or pa.types.is_large_list(arr.type)):
+        off = _np_offsets(arr)
+        lens = off[1:] - off[:-1]
+        lens = np.where(mask, lens, 0)
+        vals_arr = arr.values  # child array
+        child = vals_arr.to_numpy(zero_copy_only=False)
+        if null_count == 0 and arr.offset == 0 and off[0] == 0:
+            vals = numeric_cast(child[: off[-1]])
+            return WireColumn(kind, False, presence, off.copy(), vals)
+        starts = off[:-1][mask]
+        idx = _ragged_gather_idx(starts.astype(np.int64), lens[mask])
+        vals = numeric_cast(child[idx])
+        row_off = np.zeros(R + 1, np.int64)
+        np.cumsum(lens, out=row_off[1:])
+        return WireColumn(kind, False, presence, row_off, vals)
+
+    # General fallback: python objects
+    return column_from_values(arr.to_pylist(), dt, nullable, name)
+
+
+def table_to_batch(table: pa.Table, schema: StructType) -> RecordBatch:
+    cols = []
+    for f in wire_fields(schema):
+        if f.name not in table.column_names:
+            raise KeyError(f"column '{f.name}' not found in input data")
+        arr = table.column(f.name)
+        if isinstance(arr, pa.ChunkedArray):
+            arr = arr.combine_chunks()
+        cols.append(arrow_to_wire(arr, f.dataType, f.nullable, f.name))
+    return RecordBatch(schema, cols, table.num_rows)

@@ -1,0 +1,175 @@
+"""File layout, codecs, save modes, and partition discovery.
+
+Mirrors the on-disk conventions the reference inherits from Spark's
+FileFormatWriter / commit protocol (SURVEY.md §3.3, §5 "Failure detection"):
+  - output dir contains part files `part-<shard>-<uuid>.tfrecord[.gz]`
+  - `partitionBy` produces `col=value/` subdirectories
+  - `_SUCCESS` marker written on job completion
+  - save modes: error (default), overwrite, append, ignore
+    (TFRecordIOSuite.scala:184-237 semantics)
+  - `codec` option on write; read-side codec inferred from file extension
+    (DefaultSource.scala:95-102)
+Writes go to a temp file + atomic rename (idempotent shard retry).
+"""
+
+from __future__ import annotations
+
+import glob as _glob
+import gzip
+import os
+import shutil
+import uuid
+import zlib
+from typing import Dict, List, Optional, Tuple
+
+__all__ = ["normalize_codec", "codec_extension", "compress_bytes",
+           "decompress_file", "list_data_files", "partition_values_of",
+           "apply_save_mode", "part_file_name", "write_file_atomic",
+           "write_success_marker", "SaveModeError"]
+
+# codec option values accepted, mirroring Hadoop codec class names + shortcuts
+_CODEC_ALIASES = {
+    "gzip": "gzip",
+    "org.apache.hadoop.io.compress.gzipcodec": "gzip",
+    "deflate": "deflate",
+    "org.apache.hadoop.io.compress.deflatecodec": "deflate",
+    "org.apache.hadoop.io.compress.defaultcodec": "deflate",
+    "none": None,
+    "uncompressed": None,
+}
+
+_EXTENSIONS = {"gzip": ".gz", "deflate": ".deflate"}
+_EXT_TO_CODEC = {".gz": "gzip", ".deflate": "deflate"}
+
+
+class SaveModeError(RuntimeError):
+    pass
+
+
+def normalize_codec(codec: Optional[str]) -> Optional[str]:
+    if codec is None:
+        return None
+    key = codec.strip().lower()
+    if key in _CODEC_ALIASES:
+        return _CODEC_ALIASES[key]
+    raise ValueError(f"Unknown compression codec: {codec!r}")
+
+
+def codec_extension(codec: Optional[str]) -> str:
+    return _EXTENSIONS.get(codec, "")
+
+
+def codec_from_path(path: str) -> Optional[str]:
+    _, ext = os.path.splitext(path)
+    return _EXT_TO_CODEC.get(ext)
+
+
+def compress_bytes(data: bytes, codec: Optional[str]) -> bytes:
+    if codec is None:
+        return data
+    if codec == "gzip":
+        return gzip.compress(data, compresslevel=6)
+    if codec == "deflate":
+        return zlib.compress(data, 6)
+    raise ValueError(codec)
+
+
+def decompress_file(path: str) -> bytes:
+    codec = codec_from_path(path)
+    with open(path, "rb") as f:
+        raw = f.read()
+    if codec is None:
+        return raw
+    if codec == "gzip":
+        return gzip.decompress(raw)
+    if codec == "deflate":
+        return zlib.decompress(raw)
+    raise ValueError(codec)
+
+
+def part_file_name(shard: int, codec: Optional[str], job_id: str) -> str:
+    return f"part-{shard:05d}-{job_id}.tfrecord{codec_extension(codec)}"
+
+
+def _is_data_file(name: str) -> bool:
+    base = os.path.basename(name)
+    return not (base.startswith("_") or base.startswith("."))
+
+
+def list_data_files(path: str) -> List[str]:
+    """Resolve a path/glob/dir into a sorted list of data files, recursing
+    into partition directories."""
+    paths: List[str] = []
+    candidates = _glob.glob(path) if _glob.has_magic(path) else [path]
+    if _glob.has_magic(path) and not candidates:
+        raise FileNotFoundError(f"Path does not exist: {path}")
+    for p in candidates:
+        if os.path.isdir(p):
+            for root, dirs, files in os.walk(p):
+                dirs[:] = sorted(d for d in dirs if _is_data_file(d))
+                for fn in sorted(files):
+                    if _is_data_file(fn):
+                        paths.append(os.path.join(root, fn))
+        elif os.path.isfile(p):
+            paths.append(p)
+        else:
+            raise FileNotFoundError(f"Path does not exist: {p}")
+    return sorted(paths)
+
+
+def partition_values_of(file_path: str, base_dir: str) -> Dict[str, str]:
+    """Extract `col=value` partition components between base_dir and the file."""
+    rel = os.path.relpath(os.path.dirname(os.path.abspath(file_path)),
+                          os.path.abspath(base_dir))
+    out: Dict[str, str] = {}
+    if rel in (".", ""):
+        return out
+    for comp in rel.split(os.sep):
+        if "=" in comp:
+            k, v = comp.split("=", 1)
+            out[k] = v
+    return out
+
+
+def apply_save_mode(path: str, mode: str) -> bool:
+    """Prepare the output dir for the given save mode.
+
+    Returns True when the write should proceed, False for ignore-and-skip.
+    Mirrors Spark SaveMode semantics exercised in TFRecordIOSuite.scala:184-237.
+    """
+    mode = mode.lower().replace("_", "")
+    exists = os.path.exists(path) and (not os.path.isdir(path) or os.listdir(path))
+    if mode in ("error", "errorifexists", "default"):
+        if exists:
+            raise SaveModeError(f"path already exists: {path}")
+    elif mode == "overwrite":
+        if os.path.isdir(path):
+            shutil.rmtree(path)
+        elif os.path.exists(path):
+            os.remove(path)
+    elif mode == "append":
+        pass
+    elif mode == "ignore":
+        if exists:
+            return False
+    else:
+        raise ValueError(f"Unknown save mode: {mode}")
+    os.makedirs(path, exist_ok=True)
+    return True
+
+
+def write_file_atomic(data: bytes, final_path: str):
+    """Temp file + rename: a torn write never becomes visible (the engine's
+    stand-in for Spark's task-commit rename protocol)."""
+    os.makedirs(os.path.dirname(final_path), exist_ok=True)
+    tmp = final_path + f".__tmp.{uuid.uuid4().hex[:8]}"
+    with open(tmp, "wb") as f:
+        f.write(data)
+        f.flush()
+        os.fsync(f.fileno())
+    os.replace(tmp, final_path)
+
+
+def write_success_marker(path: str):
+    with open(os.path.join(path, "_SUCCESS"), "wb"):
+        pass

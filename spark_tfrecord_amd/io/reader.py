@@ -1,0 +1,127 @@
+"""Read path: TFRecord files -> DataFrame.
+
+Mirrors the reference's read pipeline: file discovery
+(whole files, never split — gzip streams can't be, DefaultSource.scala:26-29),
+schema inference from the first non-empty file when no schema is given,
+per-file decode, and partition-column discovery from `col=value/` directory
+names (the part Spark's planner did above the reference library).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional
+
+import numpy as np
+import pyarrow as pa
+
+from .. import engine as engine_mod
+from ..arrow_interop import batch_to_table, schema_to_arrow
+from ..engine import cpu as cpu_engine
+from ..infer import (
+    byte_array_schema,
+    infer_codes_from_buffer,
+    merge_code_maps,
+    schema_from_codes,
+)
+from ..schema import LongType, StringType, StructField, StructType
+from .. import _native
+from . import paths as P
+
+__all__ = ["read_tfrecord", "infer_schema_of_paths"]
+
+
+def _load_file(path: str) -> np.ndarray:
+    return np.frombuffer(P.decompress_file(path), np.uint8)
+
+
+def infer_schema_of_paths(files: List[str], record_type: str) -> StructType:
+    """Schema from the FIRST non-empty file (DefaultSource.scala:36-38
+    collectFirst), scanned fully."""
+    if record_type == "ByteArray":
+        return byte_array_schema()
+    for f in files:
+        data = _load_file(f)
+        if data.size == 0:
+            continue
+        off, lens = _native.scan_frames(data, False)
+        if len(off) == 0:
+            continue
+        codes = infer_codes_from_buffer(data, off, lens, record_type)
+        if codes:
+            return schema_from_codes(codes)
+        return StructType([])
+    raise ValueError("Could not infer schema: no non-empty TFRecord files found")
+
+
+def _partition_schema(files: List[str], base_dir: str) -> List[str]:
+    cols: List[str] = []
+    for f in files:
+        for k in P.partition_values_of(f, base_dir):
+            if k not in cols:
+                cols.append(k)
+    return cols
+
+
+def _partition_col_array(values: List[str], n: int):
+    """Spark-style partition value type inference: int64 if all parse, else
+    string."""
+    try:
+        return pa.array([int(v) for v in values], type=pa.int64())
+    except (ValueError, TypeError):
+        return pa.array(
+            [None if v == "__HIVE_DEFAULT_PARTITION__" else v for v in values],
+            type=pa.large_utf8())
+
+
+def read_tfrecord(path: str, schema: Optional[StructType] = None,
+                  record_type: str = "Example", engine: str = "auto",
+                  verify_crc: bool = True):
+    from ..api import DataFrame
+
+    if record_type not in ("Example", "SequenceExample", "ByteArray"):
+        raise ValueError(f"Unsupported recordType {record_type!r}")
+    files = P.list_data_files(path)
+    if not files:
+        raise FileNotFoundError(f"No TFRecord files found under {path}")
+    base_dir = path if os.path.isdir(path) else os.path.dirname(path)
+    part_cols = _partition_schema(files, base_dir) if base_dir else []
+    eng = engine_mod.resolve_engine(engine)
+
+    if schema is None:
+        schema = (byte_array_schema() if record_type == "ByteArray"
+                  else infer_schema_of_paths(files, record_type))
+    data_schema = StructType([f for f in schema.fields if f.name not in part_cols])
+
+    tables = []
+    for fpath in files:
+        data = _load_file(fpath)
+        if data.size == 0:
+            continue
+        if eng == "gpu":
+            from ..engine import gpu as gpu_engine
+            batch = gpu_engine.decode_buffer_to_cpu(data, data_schema, record_type,
+                                                    verify_crc=verify_crc)
+        else:
+            batch = cpu_engine.decode_buffer(data, data_schema, record_type,
+                                             verify_crc=verify_crc)
+        t = batch_to_table(batch)
+        pv = P.partition_values_of(fpath, base_dir)
+        for c in part_cols:
+            vals = [pv.get(c)] * t.num_rows
+            t = t.append_column(c, _partition_col_array(vals, t.num_rows))
+        tables.append(t)
+    if not tables:
+        full = StructType(list(data_schema.fields) +
+                          [StructField(c, StringType(), True) for c in part_cols])
+        empty = pa.table({f.name: [] for f in full.fields},
+                         schema=schema_to_arrow(full))
+        return DataFrame(empty, full)
+
+    # Partition column types must agree across files: rebuild as common type
+    table = pa.concat_tables(tables, promote_options="permissive")
+    full_schema = StructType(list(data_schema.fields))
+    for c in part_cols:
+        at = table.schema.field(c).type
+        full_schema.add(c, LongType() if pa.types.is_integer(at) else StringType())
+    return DataFrame(table, full_schema)

@@ -313,7 +313,8 @@ def _scalar_convert(dt: DataType, raw):
 
 def _bytes_convert(dt: DataType, b: bytes):
     if isinstance(dt, StringType):
-        return b.decode("utf-8")
+        # replacement chars on invalid UTF-8, like protobuf-java's toStringUtf8
+        return b.decode("utf-8", errors="replace")
     if isinstance(dt, BinaryType):
         return b
     raise TypeError(f"Unsupported bytes type {dt!r}")

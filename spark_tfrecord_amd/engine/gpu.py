@@ -1,0 +1,293 @@
+"""GPU engine: gfx950 kernel pipeline orchestration.
+
+Decode: host header scan -> H2D -> [crc_verify | scan_records] -> torch
+prefix sums -> extract_fields -> device wire-form columns.
+Encode: device wire-form -> size_records -> prefix sum -> emit + frame CRC
+-> device file image (D2H only at the file boundary).
+
+All launches go on the torch current stream, so the cumsum/cast glue and the
+kernels form one in-order pipeline; prefix sums over per-record counts are
+torch ops on the same stream.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from .. import _native
+from ..columnar import RecordBatch, WireColumn, schema_blob, wire_fields
+from ..schema import KIND_BYTES, KIND_FLOAT, KIND_INT64, StructType, is_sequence_field, wire_kind_of
+
+FMT = {"Example": _native.FMT_EXAMPLE, "SequenceExample": _native.FMT_SEQUENCE}
+
+_U64_MAX = 0xFFFFFFFFFFFFFFFF
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _dev(x, dtype, device="cuda") -> torch.Tensor:
+    if isinstance(x, torch.Tensor):
+        return x.to(device=device, dtype=dtype).contiguous()
+    return torch.as_tensor(np.ascontiguousarray(x), dtype=dtype).to(device)
+
+
+def check_native():
+    if not getattr(_native, "HAS_GPU_KERNELS", False):
+        raise RuntimeError("_native was built without HIP kernels; rebuild "
+                           "with build_native.py before using the GPU engine")
+
+
+# ---------------------------------------------------------------------------
+# Decode
+# ---------------------------------------------------------------------------
+
+def crc_verify_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor):
+    """Raise on any bad frame CRC (parallel over records on the GPU)."""
+    R = off.numel()
+    if R == 0:
+        return
+    err = torch.full((1,), -1, dtype=torch.int64, device=data.device)  # all-ones
+    _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(), lens.data_ptr(), R,
+                           err.data_ptr(), _stream())
+    bad = int(err.item())  # syncs
+    if bad != -1:
+        raise RuntimeError(
+            f"corrupt TFRecord: bad CRC in record {bad - 1 if bad > 0 else bad}")
+
+
+def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
+                  schema: StructType, record_type: str,
+                  verify_crc: bool = True) -> RecordBatch:
+    """Decode framed records already resident in HBM into device wire-form."""
+    check_native()
+    device = data.device
+    R = off.numel()
+    fields = wire_fields(schema)
+    F = len(fields)
+    if verify_crc:
+        crc_verify_device(data, off, lens)
+
+    if record_type == "ByteArray":
+        dst_off = torch.zeros(R + 1, dtype=torch.int64, device=device)
+        torch.cumsum(lens, 0, out=dst_off[1:])
+        total = int(dst_off[-1].item())
+        out = torch.empty(total, dtype=torch.uint8, device=device)
+        if R:
+            _native.gpu_gather_payloads(data.data_ptr(), off.data_ptr(),
+                                        lens.data_ptr(), dst_off.data_ptr(), R,
+                                        out.data_ptr(), _stream())
+        col = WireColumn(kind=KIND_BYTES, is_seq=False,
+                         presence=torch.ones(R, dtype=torch.uint8, device=device),
+                         row_off=torch.arange(R + 1, dtype=torch.int64, device=device),
+                         values=out, elem_off=dst_off)
+        return RecordBatch(schema, [col], R)
+
+    if F == 0 or R == 0:
+        cols = [WireColumn(wire_kind_of(f.dataType), is_sequence_field(f.dataType),
+                           torch.zeros(R, dtype=torch.uint8, device=device),
+                           torch.zeros(R + 1, dtype=torch.int64, device=device),
+                           torch.zeros(0, dtype=torch.int64, device=device))
+                for f in fields]
+        return RecordBatch(schema, cols, R)
+
+    blob = torch.frombuffer(bytearray(schema_blob(schema)),
+                            dtype=torch.uint8).to(device)
+    # FieldStat[R][F] as int64 [R,F,6]: pos,len,nvals,nbytes,nlists,(kind|err)
+    stats = torch.empty((R, F, 6), dtype=torch.int64, device=device)
+    err = torch.zeros(1, dtype=torch.int32, device=device)
+    _native.gpu_scan_records(data.data_ptr(), off.data_ptr(), lens.data_ptr(), R,
+                             FMT[record_type], blob.data_ptr(), F,
+                             stats.data_ptr(), err.data_ptr(), _stream())
+
+    # Per-field exclusive prefix sums ([F, R+1], row-contiguous views)
+    def excl_scan(counts_rf: torch.Tensor) -> torch.Tensor:
+        base = torch.zeros((F, R + 1), dtype=torch.int64, device=device)
+        torch.cumsum(counts_rf.transpose(0, 1), dim=1, out=base[:, 1:])
+        return base
+
+    val_base = excl_scan(stats[:, :, 2])
+    byte_base = excl_scan(stats[:, :, 3])
+    list_base = excl_scan(stats[:, :, 4])
+    totals = torch.stack([val_base[:, -1], byte_base[:, -1], list_base[:, -1]])
+    totals_h = totals.cpu()  # one sync for all allocations
+    if int(err.item()) != 0:
+        raise RuntimeError(f"TFRecord decode failed (native error "
+                           f"{int(err.item())}; kind mismatch or malformed record)")
+
+    metas = []
+    outs = []
+    for i, f in enumerate(fields):
+        kind = wire_kind_of(f.dataType)
+        seq = is_sequence_field(f.dataType)
+        nv = int(totals_h[0, i])
+        nb = int(totals_h[1, i])
+        nl = int(totals_h[2, i])
+        o = {"kind": kind, "seq": seq}
+        o["i64_vals"] = torch.empty(nv if kind == KIND_INT64 else 0,
+                                    dtype=torch.int64, device=device)
+        o["f32_vals"] = torch.empty(nv if kind == KIND_FLOAT else 0,
+                                    dtype=torch.float32, device=device)
+        o["bytes_data"] = torch.empty(nb, dtype=torch.uint8, device=device)
+        o["elem_len"] = torch.empty(nv if kind == KIND_BYTES else 0,
+                                    dtype=torch.int64, device=device)
+        o["sub_count"] = torch.empty(nl, dtype=torch.int64, device=device)
+        outs.append(o)
+        metas.append({
+            "kind": kind, "is_seq": 1 if seq else 0,
+            "i64_vals": o["i64_vals"].data_ptr(),
+            "f32_vals": o["f32_vals"].data_ptr(),
+            "bytes_data": o["bytes_data"].data_ptr(),
+            "elem_len": o["elem_len"].data_ptr(),
+            "sub_count": o["sub_count"].data_ptr(),
+            "val_base": val_base[i].data_ptr(),
+            "byte_base": byte_base[i].data_ptr(),
+            "list_base": list_base[i].data_ptr() if seq else 0,
+        })
+
+    meta_dev = torch.empty(F * _native.gpu_devmeta_bytes(), dtype=torch.uint8,
+                           device=device)
+    _native.gpu_extract_fields(data.data_ptr(), R, F, stats.data_ptr(), metas,
+                               meta_dev.data_ptr(), err.data_ptr(), _stream())
+
+    cols: List[WireColumn] = []
+    for i, (f, o) in enumerate(zip(fields, outs)):
+        kind = o["kind"]
+        seq = o["seq"]
+        presence = (stats[:, i, 0] >= 0).to(torch.uint8)
+        elem_off = None
+        if kind == KIND_BYTES:
+            elem_off = torch.zeros(o["elem_len"].numel() + 1, dtype=torch.int64,
+                                   device=device)
+            torch.cumsum(o["elem_len"], 0, out=elem_off[1:])
+        sub_off = None
+        if seq:
+            sub_off = torch.zeros(o["sub_count"].numel() + 1, dtype=torch.int64,
+                                  device=device)
+            torch.cumsum(o["sub_count"], 0, out=sub_off[1:])
+        values = (o["i64_vals"] if kind == KIND_INT64 else
+                  o["f32_vals"] if kind == KIND_FLOAT else o["bytes_data"])
+        cols.append(WireColumn(kind, seq, presence, val_base[i], values,
+                               elem_off, list_base[i] if seq else None, sub_off))
+    if int(err.item()) != 0:
+        raise RuntimeError(f"TFRecord decode failed in value extraction "
+                           f"(native error {int(err.item())})")
+    return RecordBatch(schema, cols, R)
+
+
+def decode_buffer_device(data_np: np.ndarray, schema: StructType, record_type: str,
+                         verify_crc: bool = True, device="cuda") -> RecordBatch:
+    """Host bytes -> device batch (header scan on host, all else on GPU)."""
+    off_np, len_np = _native.scan_frame_headers(np.ascontiguousarray(data_np,
+                                                                     np.uint8))
+    data = torch.as_tensor(data_np).to(device, non_blocking=True)
+    off = _dev(off_np, torch.int64, device)
+    lens = _dev(len_np, torch.int64, device)
+    return decode_device(data, off, lens, schema, record_type, verify_crc)
+
+
+# ---------------------------------------------------------------------------
+# Encode
+# ---------------------------------------------------------------------------
+
+def _col_ptrs(col: WireColumn) -> dict:
+    def p(t):
+        return t.data_ptr() if isinstance(t, torch.Tensor) and t.numel() else 0
+
+    return {
+        "kind": col.kind, "is_seq": 1 if col.is_seq else 0,
+        "presence": p(col.presence), "row_off": p(col.row_off),
+        "list_off": p(col.list_off) if col.list_off is not None else 0,
+        "sub_off": p(col.sub_off) if col.sub_off is not None else 0,
+        "elem_off": p(col.elem_off) if col.elem_off is not None else 0,
+        "values_bytes": p(col.values) if col.kind == KIND_BYTES else 0,
+        "values_i64": p(col.values) if col.kind == KIND_INT64 else 0,
+        "values_f32": p(col.values) if col.kind == KIND_FLOAT else 0,
+    }
+
+
+def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
+    """Device wire-form batch -> framed file image as a device u8 tensor."""
+    check_native()
+    R = batch.num_rows
+    device = batch.columns[0].presence.device if batch.columns else torch.device("cuda")
+    if record_type == "ByteArray":
+        col = batch.columns[0]
+        lens = col.elem_off[1:] - col.elem_off[:-1]
+        frame_off = torch.zeros(R + 1, dtype=torch.int64, device=device)
+        torch.cumsum(lens + 16, 0, out=frame_off[1:])
+        total = int(frame_off[-1].item())
+        file = torch.empty(total, dtype=torch.uint8, device=device)
+        if R:
+            _native.gpu_frame_bytes(col.values.data_ptr(), col.elem_off.data_ptr(),
+                                    frame_off.data_ptr(), R, file.data_ptr(),
+                                    _stream())
+        return file
+
+    blob = torch.frombuffer(bytearray(schema_blob(batch.schema)),
+                            dtype=torch.uint8).to(device)
+    col_dicts = [_col_ptrs(c) for c in batch.columns]
+    cols_dev = torch.empty(_native.gpu_devcols_bytes(), dtype=torch.uint8,
+                           device=device)
+    psize = torch.empty(R, dtype=torch.int64, device=device)
+    _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
+                             FMT[record_type], R, psize.data_ptr(), _stream())
+    frame_off = torch.zeros(R + 1, dtype=torch.int64, device=device)
+    torch.cumsum(psize, 0, out=frame_off[1:])
+    total = int(frame_off[-1].item())
+    file = torch.empty(total, dtype=torch.uint8, device=device)
+    err = torch.zeros(1, dtype=torch.int32, device=device)
+    _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
+                             FMT[record_type], R, frame_off.data_ptr(),
+                             file.data_ptr(), err.data_ptr(), _stream())
+    if int(err.item()) != 0:
+        raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+    return file
+
+
+def batch_to_device(batch: RecordBatch, device="cuda") -> RecordBatch:
+    cols = []
+    for c in batch.columns:
+        cols.append(WireColumn(
+            c.kind, c.is_seq,
+            _dev(c.presence, torch.uint8, device),
+            _dev(c.row_off, torch.int64, device),
+            _dev(c.values, {KIND_INT64: torch.int64, KIND_FLOAT: torch.float32,
+                            KIND_BYTES: torch.uint8}[c.kind], device),
+            _dev(c.elem_off, torch.int64, device) if c.elem_off is not None else None,
+            _dev(c.list_off, torch.int64, device) if c.list_off is not None else None,
+            _dev(c.sub_off, torch.int64, device) if c.sub_off is not None else None,
+        ))
+    return RecordBatch(batch.schema, cols, batch.num_rows)
+
+
+def batch_to_host(batch: RecordBatch) -> RecordBatch:
+    def h(t):
+        return t.cpu().numpy() if isinstance(t, torch.Tensor) else t
+
+    cols = [WireColumn(c.kind, c.is_seq, h(c.presence), h(c.row_off), h(c.values),
+                       h(c.elem_off) if c.elem_off is not None else None,
+                       h(c.list_off) if c.list_off is not None else None,
+                       h(c.sub_off) if c.sub_off is not None else None)
+            for c in batch.columns]
+    return RecordBatch(batch.schema, cols, batch.num_rows)
+
+
+# ---------------------------------------------------------------------------
+# CPU-boundary convenience wrappers (used by the file reader/writer)
+# ---------------------------------------------------------------------------
+
+def decode_buffer_to_cpu(data_np: np.ndarray, schema: StructType, record_type: str,
+                         verify_crc: bool = True) -> RecordBatch:
+    return batch_to_host(decode_buffer_device(data_np, schema, record_type,
+                                              verify_crc))
+
+
+def encode_batch_from_cpu(batch: RecordBatch, record_type: str) -> bytes:
+    dev_batch = batch_to_device(batch)
+    file = encode_device(dev_batch, record_type)
+    return file.cpu().numpy().tobytes()

@@ -342,6 +342,11 @@ void gpu_gather_payloads(uintptr_t data, uintptr_t off, uintptr_t len,
 }  // namespace
 
 void register_gpu(py::module_& m) {
+  m.def("gpu_device_count", []() {
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    return (e == hipSuccess) ? n : -static_cast<int>(e);
+  });
   m.def("gpu_crc_verify", &gpu_crc_verify);
   m.def("gpu_scan_records", &gpu_scan_records);
   m.def("gpu_extract_fields", &gpu_extract_fields);

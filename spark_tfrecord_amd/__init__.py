@@ -8,6 +8,11 @@ in C++/HIP for CDNA4 (gfx950) and multi-GPU scaling over RCCL/xGMI instead
 of a JVM + Spark. See SURVEY.md for the structural map of the reference.
 """
 
+# torch must load first so _native.so binds the same (torch-bundled) HIP
+# runtime: loading the system libamdhip64 first leaves torch and _native on
+# two different runtimes, and kernel launches fail with hipErrorNoDevice.
+import torch  # noqa: F401  (intentional import order)
+
 from .api import DataFrame, DataFrameReader, DataFrameWriter, TFRecordSession, session
 from .io.reader import read_tfrecord
 from .io.writer import write_tfrecord

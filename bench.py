@@ -72,6 +72,8 @@ def main():
     ap.add_argument("--dir", type=str, default=None,
                     help="bench storage dir (default: /dev/shm tmpfs)")
     ap.add_argument("--engine", type=str, default="auto")
+    ap.add_argument("--phases", action="store_true",
+                    help="print per-phase timing breakdown (rank 0)")
     args = ap.parse_args()
 
     import torch
@@ -107,15 +109,25 @@ def main():
         from spark_tfrecord_amd.engine import gpu as gpu_engine
 
         dev_batch = gpu_engine.batch_to_device(batch)
+        phase_t = {}
+
+        def timed(name, fn):
+            if not args.phases:
+                return fn()
+            torch.cuda.synchronize()
+            t = time.perf_counter()
+            r = fn()
+            torch.cuda.synchronize()
+            phase_t[name] = phase_t.get(name, 0.0) + time.perf_counter() - t
+            return r
 
         def step():
-            img = gpu_engine.encode_device(dev_batch, "Example")  # GPU encode
-            host = img.cpu().numpy()                              # D2H
-            with open(fpath, "wb") as f:                          # storage write
-                host.tofile(f)
-            data = np.fromfile(fpath, np.uint8)                   # storage read
-            out = gpu_engine.decode_buffer_device(data, batch.schema, "Example",
-                                                  verify_crc=True)
+            img = timed("encode", lambda: gpu_engine.encode_device(dev_batch, "Example"))
+            host = timed("d2h", lambda: img.cpu().numpy())
+            timed("fwrite", lambda: host.tofile(fpath))
+            data = timed("fread", lambda: np.fromfile(fpath, np.uint8))
+            out = timed("decode", lambda: gpu_engine.decode_buffer_device(
+                data, batch.schema, "Example", verify_crc=True))
             return out
         engine_name = "gpu"
     else:
@@ -163,6 +175,13 @@ def main():
     total_rows = rows * args.steps * world_size
     rows_per_sec = total_rows / elapsed
     mb_per_sec = file_bytes * args.steps * world_size / elapsed / 1e6
+
+    if rank == 0 and args.phases and engine_name == "gpu":
+        total = sum(phase_t.values())
+        breakdown = {k: f"{v / (args.steps + args.warmup + 1) * 1000:.1f}ms"
+                     for k, v in phase_t.items()}
+        print(f"# phase breakdown (per step, incl. warmup): {breakdown}",
+              file=sys.stderr)
 
     if rank == 0:
         print(json.dumps({

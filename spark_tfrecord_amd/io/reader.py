@@ -76,7 +76,7 @@ def _partition_col_array(values: List[str], n: int):
 
 def read_tfrecord(path: str, schema: Optional[StructType] = None,
                   record_type: str = "Example", engine: str = "auto",
-                  verify_crc: bool = True):
+                  verify_crc: bool = True, base_dir: Optional[str] = None):
     from ..api import DataFrame
 
     if record_type not in ("Example", "SequenceExample", "ByteArray"):
@@ -84,7 +84,8 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     files = P.list_data_files(path)
     if not files:
         raise FileNotFoundError(f"No TFRecord files found under {path}")
-    base_dir = path if os.path.isdir(path) else os.path.dirname(path)
+    if base_dir is None:
+        base_dir = path if os.path.isdir(path) else os.path.dirname(path)
     part_cols = _partition_schema(files, base_dir) if base_dir else []
     eng = engine_mod.resolve_engine(engine)
 

@@ -123,11 +123,11 @@ def main():
 
         def step():
             img = timed("encode", lambda: gpu_engine.encode_device(dev_batch, "Example"))
-            host = timed("d2h", lambda: img.cpu().numpy())
-            timed("fwrite", lambda: host.tofile(fpath))
-            data = timed("fread", lambda: np.fromfile(fpath, np.uint8))
-            out = timed("decode", lambda: gpu_engine.decode_buffer_device(
-                data, batch.schema, "Example", verify_crc=True))
+            timed("write", lambda: gpu_engine.device_to_file(img, fpath))
+            data = timed("read", lambda: gpu_engine.read_file_to_device(fpath))
+            ol = timed("fscan", lambda: gpu_engine.scan_frames_device(data))
+            out = timed("decode", lambda: gpu_engine.decode_device(
+                data, ol[0], ol[1], batch.schema, "Example", verify_crc=True))
             return out
         engine_name = "gpu"
     else:

@@ -202,6 +202,36 @@ __global__ void frame_bytes_kernel(const u8* __restrict__ src,
   }
 }
 
+// Parallel frame-boundary discovery (SURVEY.md §7 "hard parts"): TFRecord has
+// no sync markers, so every byte position is tested as a candidate frame head
+// (8-byte length whose masked CRC32C matches the next 4 bytes). A real head's
+// 12-byte check passes always; a random position passes with p = 2^-32, so
+// candidates ~= frames. Python sorts the candidates and validates the chain
+// pos[k+1] == pos[k] + 16 + len[k] with two tensor ops — no host pass over
+// the file bytes at all.
+__global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
+                                       i64* __restrict__ cand_pos,
+                                       i64* __restrict__ cand_len,
+                                       unsigned long long* __restrict__ count,
+                                       i64 max_cand) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i + 12 <= N;
+       i += (i64)gridDim.x * blockDim.x) {
+    u64 len;
+    __builtin_memcpy(&len, data + i, 8);
+    if ((u64)(N - i) < 16 || len > (u64)(N - i) - 16) continue;
+    u32 want;
+    __builtin_memcpy(&want, data + i + 8, 4);
+    if (mask_crc(crc32c_sw(data + i, 8, 0, tab)) != want) continue;
+    unsigned long long k = atomicAdd(count, 1ull);
+    if ((i64)k < max_cand) {
+      cand_pos[k] = i;
+      cand_len[k] = (i64)len;
+    }
+  }
+}
+
 // Raw-payload gather for the ByteArray read path (payload extents -> packed).
 __global__ void gather_payloads_kernel(const u8* __restrict__ data,
                                        const i64* __restrict__ off,
@@ -331,6 +361,14 @@ void gpu_frame_bytes(uintptr_t src, uintptr_t elem_off, uintptr_t frame_off, i64
   HIP_CHECK(hipGetLastError());
 }
 
+void gpu_frame_scan(uintptr_t data, i64 N, uintptr_t cand_pos, uintptr_t cand_len,
+                    uintptr_t count, i64 max_cand, uintptr_t stream) {
+  hipLaunchKernelGGL(frame_candidate_kernel, dim3(grid_for(N / 64)), dim3(kBlock),
+                     0, (hipStream_t)stream, (const u8*)data, N, (i64*)cand_pos,
+                     (i64*)cand_len, (unsigned long long*)count, max_cand);
+  HIP_CHECK(hipGetLastError());
+}
+
 void gpu_gather_payloads(uintptr_t data, uintptr_t off, uintptr_t len,
                          uintptr_t dst_off, i64 R, uintptr_t out, uintptr_t stream) {
   hipLaunchKernelGGL(gather_payloads_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
@@ -353,6 +391,7 @@ void register_gpu(py::module_& m) {
   m.def("gpu_size_records", &gpu_size_records);
   m.def("gpu_emit_records", &gpu_emit_records);
   m.def("gpu_frame_bytes", &gpu_frame_bytes);
+  m.def("gpu_frame_scan", &gpu_frame_scan);
   m.def("gpu_gather_payloads", &gpu_gather_payloads);
   m.def("gpu_devcols_bytes", &gpu_devcols_bytes);
   m.def("gpu_devmeta_bytes", []() { return sizeof(DevFieldDst); });

@@ -43,6 +43,116 @@ def check_native():
 
 
 # ---------------------------------------------------------------------------
+# Pinned staging buffers: host<->device copies through pageable memory run at
+# ~8 GB/s on this platform vs ~57 GB/s pinned (measured, exp/exp_probe.py).
+# One reusable pinned buffer per purpose, grown geometrically.
+# ---------------------------------------------------------------------------
+
+_pinned_pool = {}
+
+
+def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
+    buf = _pinned_pool.get(tag)
+    if buf is None or buf.numel() < nbytes:
+        cap = max(nbytes, int((buf.numel() if buf is not None else 1 << 20) * 1.5))
+        buf = torch.empty(cap, dtype=torch.uint8, pin_memory=True)
+        _pinned_pool[tag] = buf
+    return buf
+
+
+def read_file_to_device(path: str, device="cuda") -> torch.Tensor:
+    """File -> pinned staging -> HBM (one pinned H2D copy, no numpy detour)."""
+    import os as _os
+
+    n = _os.path.getsize(path)
+    buf = pinned_buffer("fread", n)
+    view = buf.numpy()[:n]
+    with open(path, "rb") as f:
+        got = f.readinto(memoryview(view))
+    if got != n:
+        raise IOError(f"short read from {path}")
+    dev = torch.empty(n, dtype=torch.uint8, device=device)
+    dev.copy_(buf[:n], non_blocking=True)
+    return dev
+
+
+def device_to_file(img: torch.Tensor, path: str):
+    """HBM file image -> pinned staging -> file (overwrite in place)."""
+    n = img.numel()
+    buf = pinned_buffer("fwrite", n)
+    buf[:n].copy_(img, non_blocking=True)
+    torch.cuda.synchronize()
+    with open(path, "wb") as f:
+        f.write(memoryview(buf.numpy()[:n]))
+
+
+def device_to_bytes(img: torch.Tensor) -> bytes:
+    n = img.numel()
+    buf = pinned_buffer("d2h", n)
+    buf[:n].copy_(img, non_blocking=True)
+    torch.cuda.synchronize()
+    return buf.numpy()[:n].tobytes()
+
+
+# ---------------------------------------------------------------------------
+# GPU frame scan: parallel frame-boundary discovery + chain validation.
+# ---------------------------------------------------------------------------
+
+def scan_frames_device(data: torch.Tensor):
+    """Device file image -> (payload_off, payload_len) device tensors.
+
+    Every byte position is CRC-tested as a candidate frame head in parallel
+    (frame_candidate_kernel); the sorted candidates must chain exactly
+    (pos[k+1] == pos[k] + 16 + len[k]) — a false positive (p=2^-32/byte) or a
+    torn file breaks the chain and raises."""
+    check_native()
+    N = data.numel()
+    device = data.device
+    if N == 0:
+        z = torch.zeros(0, dtype=torch.int64, device=device)
+        return z, z.clone()
+    max_cand = N // 16 + 64
+    cand_pos = torch.empty(max_cand, dtype=torch.int64, device=device)
+    cand_len = torch.empty(max_cand, dtype=torch.int64, device=device)
+    count = torch.zeros(1, dtype=torch.int64, device=device)
+    _native.gpu_frame_scan(data.data_ptr(), N, cand_pos.data_ptr(),
+                           cand_len.data_ptr(), count.data_ptr(), max_cand,
+                           _stream())
+    C = int(count.item())
+    if C == 0:
+        raise RuntimeError("corrupt TFRecord: no valid frame header found")
+    if C > max_cand:
+        raise RuntimeError("corrupt TFRecord: implausible candidate count")
+    pos, order = torch.sort(cand_pos[:C])
+    lens = cand_len[:C][order]
+    # chain check in two tensor ops (no host pass over the data)
+    expect_next = pos + 16 + lens
+    ok = bool((pos[0] == 0).item()) and bool((expect_next[-1] == N).item()) \
+        and bool(torch.equal(expect_next[:-1], pos[1:]))
+    if not ok:
+        # rare: false-positive candidate inside a payload — stitch on host
+        pos_h = pos.cpu().numpy()
+        len_h = lens.cpu().numpy()
+        import numpy as _np
+
+        keep = []
+        cur = 0
+        idx = 0
+        pmap = {int(p): i for i, p in enumerate(pos_h)}
+        while cur < N:
+            i = pmap.get(cur)
+            if i is None:
+                raise RuntimeError(
+                    f"corrupt TFRecord: broken frame chain at offset {cur}")
+            keep.append(i)
+            cur = int(pos_h[i] + 16 + len_h[i])
+        sel = torch.as_tensor(_np.asarray(keep, _np.int64), device=device)
+        pos = pos[sel]
+        lens = lens[sel]
+    return pos + 12, lens
+
+
+# ---------------------------------------------------------------------------
 # Decode
 # ---------------------------------------------------------------------------
 
@@ -104,10 +214,14 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                              FMT[record_type], blob.data_ptr(), F,
                              stats.data_ptr(), err.data_ptr(), _stream())
 
-    # Per-field exclusive prefix sums ([F, R+1], row-contiguous views)
+    # Per-field exclusive prefix sums ([F, R+1], row-contiguous views).
+    # The transpose is materialized first: cumsum over a strided innermost
+    # dim is ~10x slower than over a contiguous one (rocprof: 2.5 ms vs copy
+    # + contiguous scan well under 1 ms at R=1M).
     def excl_scan(counts_rf: torch.Tensor) -> torch.Tensor:
         base = torch.zeros((F, R + 1), dtype=torch.int64, device=device)
-        torch.cumsum(counts_rf.transpose(0, 1), dim=1, out=base[:, 1:])
+        torch.cumsum(counts_rf.transpose(0, 1).contiguous(), dim=1,
+                     out=base[:, 1:])
         return base
 
     val_base = excl_scan(stats[:, :, 2])
@@ -181,12 +295,22 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
 
 def decode_buffer_device(data_np: np.ndarray, schema: StructType, record_type: str,
                          verify_crc: bool = True, device="cuda") -> RecordBatch:
-    """Host bytes -> device batch (header scan on host, all else on GPU)."""
-    off_np, len_np = _native.scan_frame_headers(np.ascontiguousarray(data_np,
-                                                                     np.uint8))
-    data = torch.as_tensor(data_np).to(device, non_blocking=True)
-    off = _dev(off_np, torch.int64, device)
-    lens = _dev(len_np, torch.int64, device)
+    """Host bytes -> device batch (frame discovery AND decode on the GPU)."""
+    data_np = np.ascontiguousarray(data_np, np.uint8)
+    n = data_np.size
+    stage = pinned_buffer("h2d", n)
+    stage.numpy()[:n] = data_np
+    data = torch.empty(n, dtype=torch.uint8, device=device)
+    data.copy_(stage[:n], non_blocking=True)
+    off, lens = scan_frames_device(data)
+    return decode_device(data, off, lens, schema, record_type, verify_crc)
+
+
+def read_file_to_batch(path: str, schema: StructType, record_type: str,
+                       verify_crc: bool = True, device="cuda") -> RecordBatch:
+    """Uncompressed file -> pinned read -> GPU frame scan -> device batch."""
+    data = read_file_to_device(path, device)
+    off, lens = scan_frames_device(data)
     return decode_device(data, off, lens, schema, record_type, verify_crc)
 
 
@@ -290,4 +414,4 @@ def decode_buffer_to_cpu(data_np: np.ndarray, schema: StructType, record_type: s
 def encode_batch_from_cpu(batch: RecordBatch, record_type: str) -> bytes:
     dev_batch = batch_to_device(batch)
     file = encode_device(dev_batch, record_type)
-    return file.cpu().numpy().tobytes()
+    return device_to_bytes(file)

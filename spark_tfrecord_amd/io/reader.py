@@ -96,14 +96,23 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
 
     tables = []
     for fpath in files:
-        data = _load_file(fpath)
-        if data.size == 0:
-            continue
         if eng == "gpu":
             from ..engine import gpu as gpu_engine
-            batch = gpu_engine.decode_buffer_to_cpu(data, data_schema, record_type,
-                                                    verify_crc=verify_crc)
+            if P.codec_from_path(fpath) is None:
+                if os.path.getsize(fpath) == 0:
+                    continue
+                batch = gpu_engine.batch_to_host(gpu_engine.read_file_to_batch(
+                    fpath, data_schema, record_type, verify_crc=verify_crc))
+            else:
+                data = _load_file(fpath)
+                if data.size == 0:
+                    continue
+                batch = gpu_engine.decode_buffer_to_cpu(
+                    data, data_schema, record_type, verify_crc=verify_crc)
         else:
+            data = _load_file(fpath)
+            if data.size == 0:
+                continue
             batch = cpu_engine.decode_buffer(data, data_schema, record_type,
                                              verify_crc=verify_crc)
         t = batch_to_table(batch)

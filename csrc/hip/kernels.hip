@@ -209,6 +209,23 @@ __global__ void frame_bytes_kernel(const u8* __restrict__ src,
 // candidates ~= frames. Python sorts the candidates and validates the chain
 // pos[k+1] == pos[k] + 16 + len[k] with two tensor ops — no host pass over
 // the file bytes at all.
+// CRC32C of 8 little-endian bytes held in a register (one slicing-by-8 step).
+__device__ inline u32 crc8b_reg(u64 w, const uint32_t (*tab)[256]) {
+  w ^= 0xFFFFFFFFull;  // crc init (~0) folded into low 32 bits
+  u32 c = tab[7][w & 0xFF] ^ tab[6][(w >> 8) & 0xFF] ^ tab[5][(w >> 16) & 0xFF] ^
+          tab[4][(w >> 24) & 0xFF] ^ tab[3][(w >> 32) & 0xFF] ^
+          tab[2][(w >> 40) & 0xFF] ^ tab[1][(w >> 48) & 0xFF] ^
+          tab[0][(w >> 56) & 0xFF];
+  return ~c;
+}
+
+// Each lane owns kPosPerLane consecutive byte positions; the covering bytes
+// are loaded once as aligned u64 words and every position's (length, crc)
+// pair is reassembled with 128-bit funnel shifts — no per-byte loads. The
+// length bound check (random u64 <= N with p ~ N/2^64) culls everything but
+// real frame heads before any CRC work.
+constexpr int kPosPerLane = 16;
+
 __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
                                        i64* __restrict__ cand_pos,
                                        i64* __restrict__ cand_len,
@@ -216,18 +233,42 @@ __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
                                        i64 max_cand) {
   __shared__ uint32_t tab[8][256];
   stage_crc_tables(tab);
-  for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i + 12 <= N;
-       i += (i64)gridDim.x * blockDim.x) {
-    u64 len;
-    __builtin_memcpy(&len, data + i, 8);
-    if ((u64)(N - i) < 16 || len > (u64)(N - i) - 16) continue;
-    u32 want;
-    __builtin_memcpy(&want, data + i + 8, 4);
-    if (mask_crc(crc32c_sw(data + i, 8, 0, tab)) != want) continue;
-    unsigned long long k = atomicAdd(count, 1ull);
-    if ((i64)k < max_cand) {
-      cand_pos[k] = i;
-      cand_len[k] = (i64)len;
+  const u64* wdata = reinterpret_cast<const u64*>(data);  // data is 8B-aligned
+  i64 nchunks = (N + kPosPerLane - 1) / kPosPerLane;
+  for (i64 c = blockIdx.x * (i64)blockDim.x + threadIdx.x; c < nchunks;
+       c += (i64)gridDim.x * blockDim.x) {
+    i64 p0 = c * kPosPerLane;
+    // window bytes [p0, p0 + kPosPerLane-1 + 12] -> aligned words
+    i64 w0 = p0 >> 3;
+    u64 w[5];
+    i64 wmax = (N + 7) >> 3;
+#pragma unroll
+    for (int j = 0; j < 5; ++j)
+      w[j] = (w0 + j < wmax) ? wdata[w0 + j] : 0;
+    int base_sh = (int)(p0 & 7);
+#pragma unroll
+    for (int k = 0; k < kPosPerLane; ++k) {
+      i64 i = p0 + k;
+      if (i + 16 > N) break;
+      int d = base_sh + k;          // byte offset within w[]
+      int wi = d >> 3;
+      int sh = (d & 7) * 8;
+      u64 len = (sh == 0) ? w[wi]
+                          : (w[wi] >> sh) | (w[wi + 1] << (64 - sh));
+      if (len > (u64)(N - i) - 16) continue;
+      int d2 = d + 8;
+      int wi2 = d2 >> 3;
+      int sh2 = (d2 & 7) * 8;
+      u64 hi = (sh2 == 0) ? w[wi2]
+                          : (w[wi2] >> sh2) | ((wi2 + 1 < 5 ? w[wi2 + 1] : 0)
+                                              << (64 - sh2));
+      u32 want = (u32)hi;
+      if (mask_crc(crc8b_reg(len, tab)) != want) continue;
+      unsigned long long slot = atomicAdd(count, 1ull);
+      if ((i64)slot < max_cand) {
+        cand_pos[slot] = i;
+        cand_len[slot] = (i64)len;
+      }
     }
   }
 }
@@ -363,8 +404,9 @@ void gpu_frame_bytes(uintptr_t src, uintptr_t elem_off, uintptr_t frame_off, i64
 
 void gpu_frame_scan(uintptr_t data, i64 N, uintptr_t cand_pos, uintptr_t cand_len,
                     uintptr_t count, i64 max_cand, uintptr_t stream) {
-  hipLaunchKernelGGL(frame_candidate_kernel, dim3(grid_for(N / 64)), dim3(kBlock),
-                     0, (hipStream_t)stream, (const u8*)data, N, (i64*)cand_pos,
+  hipLaunchKernelGGL(frame_candidate_kernel,
+                     dim3(grid_for(N / kPosPerLane + 1)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)data, N, (i64*)cand_pos,
                      (i64*)cand_len, (unsigned long long*)count, max_cand);
   HIP_CHECK(hipGetLastError());
 }

@@ -60,30 +60,70 @@ def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
     return buf
 
 
+_CHUNK = 32 << 20  # 32 MiB: overlap file IO with PCIe copies chunkwise
+
+
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:
-    """File -> pinned staging -> HBM (one pinned H2D copy, no numpy detour)."""
+    """File -> pinned staging -> HBM, chunked so the H2D copy of chunk k
+    overlaps the file read of chunk k+1."""
     import os as _os
 
     n = _os.path.getsize(path)
-    buf = pinned_buffer("fread", n)
-    view = buf.numpy()[:n]
+    dev = torch.empty(max(n, 1), dtype=torch.uint8, device=device)[:n]
+    buf = pinned_buffer("fread", min(n, 2 * _CHUNK) or 1)
+    ev = [torch.cuda.Event(), torch.cuda.Event()]
     with open(path, "rb") as f:
-        got = f.readinto(memoryview(view))
-    if got != n:
-        raise IOError(f"short read from {path}")
-    dev = torch.empty(n, dtype=torch.uint8, device=device)
-    dev.copy_(buf[:n], non_blocking=True)
+        pos = 0
+        which = 0
+        while pos < n:
+            m = min(_CHUNK, n - pos)
+            half = buf[which * _CHUNK:which * _CHUNK + m]
+            ev[which].synchronize()  # prior H2D from this half must be done
+            got = f.readinto(memoryview(half.numpy()))
+            if got != m:
+                raise IOError(f"short read from {path}")
+            dev[pos:pos + m].copy_(half, non_blocking=True)
+            ev[which].record()
+            pos += m
+            which ^= 1 if n > _CHUNK else 0
     return dev
 
 
 def device_to_file(img: torch.Tensor, path: str):
-    """HBM file image -> pinned staging -> file (overwrite in place)."""
+    """HBM file image -> pinned staging -> file.
+
+    Writes in place (r+b) when the file exists so tmpfs/page-cache pages are
+    reused instead of freed+reallocated by O_TRUNC; truncates to the new
+    size afterwards. Chunked: file write of chunk k overlaps the D2H copy of
+    chunk k+1."""
+    import os as _os
+
     n = img.numel()
-    buf = pinned_buffer("fwrite", n)
-    buf[:n].copy_(img, non_blocking=True)
-    torch.cuda.synchronize()
-    with open(path, "wb") as f:
-        f.write(memoryview(buf.numpy()[:n]))
+    buf = pinned_buffer("fwrite", min(n, 2 * _CHUNK) or 1)
+    mode = "r+b" if _os.path.exists(path) else "wb"
+    ev = [torch.cuda.Event(), torch.cuda.Event()]
+    with open(path, mode) as f:
+        pos = 0
+        which = 0
+        # prefetch chunk 0
+        m0 = min(_CHUNK, n)
+        if n:
+            buf[0:m0].copy_(img[0:m0], non_blocking=True)
+            ev[0].record()
+        while pos < n:
+            m = min(_CHUNK, n - pos)
+            nxt = pos + m
+            nwhich = which ^ (1 if n > _CHUNK else 0)
+            if nxt < n:  # start next D2H before blocking on this chunk
+                m2 = min(_CHUNK, n - nxt)
+                buf[nwhich * _CHUNK:nwhich * _CHUNK + m2].copy_(
+                    img[nxt:nxt + m2], non_blocking=True)
+                ev[nwhich].record()
+            ev[which].synchronize()
+            f.write(memoryview(buf[which * _CHUNK:which * _CHUNK + m].numpy()))
+            pos = nxt
+            which = nwhich
+        f.truncate(n)
 
 
 def device_to_bytes(img: torch.Tensor) -> bytes:

@@ -235,33 +235,42 @@ __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
   stage_crc_tables(tab);
   const u64* wdata = reinterpret_cast<const u64*>(data);  // data is 8B-aligned
   i64 nchunks = (N + kPosPerLane - 1) / kPosPerLane;
+  i64 wmax = (N + 7) >> 3;
   for (i64 c = blockIdx.x * (i64)blockDim.x + threadIdx.x; c < nchunks;
        c += (i64)gridDim.x * blockDim.x) {
+    // p0 = 16*c is 8-byte aligned, so every index below is compile-time
+    // constant after unrolling (a runtime-indexed w[] would spill to scratch).
     i64 p0 = c * kPosPerLane;
-    // window bytes [p0, p0 + kPosPerLane-1 + 12] -> aligned words
     i64 w0 = p0 >> 3;
-    u64 w[5];
-    i64 wmax = (N + 7) >> 3;
+    u64 w[4];
+    if (p0 + 32 <= N) {  // interior chunk: full-word loads
 #pragma unroll
-    for (int j = 0; j < 5; ++j)
-      w[j] = (w0 + j < wmax) ? wdata[w0 + j] : 0;
-    int base_sh = (int)(p0 & 7);
+      for (int j = 0; j < 4; ++j) w[j] = wdata[w0 + j];
+    } else {  // file tail: assemble partial words byte-wise, zero-padded
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        u64 v = 0;
+        i64 boff = (w0 + j) * 8;
+        for (int b = 0; b < 8 && boff + b < N; ++b)
+          v |= (u64)data[boff + b] << (8 * b);
+        w[j] = v;
+      }
+    }
+    (void)wmax;
 #pragma unroll
     for (int k = 0; k < kPosPerLane; ++k) {
       i64 i = p0 + k;
       if (i + 16 > N) break;
-      int d = base_sh + k;          // byte offset within w[]
-      int wi = d >> 3;
-      int sh = (d & 7) * 8;
-      u64 len = (sh == 0) ? w[wi]
-                          : (w[wi] >> sh) | (w[wi + 1] << (64 - sh));
+      constexpr int _ppl = kPosPerLane;
+      static_assert(_ppl == 16, "index math below assumes 16 positions");
+      const int wi = k >> 3;
+      const int sh = (k & 7) * 8;
+      u64 len = (sh == 0) ? w[wi] : (w[wi] >> sh) | (w[wi + 1] << (64 - sh));
       if (len > (u64)(N - i) - 16) continue;
-      int d2 = d + 8;
-      int wi2 = d2 >> 3;
-      int sh2 = (d2 & 7) * 8;
+      const int wi2 = (k + 8) >> 3;
+      const int sh2 = sh;  // (k+8) & 7 == k & 7
       u64 hi = (sh2 == 0) ? w[wi2]
-                          : (w[wi2] >> sh2) | ((wi2 + 1 < 5 ? w[wi2 + 1] : 0)
-                                              << (64 - sh2));
+                          : (w[wi2] >> sh2) | (w[wi2 + 1] << (64 - sh2));
       u32 want = (u32)hi;
       if (mask_crc(crc8b_reg(len, tab)) != want) continue;
       unsigned long long slot = atomicAdd(count, 1ull);

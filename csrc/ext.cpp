@@ -13,10 +13,18 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <fcntl.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <condition_variable>
 #include <cstring>
+#include <functional>
 #include <map>
+#include <mutex>
 #include <stdexcept>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "codec_core.h"
@@ -25,6 +33,133 @@ namespace py = pybind11;
 using namespace tfrec;
 
 namespace {
+
+// ---------------------------------------------------------------------------
+// Parallel file IO: persistent pread/pwrite worker pool. A single-threaded
+// write() to tmpfs tops out at ~9 GB/s (one kernel memcpy); splitting each
+// chunk across workers scales that with cores. Used by the GPU engine to keep
+// file IO off the critical path of the H2D/D2H pipeline
+// (spark_tfrecord_amd/engine/gpu.py). Replaces what the reference delegates
+// to Hadoop FS streams (DefaultSource.scala:118-136).
+// ---------------------------------------------------------------------------
+
+class IOPool {
+ public:
+  static IOPool& instance() {
+    // leaked on purpose: workers block on the cv forever; destructing the
+    // mutex/cv under them at exit would be UB
+    static IOPool* pool = new IOPool(8);
+    return *pool;
+  }
+
+  // op: false=read, true=write. Splits [off, off+n) across the workers and
+  // blocks until all spans complete. Returns 0 or -errno.
+  int run(bool is_write, int fd, u8* ptr, i64 n, i64 off) {
+    if (n <= 0) return 0;
+    int nspan = (int)nthreads_;
+    i64 span = (n + nspan - 1) / nspan;
+    if (span < (64 << 10)) {  // don't split tiny transfers
+      nspan = 1;
+      span = n;
+    }
+    std::atomic<int> err{0};
+    std::atomic<int> left{nspan};
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      for (int t = 0; t < nspan; ++t) {
+        i64 o = (i64)t * span;
+        i64 m = std::min(span, n - o);
+        tasks_.push_back([=, &err, &left] {
+          int e = is_write ? do_write(fd, ptr + o, m, off + o)
+                           : do_read(fd, ptr + o, m, off + o);
+          if (e) err.store(e, std::memory_order_relaxed);
+          if (left.fetch_sub(1) == 1) {
+            std::lock_guard<std::mutex> dl(done_mu_);
+            done_cv_.notify_all();
+          }
+        });
+      }
+      cv_.notify_all();
+    }
+    std::unique_lock<std::mutex> dl(done_mu_);
+    done_cv_.wait(dl, [&] { return left.load() == 0; });
+    return err.load();
+  }
+
+ private:
+  explicit IOPool(int n) : nthreads_(n) {
+    for (int i = 0; i < n; ++i)
+      workers_.emplace_back([this] { worker(); });
+  }
+
+  static int do_write(int fd, const u8* p, i64 n, i64 off) {
+    while (n > 0) {
+      ssize_t w = ::pwrite(fd, p, (size_t)n, (off_t)off);
+      if (w < 0) {
+        if (errno == EINTR) continue;
+        return -errno;
+      }
+      p += w;
+      n -= w;
+      off += w;
+    }
+    return 0;
+  }
+
+  static int do_read(int fd, u8* p, i64 n, i64 off) {
+    while (n > 0) {
+      ssize_t r = ::pread(fd, p, (size_t)n, (off_t)off);
+      if (r < 0) {
+        if (errno == EINTR) continue;
+        return -errno;
+      }
+      if (r == 0) return -EIO;  // unexpected EOF
+      p += r;
+      n -= r;
+      off += r;
+    }
+    return 0;
+  }
+
+  void worker() {
+    for (;;) {
+      std::function<void()> task;
+      {
+        std::unique_lock<std::mutex> lk(mu_);
+        cv_.wait(lk, [this] { return !tasks_.empty(); });
+        task = std::move(tasks_.back());
+        tasks_.pop_back();
+      }
+      task();
+    }
+  }
+
+  size_t nthreads_;
+  std::vector<std::thread> workers_;
+  std::vector<std::function<void()>> tasks_;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::mutex done_mu_;
+  std::condition_variable done_cv_;
+};
+
+void pwrite_parallel(int fd, uintptr_t ptr, i64 n, i64 file_off) {
+  int e;
+  {
+    py::gil_scoped_release rel;
+    e = IOPool::instance().run(true, fd, (u8*)ptr, n, file_off);
+  }
+  if (e) throw std::runtime_error("pwrite failed: " + std::string(strerror(-e)));
+}
+
+void pread_parallel(int fd, uintptr_t ptr, i64 n, i64 file_off) {
+  int e;
+  {
+    py::gil_scoped_release rel;
+    e = IOPool::instance().run(false, fd, (u8*)ptr, n, file_off);
+  }
+  if (e) throw std::runtime_error("pread failed: " + std::string(strerror(-e)));
+}
 
 struct BufView {
   const u8* data;
@@ -487,6 +622,12 @@ void register_gpu(py::module_& m);  // defined in csrc/hip/kernels.hip
 
 PYBIND11_MODULE(_native, m) {
   m.doc() = "MI355X-native TFRecord codec (host + gfx950 kernels)";
+  m.def("pwrite_parallel", &pwrite_parallel, py::arg("fd"), py::arg("ptr"),
+        py::arg("n"), py::arg("file_off"),
+        "Multi-threaded pwrite of [ptr, ptr+n) to fd at file_off");
+  m.def("pread_parallel", &pread_parallel, py::arg("fd"), py::arg("ptr"),
+        py::arg("n"), py::arg("file_off"),
+        "Multi-threaded pread of n bytes at file_off into ptr");
   m.def("crc32c", &crc32c_py, "CRC32C (Castagnoli) of a byte buffer");
   m.def("masked_crc32c", &masked_crc32c_py, "TFRecord-masked CRC32C");
   m.def("scan_frames", &scan_frames, py::arg("data"), py::arg("verify_crc") = true,

@@ -18,6 +18,7 @@
 #include <hip/hip_runtime.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
+#include <rocprim/rocprim.hpp>
 
 #include <cstdint>
 #include <stdexcept>
@@ -224,7 +225,14 @@ __device__ inline u32 crc8b_reg(u64 w, const uint32_t (*tab)[256]) {
 // pair is reassembled with 128-bit funnel shifts — no per-byte loads. The
 // length bound check (random u64 <= N with p ~ N/2^64) culls everything but
 // real frame heads before any CRC work.
+//
+// Candidates are compacted through an LDS staging buffer with ONE global
+// atomicAdd per block iteration: naive per-candidate atomics on a single
+// global counter serialize at the owning L2 bank (~1M contended atomics cost
+// ~11 ms, measured — profiles/r01_kernel_stats.txt); the aggregated scheme
+// reduces that to ~8k block-level atomics.
 constexpr int kPosPerLane = 16;
+constexpr int kLdsCand = 2048;  // 32 KiB LDS; realistic max ≈ blockDim*16/24
 
 __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
                                        i64* __restrict__ cand_pos,
@@ -232,53 +240,82 @@ __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
                                        unsigned long long* __restrict__ count,
                                        i64 max_cand) {
   __shared__ uint32_t tab[8][256];
+  __shared__ i64 lpos[kLdsCand];
+  __shared__ i64 llen[kLdsCand];
+  __shared__ unsigned int lcount;
+  __shared__ unsigned long long gbase;
   stage_crc_tables(tab);
+  if (threadIdx.x == 0) lcount = 0;
+  __syncthreads();
   const u64* wdata = reinterpret_cast<const u64*>(data);  // data is 8B-aligned
   i64 nchunks = (N + kPosPerLane - 1) / kPosPerLane;
-  i64 wmax = (N + 7) >> 3;
-  for (i64 c = blockIdx.x * (i64)blockDim.x + threadIdx.x; c < nchunks;
-       c += (i64)gridDim.x * blockDim.x) {
-    // p0 = 16*c is 8-byte aligned, so every index below is compile-time
-    // constant after unrolling (a runtime-indexed w[] would spill to scratch).
-    i64 p0 = c * kPosPerLane;
-    i64 w0 = p0 >> 3;
-    u64 w[4];
-    if (p0 + 32 <= N) {  // interior chunk: full-word loads
+  // Uniform per-block trip count so every thread reaches the __syncthreads
+  // flush barrier even when its own chunk index runs past nchunks.
+  for (i64 cbase = (i64)blockIdx.x * blockDim.x; cbase < nchunks;
+       cbase += (i64)gridDim.x * blockDim.x) {
+    i64 c = cbase + threadIdx.x;
+    if (c < nchunks) {
+      // p0 = 16*c is 8-byte aligned, so every index below is compile-time
+      // constant after unrolling (runtime-indexed w[] would spill to scratch).
+      i64 p0 = c * kPosPerLane;
+      i64 w0 = p0 >> 3;
+      u64 w[4];
+      if (p0 + 32 <= N) {  // interior chunk: full-word loads
 #pragma unroll
-      for (int j = 0; j < 4; ++j) w[j] = wdata[w0 + j];
-    } else {  // file tail: assemble partial words byte-wise, zero-padded
+        for (int j = 0; j < 4; ++j) w[j] = wdata[w0 + j];
+      } else {  // file tail: assemble partial words byte-wise, zero-padded
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        u64 v = 0;
-        i64 boff = (w0 + j) * 8;
-        for (int b = 0; b < 8 && boff + b < N; ++b)
-          v |= (u64)data[boff + b] << (8 * b);
-        w[j] = v;
+        for (int j = 0; j < 4; ++j) {
+          u64 v = 0;
+          i64 boff = (w0 + j) * 8;
+          for (int b = 0; b < 8 && boff + b < N; ++b)
+            v |= (u64)data[boff + b] << (8 * b);
+          w[j] = v;
+        }
+      }
+#pragma unroll
+      for (int k = 0; k < kPosPerLane; ++k) {
+        i64 i = p0 + k;
+        if (i + 16 > N) break;
+        constexpr int _ppl = kPosPerLane;
+        static_assert(_ppl == 16, "index math below assumes 16 positions");
+        const int wi = k >> 3;
+        const int sh = (k & 7) * 8;
+        u64 len = (sh == 0) ? w[wi] : (w[wi] >> sh) | (w[wi + 1] << (64 - sh));
+        if (len > (u64)(N - i) - 16) continue;
+        const int wi2 = (k + 8) >> 3;
+        const int sh2 = sh;  // (k+8) & 7 == k & 7
+        u64 hi = (sh2 == 0) ? w[wi2]
+                            : (w[wi2] >> sh2) | (w[wi2 + 1] << (64 - sh2));
+        u32 want = (u32)hi;
+        if (mask_crc(crc8b_reg(len, tab)) != want) continue;
+        unsigned int s = atomicAdd(&lcount, 1u);
+        if (s < kLdsCand) {
+          lpos[s] = i;
+          llen[s] = (i64)len;
+        } else {  // adversarial overflow: spill straight to global (slow, rare)
+          unsigned long long slot = atomicAdd(count, 1ull);
+          if ((i64)slot < max_cand) {
+            cand_pos[slot] = i;
+            cand_len[slot] = (i64)len;
+          }
+        }
       }
     }
-    (void)wmax;
-#pragma unroll
-    for (int k = 0; k < kPosPerLane; ++k) {
-      i64 i = p0 + k;
-      if (i + 16 > N) break;
-      constexpr int _ppl = kPosPerLane;
-      static_assert(_ppl == 16, "index math below assumes 16 positions");
-      const int wi = k >> 3;
-      const int sh = (k & 7) * 8;
-      u64 len = (sh == 0) ? w[wi] : (w[wi] >> sh) | (w[wi + 1] << (64 - sh));
-      if (len > (u64)(N - i) - 16) continue;
-      const int wi2 = (k + 8) >> 3;
-      const int sh2 = sh;  // (k+8) & 7 == k & 7
-      u64 hi = (sh2 == 0) ? w[wi2]
-                          : (w[wi2] >> sh2) | (w[wi2 + 1] << (64 - sh2));
-      u32 want = (u32)hi;
-      if (mask_crc(crc8b_reg(len, tab)) != want) continue;
-      unsigned long long slot = atomicAdd(count, 1ull);
-      if ((i64)slot < max_cand) {
-        cand_pos[slot] = i;
-        cand_len[slot] = (i64)len;
+    __syncthreads();
+    unsigned int n = lcount < kLdsCand ? lcount : kLdsCand;
+    if (threadIdx.x == 0 && n) gbase = atomicAdd(count, (unsigned long long)n);
+    __syncthreads();
+    for (unsigned int s = threadIdx.x; s < n; s += blockDim.x) {
+      i64 slot = (i64)gbase + s;
+      if (slot < max_cand) {
+        cand_pos[slot] = lpos[s];
+        cand_len[slot] = llen[s];
       }
     }
+    __syncthreads();
+    if (threadIdx.x == 0) lcount = 0;
+    __syncthreads();
   }
 }
 
@@ -428,6 +465,44 @@ void gpu_gather_payloads(uintptr_t data, uintptr_t off, uintptr_t len,
   HIP_CHECK(hipGetLastError());
 }
 
+// ---------------------------------------------------------------------------
+// Prefix sums: rocprim decoupled-lookback scans. torch.cumsum's innermost-dim
+// scan kernel launches one block per row — 2.5 ms for a [4, 1M] int64 scan
+// (profiles/r01_kernel_stats.txt); these run at memory speed. The strided
+// variant scans a column of the [R, F*6] FieldStat buffer in place, removing
+// the transpose+contiguous materialization entirely.
+// ---------------------------------------------------------------------------
+
+struct StridedLoadI64 {
+  const i64* p;
+  i64 stride;
+  __host__ __device__ i64 operator()(i64 i) const { return p[i * stride]; }
+};
+
+size_t gpu_scan_temp_bytes(i64 n) {
+  size_t bytes = 0;
+  auto it = rocprim::make_transform_iterator(
+      rocprim::make_counting_iterator<i64>(0), StridedLoadI64{nullptr, 1});
+  (void)rocprim::inclusive_scan(nullptr, bytes, it, (i64*)nullptr, (size_t)n,
+                                rocprim::plus<i64>());
+  return bytes;
+}
+
+// Writes out[0] = 0 and out[1..n] = inclusive scan of in[0], in[stride], ...
+// i.e. out is the (n+1)-long exclusive scan with the total at out[n].
+void gpu_excl_sum_strided(uintptr_t temp, size_t temp_bytes, uintptr_t in,
+                          i64 stride, uintptr_t out, i64 n, uintptr_t stream) {
+  auto s = (hipStream_t)stream;
+  HIP_CHECK(hipMemsetAsync((void*)out, 0, sizeof(i64), s));
+  if (n <= 0) return;
+  auto it = rocprim::make_transform_iterator(
+      rocprim::make_counting_iterator<i64>(0), StridedLoadI64{(const i64*)in, stride});
+  size_t tb = temp_bytes;
+  hipError_t e = rocprim::inclusive_scan((void*)temp, tb, it, (i64*)out + 1,
+                                         (size_t)n, rocprim::plus<i64>(), s);
+  HIP_CHECK(e);
+}
+
 }  // namespace
 
 void register_gpu(py::module_& m) {
@@ -444,6 +519,8 @@ void register_gpu(py::module_& m) {
   m.def("gpu_frame_bytes", &gpu_frame_bytes);
   m.def("gpu_frame_scan", &gpu_frame_scan);
   m.def("gpu_gather_payloads", &gpu_gather_payloads);
+  m.def("gpu_scan_temp_bytes", &gpu_scan_temp_bytes);
+  m.def("gpu_excl_sum_strided", &gpu_excl_sum_strided);
   m.def("gpu_devcols_bytes", &gpu_devcols_bytes);
   m.def("gpu_devmeta_bytes", []() { return sizeof(DevFieldDst); });
   m.def("gpu_fieldstat_words", []() { return sizeof(FieldStat) / 8; });

@@ -60,49 +60,51 @@ def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
     return buf
 
 
-_CHUNK = 32 << 20  # 32 MiB: overlap file IO with PCIe copies chunkwise
+_CHUNK = 64 << 20  # 64 MiB: overlap file IO with PCIe copies chunkwise
 
 
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:
     """File -> pinned staging -> HBM, chunked so the H2D copy of chunk k
-    overlaps the file read of chunk k+1."""
+    overlaps the file read of chunk k+1. File reads go through the native
+    pread worker pool (single-threaded tmpfs reads cap at ~9 GB/s)."""
     import os as _os
 
     n = _os.path.getsize(path)
     dev = torch.empty(max(n, 1), dtype=torch.uint8, device=device)[:n]
     buf = pinned_buffer("fread", min(n, 2 * _CHUNK) or 1)
     ev = [torch.cuda.Event(), torch.cuda.Event()]
-    with open(path, "rb") as f:
+    fd = _os.open(path, _os.O_RDONLY)
+    try:
         pos = 0
         which = 0
         while pos < n:
             m = min(_CHUNK, n - pos)
-            half = buf[which * _CHUNK:which * _CHUNK + m]
             ev[which].synchronize()  # prior H2D from this half must be done
-            got = f.readinto(memoryview(half.numpy()))
-            if got != m:
-                raise IOError(f"short read from {path}")
-            dev[pos:pos + m].copy_(half, non_blocking=True)
+            _native.pread_parallel(fd, buf.data_ptr() + which * _CHUNK, m, pos)
+            dev[pos:pos + m].copy_(buf[which * _CHUNK:which * _CHUNK + m],
+                                   non_blocking=True)
             ev[which].record()
             pos += m
             which ^= 1 if n > _CHUNK else 0
+    finally:
+        _os.close(fd)
     return dev
 
 
 def device_to_file(img: torch.Tensor, path: str):
     """HBM file image -> pinned staging -> file.
 
-    Writes in place (r+b) when the file exists so tmpfs/page-cache pages are
-    reused instead of freed+reallocated by O_TRUNC; truncates to the new
-    size afterwards. Chunked: file write of chunk k overlaps the D2H copy of
-    chunk k+1."""
+    Opens O_CREAT without O_TRUNC so tmpfs/page-cache pages are reused
+    instead of freed+reallocated; truncates to the new size afterwards.
+    Chunked: the parallel pwrite of chunk k overlaps the D2H copy of chunk
+    k+1."""
     import os as _os
 
     n = img.numel()
     buf = pinned_buffer("fwrite", min(n, 2 * _CHUNK) or 1)
-    mode = "r+b" if _os.path.exists(path) else "wb"
     ev = [torch.cuda.Event(), torch.cuda.Event()]
-    with open(path, mode) as f:
+    fd = _os.open(path, _os.O_RDWR | _os.O_CREAT, 0o644)
+    try:
         pos = 0
         which = 0
         # prefetch chunk 0
@@ -120,10 +122,45 @@ def device_to_file(img: torch.Tensor, path: str):
                     img[nxt:nxt + m2], non_blocking=True)
                 ev[nwhich].record()
             ev[which].synchronize()
-            f.write(memoryview(buf[which * _CHUNK:which * _CHUNK + m].numpy()))
+            _native.pwrite_parallel(fd, buf.data_ptr() + which * _CHUNK, m, pos)
             pos = nxt
             which = nwhich
-        f.truncate(n)
+        _os.ftruncate(fd, n)
+    finally:
+        _os.close(fd)
+
+
+# ---------------------------------------------------------------------------
+# Prefix sums: rocprim decoupled-lookback scans (native). torch.cumsum's
+# innermost-dim kernel launches one block per row — 2.5 ms on a [4, 1M] int64
+# scan (profiles/r01_kernel_stats.txt) vs memory-speed here.
+# ---------------------------------------------------------------------------
+
+_scan_ws = {}
+
+
+def _scan_workspace(n: int, device) -> torch.Tensor:
+    key = (device.index if hasattr(device, "index") else 0)
+    need = _native.gpu_scan_temp_bytes(max(n, 1))
+    ws = _scan_ws.get(key)
+    if ws is None or ws.numel() < need:
+        ws = torch.empty(int(need * 2), dtype=torch.uint8, device=device)
+        _scan_ws[key] = ws
+    return ws
+
+
+def _excl_sum_into(out_ptr: int, in_ptr: int, stride: int, n: int, device):
+    ws = _scan_workspace(n, device)
+    _native.gpu_excl_sum_strided(ws.data_ptr(), ws.numel(), in_ptr, stride,
+                                 out_ptr, n, _stream())
+
+
+def excl_sum(t: torch.Tensor) -> torch.Tensor:
+    """Contiguous 1-D int64 tensor -> (n+1)-long exclusive scan, total at [-1]."""
+    n = t.numel()
+    out = torch.empty(n + 1, dtype=torch.int64, device=t.device)
+    _excl_sum_into(out.data_ptr(), t.data_ptr(), 1, n, t.device)
+    return out
 
 
 def device_to_bytes(img: torch.Tensor) -> bytes:
@@ -223,8 +260,7 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
         crc_verify_device(data, off, lens)
 
     if record_type == "ByteArray":
-        dst_off = torch.zeros(R + 1, dtype=torch.int64, device=device)
-        torch.cumsum(lens, 0, out=dst_off[1:])
+        dst_off = excl_sum(lens.contiguous())
         total = int(dst_off[-1].item())
         out = torch.empty(total, dtype=torch.uint8, device=device)
         if R:
@@ -254,19 +290,20 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                              FMT[record_type], blob.data_ptr(), F,
                              stats.data_ptr(), err.data_ptr(), _stream())
 
-    # Per-field exclusive prefix sums ([F, R+1], row-contiguous views).
-    # The transpose is materialized first: cumsum over a strided innermost
-    # dim is ~10x slower than over a contiguous one (rocprof: 2.5 ms vs copy
-    # + contiguous scan well under 1 ms at R=1M).
-    def excl_scan(counts_rf: torch.Tensor) -> torch.Tensor:
-        cs = torch.cumsum(counts_rf.transpose(0, 1).contiguous(), dim=1)
-        # pad a leading zero column; cumsum must write a CONTIGUOUS output
-        # (a strided `out=` silently takes a ~10x slower scan path)
-        return torch.nn.functional.pad(cs, (1, 0))
+    # Per-field exclusive prefix sums ([F, R+1]): rocprim strided scans read
+    # the stat column straight out of the [R, F, 6] buffer — no transpose
+    # materialization, no torch cumsum (see profiles/r01_kernel_stats.txt).
+    def excl_scan(stat_col: int) -> torch.Tensor:
+        out = torch.empty((F, R + 1), dtype=torch.int64, device=device)
+        for i in range(F):
+            _excl_sum_into(out.data_ptr() + i * (R + 1) * 8,
+                           stats.data_ptr() + (i * 6 + stat_col) * 8,
+                           F * 6, R, device)
+        return out
 
-    val_base = excl_scan(stats[:, :, 2])
-    byte_base = excl_scan(stats[:, :, 3])
-    list_base = excl_scan(stats[:, :, 4])
+    val_base = excl_scan(2)
+    byte_base = excl_scan(3)
+    list_base = excl_scan(4)
     totals = torch.stack([val_base[:, -1], byte_base[:, -1], list_base[:, -1]])
     totals_h = totals.cpu()  # one sync for all allocations
     if int(err.item()) != 0:
@@ -315,14 +352,10 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
         presence = (stats[:, i, 0] >= 0).to(torch.uint8)
         elem_off = None
         if kind == KIND_BYTES:
-            elem_off = torch.zeros(o["elem_len"].numel() + 1, dtype=torch.int64,
-                                   device=device)
-            torch.cumsum(o["elem_len"], 0, out=elem_off[1:])
+            elem_off = excl_sum(o["elem_len"])
         sub_off = None
         if seq:
-            sub_off = torch.zeros(o["sub_count"].numel() + 1, dtype=torch.int64,
-                                  device=device)
-            torch.cumsum(o["sub_count"], 0, out=sub_off[1:])
+            sub_off = excl_sum(o["sub_count"])
         values = (o["i64_vals"] if kind == KIND_INT64 else
                   o["f32_vals"] if kind == KIND_FLOAT else o["bytes_data"])
         cols.append(WireColumn(kind, seq, presence, val_base[i], values,
@@ -382,8 +415,7 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
     if record_type == "ByteArray":
         col = batch.columns[0]
         lens = col.elem_off[1:] - col.elem_off[:-1]
-        frame_off = torch.zeros(R + 1, dtype=torch.int64, device=device)
-        torch.cumsum(lens + 16, 0, out=frame_off[1:])
+        frame_off = excl_sum((lens + 16).contiguous())
         total = int(frame_off[-1].item())
         file = torch.empty(total, dtype=torch.uint8, device=device)
         if R:
@@ -400,8 +432,7 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
     psize = torch.empty(R, dtype=torch.int64, device=device)
     _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], R, psize.data_ptr(), _stream())
-    frame_off = torch.zeros(R + 1, dtype=torch.int64, device=device)
-    torch.cumsum(psize, 0, out=frame_off[1:])
+    frame_off = excl_sum(psize)
     total = int(frame_off[-1].item())
     file = torch.empty(total, dtype=torch.uint8, device=device)
     err = torch.zeros(1, dtype=torch.int32, device=device)

@@ -20,7 +20,14 @@
 #include <pybind11/stl.h>
 #include <rocprim/rocprim.hpp>
 
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
 #include <cstdint>
+#include <map>
+#include <mutex>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -503,6 +510,109 @@ void gpu_excl_sum_strided(uintptr_t temp, size_t temp_bytes, uintptr_t in,
   HIP_CHECK(e);
 }
 
+// ---------------------------------------------------------------------------
+// Pinned file mappings: mmap a tmpfs/page-cache file and hipHostRegister the
+// mapping so hipMemcpyAsync can DMA directly between HBM and the file's own
+// pages — no pinned bounce buffer, no host memcpy, no inode-locked write()
+// (parallel pwrite to one tmpfs file serializes on the inode mutex; mapped
+// DMA measured this at 27.7 ms -> ~4 ms for a 215 MB file image).
+// Mappings are cached per path keyed by (dev, ino, size) so the registration
+// cost amortizes across steps; an unlink+recreate changes the inode and
+// evicts the stale entry.
+// ---------------------------------------------------------------------------
+
+struct MappedFile {
+  void* ptr = nullptr;
+  size_t n = 0;
+  dev_t dev = 0;
+  ino_t ino = 0;
+  bool pinned = false;
+  bool writable = false;
+};
+
+std::map<std::string, MappedFile> g_mmap_cache;
+std::mutex g_mmap_mu;
+size_t g_mmap_bytes = 0;
+constexpr size_t kMmapCacheCap = 8ull << 30;  // 8 GiB of cached mappings
+
+void drop_mapping_locked(const std::string& key) {
+  auto it = g_mmap_cache.find(key);
+  if (it == g_mmap_cache.end()) return;
+  if (it->second.pinned) {
+    (void)hipDeviceSynchronize();  // a DMA into this mapping may be in flight
+    (void)hipHostUnregister(it->second.ptr);
+  }
+  munmap(it->second.ptr, it->second.n);
+  g_mmap_bytes -= it->second.n;
+  g_mmap_cache.erase(it);
+}
+
+// Returns (ptr, pinned). With writable=true the file is created/resized to n
+// first. pinned=false means hipHostRegister failed (fall back to the staged
+// pread/pwrite path); ptr is 0 iff n == 0.
+py::tuple file_mmap_pinned(const std::string& path, i64 n, bool writable) {
+  if (n <= 0) return py::make_tuple((uintptr_t)0, true);
+  std::lock_guard<std::mutex> lk(g_mmap_mu);
+  int flags = writable ? (O_RDWR | O_CREAT) : O_RDWR;
+  int fd = ::open(path.c_str(), flags, 0644);
+  bool rdonly = false;
+  if (fd < 0 && !writable) {  // read path on a read-only file
+    fd = ::open(path.c_str(), O_RDONLY);
+    rdonly = true;
+  }
+  if (fd < 0)
+    throw std::runtime_error("open failed: " + path + ": " + strerror(errno));
+  struct stat st {};
+  if (writable && ftruncate(fd, (off_t)n) != 0) {
+    int e = errno;
+    ::close(fd);
+    throw std::runtime_error("ftruncate failed: " + path + ": " + strerror(e));
+  }
+  if (fstat(fd, &st) != 0 || st.st_size < (off_t)n) {
+    ::close(fd);
+    throw std::runtime_error("stat/short file: " + path);
+  }
+  auto it = g_mmap_cache.find(path);
+  if (it != g_mmap_cache.end()) {
+    MappedFile& m = it->second;
+    if (m.dev == st.st_dev && m.ino == st.st_ino && m.n == (size_t)n &&
+        (m.writable || !writable)) {
+      ::close(fd);
+      return py::make_tuple((uintptr_t)m.ptr, m.pinned);
+    }
+    drop_mapping_locked(path);
+  }
+  while (g_mmap_bytes + (size_t)n > kMmapCacheCap && !g_mmap_cache.empty())
+    drop_mapping_locked(g_mmap_cache.begin()->first);
+  int prot = PROT_READ | ((writable || !rdonly) ? PROT_WRITE : 0);
+  void* p = mmap(nullptr, (size_t)n, prot, MAP_SHARED, fd, 0);
+  ::close(fd);
+  if (p == MAP_FAILED)
+    throw std::runtime_error("mmap failed: " + path + ": " + strerror(errno));
+  bool pinned = hipHostRegister(p, (size_t)n, hipHostRegisterDefault) == hipSuccess;
+  if (!pinned) (void)hipGetLastError();  // clear sticky error
+  MappedFile m{p, (size_t)n, st.st_dev, st.st_ino, pinned,
+               writable || !rdonly};
+  g_mmap_cache[path] = m;
+  g_mmap_bytes += (size_t)n;
+  return py::make_tuple((uintptr_t)p, pinned);
+}
+
+void file_mmap_drop(const std::string& path) {
+  std::lock_guard<std::mutex> lk(g_mmap_mu);
+  drop_mapping_locked(path);
+}
+
+void gpu_memcpy_d2h(uintptr_t dst, uintptr_t src, i64 n, uintptr_t stream) {
+  HIP_CHECK(hipMemcpyAsync((void*)dst, (const void*)src, (size_t)n,
+                           hipMemcpyDeviceToHost, (hipStream_t)stream));
+}
+
+void gpu_memcpy_h2d(uintptr_t dst, uintptr_t src, i64 n, uintptr_t stream) {
+  HIP_CHECK(hipMemcpyAsync((void*)dst, (const void*)src, (size_t)n,
+                           hipMemcpyHostToDevice, (hipStream_t)stream));
+}
+
 }  // namespace
 
 void register_gpu(py::module_& m) {
@@ -521,6 +631,11 @@ void register_gpu(py::module_& m) {
   m.def("gpu_gather_payloads", &gpu_gather_payloads);
   m.def("gpu_scan_temp_bytes", &gpu_scan_temp_bytes);
   m.def("gpu_excl_sum_strided", &gpu_excl_sum_strided);
+  m.def("file_mmap_pinned", &file_mmap_pinned, py::arg("path"), py::arg("n"),
+        py::arg("writable"));
+  m.def("file_mmap_drop", &file_mmap_drop);
+  m.def("gpu_memcpy_d2h", &gpu_memcpy_d2h);
+  m.def("gpu_memcpy_h2d", &gpu_memcpy_h2d);
   m.def("gpu_devcols_bytes", &gpu_devcols_bytes);
   m.def("gpu_devmeta_bytes", []() { return sizeof(DevFieldDst); });
   m.def("gpu_fieldstat_words", []() { return sizeof(FieldStat) / 8; });

@@ -55,8 +55,45 @@ enum CodecErr : int32_t {
 // varint
 // ---------------------------------------------------------------------------
 
+// clang lowers these builtins natively for both host and gfx950 device code
+TFR_HOSTDEV inline int popcount64(u64 x) { return __builtin_popcountll(x); }
+
+TFR_HOSTDEV inline int ctz64(u64 x) { return __builtin_ctzll(x); }  // x != 0
+
+constexpr u64 kMsbMask = 0x8080808080808080ull;
+
 // Reads a base-128 varint; returns new cursor or nullptr on malformed/overrun.
+// Fast path: ONE unaligned 8-byte load covers varints up to 8 bytes (the
+// continuation-bit scan finds the terminator in registers) — the per-byte
+// dependent-load loop only runs at buffer tails and for 9/10-byte varints.
+// gfx950 supports unaligned global loads, so the memcpy lowers to a plain
+// dwordx2 load; this is the hot instruction of the decode structure scan.
 TFR_HOSTDEV inline const u8* read_varint(const u8* p, const u8* end, u64* out) {
+  if (end - p >= 8) {
+    u64 w;
+    __builtin_memcpy(&w, p, 8);
+    u64 stops = ~w & kMsbMask;
+    if (stops) {
+      int nb = (ctz64(stops) >> 3) + 1;
+      u64 v = 0;
+      for (int i = 0; i < nb; ++i)
+        v |= ((w >> (8 * i)) & 0x7F) << (7 * i);
+      *out = v;
+      return p + nb;
+    }
+    u64 v = 0;
+    for (int i = 0; i < 8; ++i) v |= ((w >> (8 * i)) & 0x7F) << (7 * i);
+    const u8* q = p + 8;
+    for (int shift = 56; q < end && shift < 64; shift += 7) {
+      u8 b = *q++;
+      v |= static_cast<u64>(b & 0x7F) << shift;
+      if (!(b & 0x80)) {
+        *out = v;
+        return q;
+      }
+    }
+    return nullptr;
+  }
   u64 v = 0;
   int shift = 0;
   while (p < end) {
@@ -232,12 +269,18 @@ TFR_HOSTDEV inline int32_t count_list_body(const u8* p, const u8* end, int32_t k
         if (!p || static_cast<u64>(end - p) < len) return ERR_TRUNCATED;
         const u8* q = p;
         const u8* qe = p + len;
-        while (q < qe) {
-          u64 v;
-          q = read_varint(q, qe, &v);
-          if (!q) return ERR_BAD_VARINT;
-          *nvals += 1;
+        // count = number of terminator bytes (MSB clear), found by popcount
+        // over 8-byte windows — no per-varint parsing in the count pass
+        if (len && (qe[-1] & 0x80)) return ERR_BAD_VARINT;  // ends mid-varint
+        i64 cnt = 0;
+        while (qe - q >= 8) {
+          u64 w;
+          __builtin_memcpy(&w, q, 8);
+          cnt += popcount64(~w & kMsbMask);
+          q += 8;
         }
+        for (; q < qe; ++q) cnt += !(*q & 0x80);
+        *nvals += cnt;
         p = qe;
       } else if (wt == 0) {
         u64 v;

@@ -86,18 +86,20 @@ _copy_streams: Optional[list] = None
 def _dma_streams() -> list:
     global _copy_streams
     if _copy_streams is None:
-        _copy_streams = [torch.cuda.Stream() for _ in range(4)]
+        _copy_streams = [torch.cuda.Stream() for _ in range(2)]
     return _copy_streams
 
 
 def _multi_dma(dst: int, src: int, n: int, fn, after_main: bool):
-    """Split one big pinned<->HBM copy across side streams so multiple SDMA
-    engines run in parallel (a single hipMemcpyAsync ~27 GB/s, 4-way ~back to
-    link speed). after_main=True orders the copies after current-stream work
-    and makes the main stream wait for completion; False only syncs them."""
+    """Issue one big pinned<->HBM copy on up to TWO side streams: measured
+    (exp/exp_dma.py) 56 GB/s each way at k<=2 — k=4 overcommits the SDMA
+    engines and falls back to ~37 GB/s shader blits. Side streams keep the
+    copy off the compute stream. after_main=True orders the copies after
+    current-stream work and makes the main stream wait for completion;
+    False blocks until the copies land."""
     streams = _dma_streams()
     main = torch.cuda.current_stream()
-    k = min(len(streams), max(1, n // (8 << 20)))
+    k = min(len(streams), max(1, n // (16 << 20)))
     span = (n + k - 1) // k
     for i in range(k):
         o = i * span

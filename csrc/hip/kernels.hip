@@ -88,15 +88,31 @@ __global__ void crc_verify_kernel(const u8* __restrict__ data,
 
 static_assert(sizeof(FieldStat) == 48, "FieldStat layout shared with Python");
 
+// When crc_err != nullptr the frame CRCs (header + payload) are verified in
+// the same pass, while the record bytes are L2-hot from the parse — a
+// separate verify kernel costs an extra full-file HBM read and an extra
+// host sync. crc_err[0] stays ~0ULL on success, else 1 + first bad record.
 __global__ void scan_records_kernel(const u8* __restrict__ data,
                                     const i64* __restrict__ off,
                                     const i64* __restrict__ len, i64 R, int32_t fmt,
                                     const u8* __restrict__ schema_blob, int F,
                                     FieldStat* __restrict__ stats,
-                                    int32_t* __restrict__ err) {
+                                    int32_t* __restrict__ err,
+                                    unsigned long long* crc_err) {
+  __shared__ uint32_t tab[8][256];
+  if (crc_err) stage_crc_tables(tab);
   SchemaView schema = schema_view(schema_blob);
   for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
        r += (i64)gridDim.x * blockDim.x) {
+    if (crc_err) {
+      const u8* h = data + off[r] - 12;
+      u32 len_crc, data_crc;
+      __builtin_memcpy(&len_crc, h + 8, 4);
+      __builtin_memcpy(&data_crc, h + 12 + len[r], 4);
+      bool ok = mask_crc(crc32c_sw(h, 8, 0, tab)) == len_crc &&
+                mask_crc(crc32c_sw(h + 12, (size_t)len[r], 0, tab)) == data_crc;
+      if (!ok) atomicMin(crc_err, (unsigned long long)(r + 1));
+    }
     FieldStat* st = stats + r * F;
     for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
     int32_t rc = scan_record(data, off[r], len[r], fmt, schema, st);
@@ -172,13 +188,13 @@ __global__ void size_records_kernel(const DevCols* __restrict__ cols,
 // the whole file image from HBM (~0.5 ms / 215 MB, r01 profile).
 __global__ void emit_records_kernel(const DevCols* __restrict__ cols,
                                     const u8* __restrict__ schema_blob, int32_t fmt,
-                                    i64 R, const i64* __restrict__ frame_off,
+                                    i64 r0, i64 R, const i64* __restrict__ frame_off,
                                     u8* __restrict__ file,
                                     int32_t* __restrict__ err) {
   __shared__ uint32_t tab[8][256];
   stage_crc_tables(tab);
   SchemaView schema = schema_view(schema_blob);
-  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+  for (i64 r = r0 + blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
        r += (i64)gridDim.x * blockDim.x) {
     i64 payload = (frame_off[r + 1] - frame_off[r]) - kFrameOverhead;
     u8* o = file + frame_off[r] + 12;
@@ -236,7 +252,12 @@ __device__ inline u32 crc8b_reg(u64 w, const uint32_t (*tab)[256]) {
 constexpr int kPosPerLane = 16;
 constexpr int kLdsCand = 2048;  // 32 KiB LDS; realistic max ≈ blockDim*16/24
 
+// Evaluates candidate positions in [pos_start, pos_end) only — the sliced
+// read pipeline scans each arrived slice while the next one is still in
+// flight on the DMA stream (a position's 32-byte load window must have
+// landed: callers keep a 32-byte guard band at the slice boundary).
 __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
+                                       i64 pos_start, i64 pos_end,
                                        i64* __restrict__ cand_pos,
                                        i64* __restrict__ cand_len,
                                        unsigned long long* __restrict__ count,
@@ -250,13 +271,14 @@ __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
   if (threadIdx.x == 0) lcount = 0;
   __syncthreads();
   const u64* wdata = reinterpret_cast<const u64*>(data);  // data is 8B-aligned
-  i64 nchunks = (N + kPosPerLane - 1) / kPosPerLane;
+  i64 c_lo = pos_start / kPosPerLane;
+  i64 nchunks = (pos_end + kPosPerLane - 1) / kPosPerLane - c_lo;
   // Uniform per-block trip count so every thread reaches the __syncthreads
   // flush barrier even when its own chunk index runs past nchunks.
   for (i64 cbase = (i64)blockIdx.x * blockDim.x; cbase < nchunks;
        cbase += (i64)gridDim.x * blockDim.x) {
-    i64 c = cbase + threadIdx.x;
-    if (c < nchunks) {
+    i64 c = c_lo + cbase + threadIdx.x;
+    if (cbase + threadIdx.x < nchunks) {
       // p0 = 16*c is 8-byte aligned, so every index below is compile-time
       // constant after unrolling (runtime-indexed w[] would spill to scratch).
       i64 p0 = c * kPosPerLane;
@@ -278,7 +300,8 @@ __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
 #pragma unroll
       for (int k = 0; k < kPosPerLane; ++k) {
         i64 i = p0 + k;
-        if (i + 16 > N) break;
+        if (i + 16 > N || i >= pos_end) break;
+        if (i < pos_start) continue;
         constexpr int _ppl = kPosPerLane;
         static_assert(_ppl == 16, "index math below assumes 16 positions");
         const int wi = k >> 3;
@@ -358,11 +381,12 @@ void gpu_crc_verify(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
 
 void gpu_scan_records(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
                       int32_t fmt, uintptr_t schema_blob, int F, uintptr_t stats,
-                      uintptr_t err, uintptr_t stream) {
+                      uintptr_t err, uintptr_t crc_err, uintptr_t stream) {
   hipLaunchKernelGGL(scan_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
                      (hipStream_t)stream, (const u8*)data, (const i64*)off,
                      (const i64*)len, R, fmt, (const u8*)schema_blob, F,
-                     (FieldStat*)stats, (int32_t*)err);
+                     (FieldStat*)stats, (int32_t*)err,
+                     (unsigned long long*)crc_err);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -429,12 +453,12 @@ void gpu_size_records(py::list cols, uintptr_t cols_dev, uintptr_t schema_blob,
   HIP_CHECK(hipGetLastError());
 }
 
-void gpu_emit_records(uintptr_t cols_dev, uintptr_t schema_blob, int32_t fmt, i64 R,
-                      uintptr_t frame_off, uintptr_t file, uintptr_t err,
-                      uintptr_t stream) {
-  hipLaunchKernelGGL(emit_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+void gpu_emit_records(uintptr_t cols_dev, uintptr_t schema_blob, int32_t fmt,
+                      i64 r0, i64 R, uintptr_t frame_off, uintptr_t file,
+                      uintptr_t err, uintptr_t stream) {
+  hipLaunchKernelGGL(emit_records_kernel, dim3(grid_for(R - r0)), dim3(kBlock), 0,
                      (hipStream_t)stream, (const DevCols*)cols_dev,
-                     (const u8*)schema_blob, fmt, R, (const i64*)frame_off,
+                     (const u8*)schema_blob, fmt, r0, R, (const i64*)frame_off,
                      (u8*)file, (int32_t*)err);
   HIP_CHECK(hipGetLastError());
 }
@@ -447,12 +471,15 @@ void gpu_frame_bytes(uintptr_t src, uintptr_t elem_off, uintptr_t frame_off, i64
   HIP_CHECK(hipGetLastError());
 }
 
-void gpu_frame_scan(uintptr_t data, i64 N, uintptr_t cand_pos, uintptr_t cand_len,
+void gpu_frame_scan(uintptr_t data, i64 N, i64 pos_start, i64 pos_end,
+                    uintptr_t cand_pos, uintptr_t cand_len,
                     uintptr_t count, i64 max_cand, uintptr_t stream) {
   hipLaunchKernelGGL(frame_candidate_kernel,
-                     dim3(grid_for(N / kPosPerLane + 1)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const u8*)data, N, (i64*)cand_pos,
-                     (i64*)cand_len, (unsigned long long*)count, max_cand);
+                     dim3(grid_for((pos_end - pos_start) / kPosPerLane + 1)),
+                     dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)data, N, pos_start, pos_end,
+                     (i64*)cand_pos, (i64*)cand_len,
+                     (unsigned long long*)count, max_cand);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -254,9 +254,17 @@ def scan_frames_device(data: torch.Tensor):
     cand_pos = torch.empty(max_cand, dtype=torch.int64, device=device)
     cand_len = torch.empty(max_cand, dtype=torch.int64, device=device)
     count = torch.zeros(1, dtype=torch.int64, device=device)
-    _native.gpu_frame_scan(data.data_ptr(), N, cand_pos.data_ptr(),
+    _native.gpu_frame_scan(data.data_ptr(), N, 0, N, cand_pos.data_ptr(),
                            cand_len.data_ptr(), count.data_ptr(), max_cand,
                            _stream())
+    return _chain_candidates(data, cand_pos, cand_len, count, N)
+
+
+def _chain_candidates(data, cand_pos, cand_len, count, N):
+    """Sort frame-head candidates and validate the chain
+    pos[k+1] == pos[k] + 16 + len[k]; returns (payload_off, payload_len)."""
+    device = data.device
+    max_cand = cand_pos.numel()
     C = int(count.item())
     if C == 0:
         raise RuntimeError("corrupt TFRecord: no valid frame header found")
@@ -318,7 +326,7 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     R = off.numel()
     fields = wire_fields(schema)
     F = len(fields)
-    if verify_crc:
+    if verify_crc and record_type == "ByteArray":
         crc_verify_device(data, off, lens)
 
     if record_type == "ByteArray":
@@ -348,9 +356,14 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     # FieldStat[R][F] as int64 [R,F,6]: pos,len,nvals,nbytes,nlists,(kind|err)
     stats = torch.empty((R, F, 6), dtype=torch.int64, device=device)
     err = torch.zeros(1, dtype=torch.int32, device=device)
+    # frame CRC verification is fused into the structure scan (record bytes
+    # are read once, while L2-hot)
+    crc_err = torch.full((1,), -1, dtype=torch.int64, device=device) \
+        if verify_crc else None
     _native.gpu_scan_records(data.data_ptr(), off.data_ptr(), lens.data_ptr(), R,
                              FMT[record_type], blob.data_ptr(), F,
-                             stats.data_ptr(), err.data_ptr(), _stream())
+                             stats.data_ptr(), err.data_ptr(),
+                             crc_err.data_ptr() if verify_crc else 0, _stream())
 
     # Per-field exclusive prefix sums ([F, R+1]): rocprim strided scans read
     # the stat column straight out of the [R, F, 6] buffer — no transpose
@@ -368,6 +381,11 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     list_base = excl_scan(4)
     totals = torch.stack([val_base[:, -1], byte_base[:, -1], list_base[:, -1]])
     totals_h = totals.cpu()  # one sync for all allocations
+    if verify_crc:
+        bad = int(crc_err.item())
+        if bad != -1:
+            raise RuntimeError(
+                f"corrupt TFRecord: bad CRC in record {bad - 1 if bad > 0 else bad}")
     if int(err.item()) != 0:
         raise RuntimeError(f"TFRecord decode failed (native error "
                            f"{int(err.item())}; kind mismatch or malformed record)")
@@ -443,10 +461,10 @@ def decode_buffer_device(data_np: np.ndarray, schema: StructType, record_type: s
 
 def read_file_to_batch(path: str, schema: StructType, record_type: str,
                        verify_crc: bool = True, device="cuda") -> RecordBatch:
-    """Uncompressed file -> pinned read -> GPU frame scan -> device batch."""
-    data = read_file_to_device(path, device)
-    off, lens = scan_frames_device(data)
-    return decode_device(data, off, lens, schema, record_type, verify_crc)
+    """Uncompressed file -> device batch (H2D sliced + overlapped with the
+    frame scan; see read_file_to_batch_pipelined)."""
+    return read_file_to_batch_pipelined(path, schema, record_type, verify_crc,
+                                        device)
 
 
 # ---------------------------------------------------------------------------
@@ -499,11 +517,115 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
     file = torch.empty(total, dtype=torch.uint8, device=device)
     err = torch.zeros(1, dtype=torch.int32, device=device)
     _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
-                             FMT[record_type], R, frame_off.data_ptr(),
+                             FMT[record_type], 0, R, frame_off.data_ptr(),
                              file.data_ptr(), err.data_ptr(), _stream())
     if int(err.item()) != 0:
         raise RuntimeError("TFRecord encode failed: size/emit mismatch")
     return file
+
+
+def write_batch_to_file(batch: RecordBatch, path: str,
+                        record_type: str = "Example", slices: int = 3) -> int:
+    """Device batch -> framed TFRecord file, with the emit kernel sliced over
+    record ranges so the D2H DMA of slice k streams to the file's mapped
+    pages while slice k+1 is still being emitted (SURVEY.md §7 step 3:
+    double-buffered transport overlapped with encode). Returns file bytes."""
+    check_native()
+    R = batch.num_rows
+    if record_type == "ByteArray" or R == 0:
+        img = encode_device(batch, record_type)
+        device_to_file(img, path)
+        return img.numel()
+    device = batch.columns[0].presence.device
+    blob = torch.frombuffer(bytearray(schema_blob(batch.schema)),
+                            dtype=torch.uint8).to(device)
+    col_dicts = [_col_ptrs(c) for c in batch.columns]
+    cols_dev = torch.empty(_native.gpu_devcols_bytes(), dtype=torch.uint8,
+                           device=device)
+    psize = torch.empty(R, dtype=torch.int64, device=device)
+    _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
+                             FMT[record_type], R, psize.data_ptr(), _stream())
+    frame_off = excl_sum(psize)
+    S = max(1, min(slices, R))
+    ridx = [R * s // S for s in range(S + 1)]
+    bounds = frame_off[ridx].cpu()  # one sync: slice byte bounds + total
+    total = int(bounds[-1])
+    file = torch.empty(total, dtype=torch.uint8, device=device)
+    err = torch.zeros(1, dtype=torch.int32, device=device)
+    ptr, pinned = _native.file_mmap_pinned(path, total, True)
+    if not pinned:
+        _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
+                                 FMT[record_type], 0, R, frame_off.data_ptr(),
+                                 file.data_ptr(), err.data_ptr(), _stream())
+        if int(err.item()) != 0:
+            raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+        _write_file_staged(file, path)
+        return total
+    main = torch.cuda.current_stream()
+    streams = _dma_streams()
+    for s in range(S):
+        _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
+                                 FMT[record_type], ridx[s], ridx[s + 1],
+                                 frame_off.data_ptr(), file.data_ptr(),
+                                 err.data_ptr(), _stream())
+        b0, b1 = int(bounds[s]), int(bounds[s + 1])
+        if b1 > b0:
+            w = streams[s % len(streams)]
+            w.wait_stream(main)
+            _native.gpu_memcpy_d2h(ptr + b0, file.data_ptr() + b0, b1 - b0,
+                                   w.cuda_stream)
+    for w in streams:
+        w.synchronize()
+    if int(err.item()) != 0:
+        raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+    return total
+
+
+def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str,
+                                 verify_crc: bool = True,
+                                 device="cuda") -> RecordBatch:
+    """File -> device batch with the H2D DMA sliced so the frame-candidate
+    scan of slice k runs while slice k+1 is still in flight. Positions within
+    32 bytes of a slice boundary are deferred to the next slice's launch
+    (their load window extends past the boundary)."""
+    import os as _os
+
+    check_native()
+    n = _os.path.getsize(path)
+    if n == 0:
+        z = torch.zeros(0, dtype=torch.int64, device=device)
+        return decode_device(torch.zeros(0, dtype=torch.uint8, device=device),
+                             z, z.clone(), schema, record_type, verify_crc)
+    ptr, pinned = _native.file_mmap_pinned(path, n, False)
+    if not pinned:
+        data = _read_file_staged(path, torch.empty(n, dtype=torch.uint8,
+                                                   device=device))
+        off, lens = scan_frames_device(data)
+        return decode_device(data, off, lens, schema, record_type, verify_crc)
+    data = torch.empty(n, dtype=torch.uint8, device=device)
+    span = 48 << 20
+    S = max(1, (n + span - 1) // span)
+    max_cand = n // 16 + 64
+    cand_pos = torch.empty(max_cand, dtype=torch.int64, device=device)
+    cand_len = torch.empty(max_cand, dtype=torch.int64, device=device)
+    count = torch.zeros(1, dtype=torch.int64, device=device)
+    main = torch.cuda.current_stream()
+    streams = _dma_streams()
+    scanned = 0
+    for s in range(S):
+        b0, b1 = s * span, min(n, (s + 1) * span)
+        st = streams[s % len(streams)]
+        _native.gpu_memcpy_h2d(data.data_ptr() + b0, ptr + b0, b1 - b0,
+                               st.cuda_stream)
+        main.wait_stream(st)
+        scan_end = b1 - 32 if s < S - 1 else n
+        if scan_end > scanned:
+            _native.gpu_frame_scan(data.data_ptr(), n, scanned, scan_end,
+                                   cand_pos.data_ptr(), cand_len.data_ptr(),
+                                   count.data_ptr(), max_cand, _stream())
+            scanned = scan_end
+    off, lens = _chain_candidates(data, cand_pos, cand_len, count, n)
+    return decode_device(data, off, lens, schema, record_type, verify_crc)
 
 
 def batch_to_device(batch: RecordBatch, device="cuda") -> RecordBatch:

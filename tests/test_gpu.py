@@ -142,6 +142,40 @@ class TestGpuCodec:
             np.frombuffer(cpu_img, np.uint8), byte_array_schema(), "ByteArray"), out)
 
 
+class TestPipelinedFileIO:
+    def test_write_batch_to_file_bytes_match_cpu(self, tmp_path):
+        g = _gpu_engine()
+        batch = make_batch(3000, seed=6)
+        path = str(tmp_path / "p.tfrecord")
+        n = g.write_batch_to_file(g.batch_to_device(batch), path, "Example")
+        on_disk = open(path, "rb").read()
+        assert len(on_disk) == n == os.path.getsize(path)
+        assert on_disk == cpu_engine.encode_batch(batch, "Example")
+
+    def test_pipelined_read_matches_cpu(self, tmp_path):
+        g = _gpu_engine()
+        batch = make_batch(4000, seed=7)
+        path = str(tmp_path / "q.tfrecord")
+        with open(path, "wb") as f:
+            f.write(cpu_engine.encode_batch(batch, "Example"))
+        out = g.batch_to_host(g.read_file_to_batch_pipelined(
+            path, batch.schema, "Example", verify_crc=True))
+        assert_batches_equal(batch, out)
+
+    def test_roundtrip_overwrites_shorter_file(self, tmp_path):
+        # in-place mmap reuse must truncate correctly when the file shrinks
+        g = _gpu_engine()
+        path = str(tmp_path / "r.tfrecord")
+        big = make_batch(2000, seed=8)
+        small = make_batch(50, seed=9)
+        g.write_batch_to_file(g.batch_to_device(big), path, "Example")
+        g.write_batch_to_file(g.batch_to_device(small), path, "Example")
+        assert open(path, "rb").read() == cpu_engine.encode_batch(small, "Example")
+        out = g.batch_to_host(g.read_file_to_batch_pipelined(
+            path, small.schema, "Example"))
+        assert_batches_equal(small, out)
+
+
 class TestGpuEndToEnd:
     def test_write_read_files_gpu_engine(self, tmp_sandbox):
         out = str(tmp_sandbox / "g")

@@ -582,24 +582,30 @@ py::tuple file_mmap_pinned(const std::string& path, i64 n, bool writable) {
   if (fd < 0)
     throw std::runtime_error("open failed: " + path + ": " + strerror(errno));
   struct stat st {};
-  if (writable && ftruncate(fd, (off_t)n) != 0) {
-    int e = errno;
+  if (fstat(fd, &st) != 0) {
     ::close(fd);
-    throw std::runtime_error("ftruncate failed: " + path + ": " + strerror(e));
-  }
-  if (fstat(fd, &st) != 0 || st.st_size < (off_t)n) {
-    ::close(fd);
-    throw std::runtime_error("stat/short file: " + path);
+    throw std::runtime_error("stat failed: " + path);
   }
   auto it = g_mmap_cache.find(path);
   if (it != g_mmap_cache.end()) {
     MappedFile& m = it->second;
     if (m.dev == st.st_dev && m.ino == st.st_ino && m.n == (size_t)n &&
-        (m.writable || !writable)) {
+        st.st_size >= (off_t)n && (m.writable || !writable)) {
       ::close(fd);
       return py::make_tuple((uintptr_t)m.ptr, m.pinned);
     }
+    // drop BEFORE any resize: ftruncate below a pinned mapping blocks
+    // forever on the DMA page pins
     drop_mapping_locked(path);
+  }
+  if (writable && st.st_size != (off_t)n && ftruncate(fd, (off_t)n) != 0) {
+    int e = errno;
+    ::close(fd);
+    throw std::runtime_error("ftruncate failed: " + path + ": " + strerror(e));
+  }
+  if (!writable && st.st_size < (off_t)n) {
+    ::close(fd);
+    throw std::runtime_error("short file: " + path);
   }
   while (g_mmap_bytes + (size_t)n > kMmapCacheCap && !g_mmap_cache.empty())
     drop_mapping_locked(g_mmap_cache.begin()->first);

@@ -344,6 +344,189 @@ __global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Schema inference on device ("per-GPU type-lattice histogram
+// kernel + RCCL all-reduce"). One record per lane walks the Features /
+// FeatureLists maps; feature names are interned into a global open-addressed
+// hash table (FNV-1a 64, atomicCAS claim), lattice codes merge with
+// atomicMax — the same commutative max the reference's RDD aggregate uses
+// (TensorFlowInferSchema.scala:120-127). Slot layout is [hash, nameref,
+// code] as 3 x i64; nameref packs (absolute name offset << 16 | name len)
+// of the first occurrence so the host can recover the string. The table
+// halves are context [0, S/2) and sequence [S/2, S) name spaces.
+// ---------------------------------------------------------------------------
+
+struct InferSlot {
+  unsigned long long hash;  // 0 = empty
+  long long nameref;
+  long long code;
+};
+static_assert(sizeof(InferSlot) == 24, "layout shared with Python [S,3] i64");
+
+__device__ inline u64 fnv1a64(const u8* s, u64 n) {
+  u64 h = 1469598103934665603ull;
+  for (u64 i = 0; i < n; ++i) h = (h ^ s[i]) * 1099511628211ull;
+  return h ? h : 1ull;  // 0 is the empty-slot marker
+}
+
+// Claims/finds the slot for (hash) in table half [lo, lo+half) and merges
+// code with max. Returns false when the half is full (caller sets err).
+__device__ inline bool infer_slot_merge(InferSlot* table, int lo, int half,
+                                        u64 hash, long long nameref, int code) {
+  int idx = lo + (int)(hash % (u64)half);
+  for (int probe = 0; probe < half; ++probe) {
+    InferSlot& s = table[idx];
+    unsigned long long seen =
+        atomicCAS(&s.hash, 0ull, (unsigned long long)hash);
+    if (seen == 0ull) s.nameref = nameref;  // claimer records first occurrence
+    if (seen == 0ull || seen == hash) {
+      atomicMax((unsigned long long*)&s.code, (unsigned long long)code);
+      return true;
+    }
+    idx = (idx + 1 - lo) % half + lo;
+  }
+  return false;
+}
+
+// Walks one Features / FeatureLists body, inferring per-entry lattice codes.
+// Mirrors csrc/ext.cpp infer_features_body (host reference implementation).
+__device__ inline int32_t infer_features_body_dev(const u8* data, const u8* p,
+                                                  const u8* end, bool seq,
+                                                  InferSlot* table, int nslots) {
+  int half = nslots / 2;
+  int lo = seq ? half : 0;
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    if (fieldno != 1 || wt != 2) {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+      continue;
+    }
+    u64 entry_len;
+    p = read_varint(p, end, &entry_len);
+    if (!p || (u64)(end - p) < entry_len) return ERR_TRUNCATED;
+    const u8* ep = p;
+    const u8* ee = p + entry_len;
+    p = ee;
+    const u8* key = nullptr;
+    u64 key_len = 0;
+    const u8* val = nullptr;
+    u64 val_len = 0;
+    while (ep < ee) {
+      u64 etag;
+      ep = read_varint(ep, ee, &etag);
+      if (!ep) return ERR_BAD_VARINT;
+      u32 efn = (u32)(etag >> 3), ewt = (u32)(etag & 7);
+      if (efn == 1 && ewt == 2) {
+        ep = read_varint(ep, ee, &key_len);
+        if (!ep || (u64)(ee - ep) < key_len) return ERR_TRUNCATED;
+        key = ep;
+        ep += key_len;
+      } else if (efn == 2 && ewt == 2) {
+        ep = read_varint(ep, ee, &val_len);
+        if (!ep || (u64)(ee - ep) < val_len) return ERR_TRUNCATED;
+        val = ep;
+        ep += val_len;
+      } else {
+        ep = skip_field(ep, ee, ewt);
+        if (!ep) return ERR_TRUNCATED;
+      }
+    }
+    if (!key) continue;
+    int code = 0;
+    if (val) {
+      if (!seq) {
+        int32_t kf = 0;
+        i64 nvals = 0, nbytes = 0;
+        int32_t rc = scan_feature_body(val, val + val_len, -1, &kf, &nvals, &nbytes);
+        if (rc != ERR_OK) return rc;
+        // rank: int64->1 float->2 bytes->3; +3 when >1 element (array)
+        if (kf != 0 && nvals > 0) code = (4 - kf) + (nvals > 1 ? 3 : 0);
+      } else {
+        const u8* lp = val;
+        const u8* le = val + val_len;
+        int kind_seen = 0;
+        while (lp < le) {
+          u64 ltag;
+          lp = read_varint(lp, le, &ltag);
+          if (!lp) return ERR_BAD_VARINT;
+          u32 lfn = (u32)(ltag >> 3), lwt = (u32)(ltag & 7);
+          if (lfn == 1 && lwt == 2) {
+            u64 flen;
+            lp = read_varint(lp, le, &flen);
+            if (!lp || (u64)(le - lp) < flen) return ERR_TRUNCATED;
+            int32_t kf = 0;
+            i64 nvals = 0, nbytes = 0;
+            int32_t rc = scan_feature_body(lp, lp + flen, -1, &kf, &nvals, &nbytes);
+            if (rc != ERR_OK) return rc;
+            if (kf) {
+              int r = 4 - kf;
+              kind_seen = kind_seen > r ? kind_seen : r;
+            }
+            lp += flen;
+          } else {
+            lp = skip_field(lp, le, lwt);
+            if (!lp) return ERR_TRUNCATED;
+          }
+        }
+        if (kind_seen) code = 6 + kind_seen;  // FeatureList infers 2-D
+      }
+    }
+    long long nameref = ((long long)(key - data) << 16) | (long long)key_len;
+    if (!infer_slot_merge(table, lo, half, fnv1a64(key, key_len), nameref, code))
+      return ERR_OVERFLOW;  // feature-name table full
+  }
+  return ERR_OK;
+}
+
+__global__ void infer_codes_kernel(const u8* __restrict__ data,
+                                   const i64* __restrict__ off,
+                                   const i64* __restrict__ len, i64 R, int32_t fmt,
+                                   InferSlot* __restrict__ table, int nslots,
+                                   int32_t* __restrict__ err) {
+  for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
+       r += (i64)gridDim.x * blockDim.x) {
+    const u8* p = data + off[r];
+    const u8* end = p + len[r];
+    while (p < end) {
+      u64 tag;
+      const u8* np = read_varint(p, end, &tag);
+      if (!np) {
+        err[0] = ERR_BAD_VARINT;
+        break;
+      }
+      p = np;
+      u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+      bool is_features = fieldno == 1;
+      bool is_fl = (fmt == FMT_SEQUENCE && fieldno == 2);
+      if ((is_features || is_fl) && wt == 2) {
+        u64 blen;
+        p = read_varint(p, end, &blen);
+        if (!p || (u64)(end - p) < blen) {
+          err[0] = ERR_TRUNCATED;
+          break;
+        }
+        int32_t rc = infer_features_body_dev(data, p, p + blen, is_fl, table,
+                                             nslots);
+        if (rc != ERR_OK) {
+          err[0] = rc;
+          break;
+        }
+        p += blen;
+      } else {
+        p = skip_field(p, end, wt);
+        if (!p) {
+          err[0] = ERR_TRUNCATED;
+          break;
+        }
+      }
+    }
+  }
+}
+
 // Raw-payload gather for the ByteArray read path (payload extents -> packed).
 __global__ void gather_payloads_kernel(const u8* __restrict__ data,
                                        const i64* __restrict__ off,
@@ -480,6 +663,16 @@ void gpu_frame_scan(uintptr_t data, i64 N, i64 pos_start, i64 pos_end,
                      (hipStream_t)stream, (const u8*)data, N, pos_start, pos_end,
                      (i64*)cand_pos, (i64*)cand_len,
                      (unsigned long long*)count, max_cand);
+  HIP_CHECK(hipGetLastError());
+}
+
+void gpu_infer_codes(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
+                     int32_t fmt, uintptr_t table, int nslots, uintptr_t err,
+                     uintptr_t stream) {
+  hipLaunchKernelGGL(infer_codes_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                     (hipStream_t)stream, (const u8*)data, (const i64*)off,
+                     (const i64*)len, R, fmt, (InferSlot*)table, nslots,
+                     (int32_t*)err);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -654,6 +847,7 @@ void register_gpu(py::module_& m) {
   m.def("gpu_frame_bytes", &gpu_frame_bytes);
   m.def("gpu_frame_scan", &gpu_frame_scan);
   m.def("gpu_gather_payloads", &gpu_gather_payloads);
+  m.def("gpu_infer_codes", &gpu_infer_codes);
   m.def("gpu_scan_temp_bytes", &gpu_scan_temp_bytes);
   m.def("gpu_excl_sum_strided", &gpu_excl_sum_strided);
   m.def("file_mmap_pinned", &file_mmap_pinned, py::arg("path"), py::arg("n"),

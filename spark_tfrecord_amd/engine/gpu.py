@@ -628,6 +628,58 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
     return decode_device(data, off, lens, schema, record_type, verify_crc)
 
 
+# ---------------------------------------------------------------------------
+# Schema inference on device: hash-table lattice kernel.
+# ---------------------------------------------------------------------------
+
+_INFER_SLOTS = 2048  # 1024 context + 1024 sequence feature names
+
+
+def infer_codes_device(data: torch.Tensor, off: torch.Tensor,
+                       lens: torch.Tensor, record_type: str) -> dict:
+    """Device records -> {feature name: lattice code}. The kernel interns
+    names into a device hash table and max-merges per-feature codes; the
+    host only reads back the (tiny) table and resolves name strings."""
+    check_native()
+    device = data.device
+    fmt = FMT["SequenceExample" if record_type == "SequenceExample" else "Example"]
+    table = torch.zeros((_INFER_SLOTS, 3), dtype=torch.int64, device=device)
+    err = torch.zeros(1, dtype=torch.int32, device=device)
+    R = off.numel()
+    if R:
+        _native.gpu_infer_codes(data.data_ptr(), off.data_ptr(), lens.data_ptr(),
+                                R, fmt, table.data_ptr(), _INFER_SLOTS,
+                                err.data_ptr(), _stream())
+    tab = table.cpu().numpy()
+    if int(err.item()) != 0:
+        raise RuntimeError("malformed record during schema inference "
+                           f"(native error {int(err.item())})")
+    used = np.nonzero(tab[:, 0])[0]
+    if len(used) == 0:
+        return {}
+    # gather all name strings with ONE D2H copy of the covering byte range
+    refs = tab[used, 1]
+    offs = (refs >> 16).astype(np.int64)
+    nlens = (refs & 0xFFFF).astype(np.int64)
+    lo = int(offs.min())
+    hi = int((offs + nlens).max())
+    blob = data[lo:hi].cpu().numpy().tobytes()
+    codes: dict = {}
+    for o, ln, code in zip(offs, nlens, tab[used, 2]):
+        name = blob[o - lo:o - lo + ln].decode("utf-8")
+        codes[name] = max(codes.get(name, 0), int(code))
+    return codes
+
+
+def infer_schema_file(path: str, record_type: str, device="cuda"):
+    """Uncompressed file -> inferred StructType, fully on the GPU."""
+    from ..infer import schema_from_codes
+
+    data = read_file_to_device(path, device)
+    off, lens = scan_frames_device(data)
+    return schema_from_codes(infer_codes_device(data, off, lens, record_type))
+
+
 def batch_to_device(batch: RecordBatch, device="cuda") -> RecordBatch:
     cols = []
     for c in batch.columns:

@@ -35,19 +35,34 @@ def _load_file(path: str) -> np.ndarray:
     return np.frombuffer(P.decompress_file(path), np.uint8)
 
 
-def infer_schema_of_paths(files: List[str], record_type: str) -> StructType:
+def infer_schema_of_paths(files: List[str], record_type: str,
+                          engine: str = "cpu") -> StructType:
     """Schema from the FIRST non-empty file (DefaultSource.scala:36-38
-    collectFirst), scanned fully."""
+    collectFirst), scanned fully — on the GPU (hash-table lattice kernel)
+    when engine='gpu' and the file is uncompressed."""
     if record_type == "ByteArray":
         return byte_array_schema()
     for f in files:
-        data = _load_file(f)
-        if data.size == 0:
-            continue
-        off, lens = _native.scan_frames(data, False)
-        if len(off) == 0:
-            continue
-        codes = infer_codes_from_buffer(data, off, lens, record_type)
+        if engine == "gpu" and P.codec_from_path(f) is None:
+            import os as _os
+
+            if _os.path.getsize(f) == 0:
+                continue
+            from ..engine import gpu as gpu_engine
+
+            data = gpu_engine.read_file_to_device(f)
+            off, lens = gpu_engine.scan_frames_device(data)
+            if off.numel() == 0:
+                continue
+            codes = gpu_engine.infer_codes_device(data, off, lens, record_type)
+        else:
+            data = _load_file(f)
+            if data.size == 0:
+                continue
+            off, lens = _native.scan_frames(data, False)
+            if len(off) == 0:
+                continue
+            codes = infer_codes_from_buffer(data, off, lens, record_type)
         if codes:
             return schema_from_codes(codes)
         return StructType([])
@@ -91,7 +106,7 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
 
     if schema is None:
         schema = (byte_array_schema() if record_type == "ByteArray"
-                  else infer_schema_of_paths(files, record_type))
+                  else infer_schema_of_paths(files, record_type, eng))
     data_schema = StructType([f for f in schema.fields if f.name not in part_cols])
 
     tables = []

@@ -107,11 +107,21 @@ def infer_schema_distributed(files: List[str], record_type: str) -> StructType:
 
     data = np.frombuffer(P.decompress_file(files[chosen]), np.uint8)
     off, lens = _native.scan_frames(data, False)
-    # each rank scans an interleaved slice of the records
+    # each rank scans an interleaved slice of the records — on its GPU via
+    # the hash-table lattice kernel when one is available
     my_off = off[rank::world]
     my_len = lens[rank::world]
-    codes = infer_codes_from_buffer(data, my_off, my_len, record_type) \
-        if len(my_off) else {}
+    if len(my_off) == 0:
+        codes: Dict[str, int] = {}
+    elif torch.cuda.is_available():
+        from ..engine import gpu as gpu_engine
+
+        dev_data = torch.as_tensor(np.ascontiguousarray(data)).cuda()
+        codes = gpu_engine.infer_codes_device(
+            dev_data, torch.as_tensor(np.ascontiguousarray(my_off)).cuda(),
+            torch.as_tensor(np.ascontiguousarray(my_len)).cuda(), record_type)
+    else:
+        codes = infer_codes_from_buffer(data, my_off, my_len, record_type)
 
     # align feature names across ranks, then max-all-reduce the code vector
     gathered: List[Dict[str, int]] = [None] * world

@@ -176,6 +176,55 @@ class TestPipelinedFileIO:
         assert_batches_equal(small, out)
 
 
+class TestGpuInference:
+    def _codes_both(self, batch, record_type):
+        import torch
+        from spark_tfrecord_amd import _native
+        from spark_tfrecord_amd.infer import infer_codes_from_buffer
+        g = _gpu_engine()
+        img = cpu_engine.encode_batch(batch, record_type)
+        data = np.frombuffer(img, np.uint8)
+        off, lens = _native.scan_frames(data, False)
+        cpu_codes = infer_codes_from_buffer(data, off, lens, record_type)
+        dev = torch.as_tensor(np.ascontiguousarray(data)).cuda()
+        gpu_codes = g.infer_codes_device(
+            dev, torch.as_tensor(off).cuda(), torch.as_tensor(lens).cuda(),
+            record_type)
+        return cpu_codes, gpu_codes
+
+    def test_infer_example_matches_host(self):
+        cpu_codes, gpu_codes = self._codes_both(make_batch(777, seed=11),
+                                                "Example")
+        assert cpu_codes == gpu_codes
+
+    def test_infer_sequence_matches_host(self):
+        rng = np.random.default_rng(12)
+        schema = stf.StructType([
+            stf.StructField("ctx", stf.LongType(), True),
+            stf.StructField("rag", stf.ArrayType(stf.ArrayType(stf.FloatType())), True),
+        ])
+        n = 300
+        ctx = list(rng.integers(0, 10, n))
+        rag = [[list(rng.random(rng.integers(0, 4)).astype(float))
+                for _ in range(rng.integers(0, 3))] for _ in range(n)]
+        cols = [column_from_values(ctx, stf.LongType(), True, "ctx"),
+                column_from_values(rag, schema[1].dataType, True, "rag")]
+        batch = RecordBatch(schema, cols, n)
+        cpu_codes, gpu_codes = self._codes_both(batch, "SequenceExample")
+        assert cpu_codes == gpu_codes
+
+    def test_read_with_gpu_inference(self, tmp_sandbox):
+        out = str(tmp_sandbox / "gi")
+        data = {"a": np.arange(100, dtype=np.int64),
+                "b": [[1.5, 2.5]] * 100,
+                "c": [f"s{i}" for i in range(100)]}
+        stf.write_tfrecord(data, out, engine="gpu")
+        df = stf.read_tfrecord(out, engine="gpu").sort("a")  # schema inferred
+        rows = df.collect()
+        assert rows[3]["a"] == 3 and rows[3]["b"] == [1.5, 2.5]
+        assert rows[3]["c"] == "s3"
+
+
 class TestGpuEndToEnd:
     def test_write_read_files_gpu_engine(self, tmp_sandbox):
         out = str(tmp_sandbox / "g")

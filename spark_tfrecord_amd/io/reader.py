@@ -25,6 +25,7 @@ from ..infer import (
     schema_from_codes,
 )
 from ..schema import LongType, StringType, StructField, StructType
+from ..utils import IOMetrics, StageTimer
 from .. import _native
 from . import paths as P
 
@@ -104,13 +105,16 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     part_cols = _partition_schema(files, base_dir) if base_dir else []
     eng = engine_mod.resolve_engine(engine)
 
+    metrics = IOMetrics("read")
     if schema is None:
-        schema = (byte_array_schema() if record_type == "ByteArray"
-                  else infer_schema_of_paths(files, record_type, eng))
+        with StageTimer(metrics, "infer_schema"):
+            schema = (byte_array_schema() if record_type == "ByteArray"
+                      else infer_schema_of_paths(files, record_type, eng))
     data_schema = StructType([f for f in schema.fields if f.name not in part_cols])
 
     tables = []
     for fpath in files:
+        metrics.add(files=1, nbytes=os.path.getsize(fpath))
         if eng == "gpu":
             from ..engine import gpu as gpu_engine
             if P.codec_from_path(fpath) is None:
@@ -149,4 +153,6 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     for c in part_cols:
         at = table.schema.field(c).type
         full_schema.add(c, LongType() if pa.types.is_integer(at) else StringType())
+    metrics.add(rows=table.num_rows)
+    metrics.finish()
     return DataFrame(table, full_schema)

@@ -21,6 +21,7 @@ from ..arrow_interop import schema_from_arrow, schema_to_arrow, table_to_batch
 from ..columnar import RecordBatch
 from ..engine import cpu as cpu_engine
 from ..schema import BinaryType, StructType
+from ..utils import IOMetrics, StageTimer
 from . import paths as P
 
 __all__ = ["write_tfrecord", "normalize_input"]
@@ -67,7 +68,8 @@ def _partition_dir_value(v) -> str:
 
 def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
                       out_dir: str, codec: Optional[str], job_id: str,
-                      num_shards: int, shard_offset: int, eng: str):
+                      num_shards: int, shard_offset: int, eng: str,
+                      metrics: Optional[IOMetrics] = None):
     """Encode `table` into `num_shards` part files under out_dir."""
     R = table.num_rows
     bounds = np.linspace(0, R, num_shards + 1).astype(np.int64)
@@ -85,6 +87,8 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
         payload = P.compress_bytes(raw, codec)
         fname = P.part_file_name(shard_offset + s, codec, job_id)
         P.write_file_atomic(payload, os.path.join(out_dir, fname))
+        if metrics is not None:
+            metrics.add(rows=hi - lo, nbytes=len(payload), files=1)
 
 
 def write_tfrecord(data, path: str, record_type: str = "Example",
@@ -113,6 +117,7 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
         if not P.apply_save_mode(path, mode):
             return
     job_id = uuid.uuid4().hex[:12]
+    metrics = IOMetrics("write")
 
     if partition_by:
         for c in partition_by:
@@ -133,10 +138,11 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
                         for c, v in zip(partition_by, combo)))
             os.makedirs(sub_dir, exist_ok=True)
             _encode_and_write(sub, data_schema, record_type, sub_dir, codec, job_id,
-                              num_shards, shard_offset, eng)
+                              num_shards, shard_offset, eng, metrics)
     else:
         _encode_and_write(table, schema, record_type, path, codec, job_id,
-                          num_shards, shard_offset, eng)
+                          num_shards, shard_offset, eng, metrics)
 
     if write_success:
         P.write_success_marker(path)
+    metrics.finish()

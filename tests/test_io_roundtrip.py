@@ -256,3 +256,23 @@ class TestErrors:
             stf.read_tfrecord(out, schema=schema)
         with pytest.raises(RuntimeError):  # inference path also rejects it
             stf.read_tfrecord(out)
+
+
+class TestMetrics:
+    def test_metrics_reported(self, tmp_sandbox):
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+        from spark_tfrecord_amd.utils import last_metrics
+
+        out = str(tmp_sandbox / "metrics")
+        data = {"x": np.arange(500, dtype=np.int64)}
+        stf.write_tfrecord(data, out, engine="cpu")
+        m = last_metrics()
+        assert m is not None and m.op == "write"
+        assert m.rows == 500 and m.files == 1 and m.bytes > 0
+        assert m.rows_per_sec > 0
+        stf.read_tfrecord(out, engine="cpu")
+        m = last_metrics()
+        assert m.op == "read" and m.rows == 500
+        assert "infer_schema" in m.stages

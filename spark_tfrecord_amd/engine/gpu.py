@@ -657,17 +657,24 @@ def infer_codes_device(data: torch.Tensor, off: torch.Tensor,
     used = np.nonzero(tab[:, 0])[0]
     if len(used) == 0:
         return {}
-    # gather all name strings with ONE D2H copy of the covering byte range
+    # recover name strings: one D2H of the covering range when it is small,
+    # else one tiny D2H per distinct name (names are first occurrences and
+    # can be anywhere in the file)
     refs = tab[used, 1]
     offs = (refs >> 16).astype(np.int64)
     nlens = (refs & 0xFFFF).astype(np.int64)
     lo = int(offs.min())
     hi = int((offs + nlens).max())
-    blob = data[lo:hi].cpu().numpy().tobytes()
     codes: dict = {}
-    for o, ln, code in zip(offs, nlens, tab[used, 2]):
-        name = blob[o - lo:o - lo + ln].decode("utf-8")
-        codes[name] = max(codes.get(name, 0), int(code))
+    if hi - lo <= (4 << 20):
+        blob = data[lo:hi].cpu().numpy().tobytes()
+        for o, ln, code in zip(offs, nlens, tab[used, 2]):
+            name = blob[o - lo:o - lo + ln].decode("utf-8")
+            codes[name] = max(codes.get(name, 0), int(code))
+    else:
+        for o, ln, code in zip(offs, nlens, tab[used, 2]):
+            name = data[o:o + ln].cpu().numpy().tobytes().decode("utf-8")
+            codes[name] = max(codes.get(name, 0), int(code))
     return codes
 
 

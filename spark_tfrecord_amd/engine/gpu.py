@@ -628,6 +628,56 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
     return decode_device(data, off, lens, schema, record_type, verify_crc)
 
 
+def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,
+                             num_parts: int, record_type: str):
+    """Encode ALL rows once, then split into per-partition file images by
+    GATHERING framed records (TFRecord frames are concatenable, so a
+    partitioned write never re-serializes a row — the partition step is a
+    pure byte gather in HBM). Returns (image, [(part_code, lo, hi), ...])
+    where image[lo:hi] is partition part_code's complete file image.
+
+    The reference delegates this grouping to Spark's FileFormatWriter sort
+    (SURVEY.md §3.3); here the grouping is a device argsort of partition
+    codes + one gather kernel pass."""
+    check_native()
+    R = batch.num_rows
+    device = batch.columns[0].presence.device if batch.columns else torch.device("cuda")
+    blob = torch.frombuffer(bytearray(schema_blob(batch.schema)),
+                            dtype=torch.uint8).to(device)
+    col_dicts = [_col_ptrs(c) for c in batch.columns]
+    cols_dev = torch.empty(_native.gpu_devcols_bytes(), dtype=torch.uint8,
+                           device=device)
+    psize = torch.empty(R, dtype=torch.int64, device=device)
+    _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
+                             FMT[record_type], R, psize.data_ptr(), _stream())
+    frame_off = excl_sum(psize)
+    total = int(frame_off[-1].item())
+    file = torch.empty(total, dtype=torch.uint8, device=device)
+    err = torch.zeros(1, dtype=torch.int32, device=device)
+    _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
+                             FMT[record_type], 0, R, frame_off.data_ptr(),
+                             file.data_ptr(), err.data_ptr(), _stream())
+    codes = torch.as_tensor(np.ascontiguousarray(part_codes, np.int64),
+                            device=device)
+    order = torch.argsort(codes, stable=True)
+    sizes = psize[order].contiguous()
+    dst_off = excl_sum(sizes)
+    src_off = frame_off[:-1][order].contiguous()
+    out = torch.empty(total, dtype=torch.uint8, device=device)
+    _native.gpu_gather_payloads(file.data_ptr(), src_off.data_ptr(),
+                                sizes.data_ptr(), dst_off.data_ptr(), R,
+                                out.data_ptr(), _stream())
+    # partition boundaries in the ordered space
+    counts = torch.bincount(codes, minlength=num_parts)
+    row_bound = torch.nn.functional.pad(torch.cumsum(counts, 0), (1, 0))
+    byte_bound = dst_off[row_bound].cpu().numpy()
+    if int(err.item()) != 0:
+        raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+    ranges = [(p, int(byte_bound[p]), int(byte_bound[p + 1]))
+              for p in range(num_parts) if byte_bound[p + 1] > byte_bound[p]]
+    return out, ranges
+
+
 # ---------------------------------------------------------------------------
 # Schema inference on device: hash-table lattice kernel (SURVEY.md §2b).
 # ---------------------------------------------------------------------------

@@ -66,6 +66,39 @@ def _partition_dir_value(v) -> str:
     return str(v)
 
 
+def _factorize_partitions(table: pa.Table, partition_by: Sequence[str]):
+    """Vectorized grouping: per-row partition code (np.int64) + the list of
+    value tuples, combos[code] = (v1, v2, ...). Nulls map to the Hive
+    default-partition value like Spark's dynamic partition insert."""
+    import pyarrow.compute as pc
+
+    idx_cols = []
+    val_lists = []
+    for c in partition_by:
+        arr = table.column(c).combine_chunks()
+        if isinstance(arr, pa.ChunkedArray):
+            arr = arr.chunk(0) if arr.num_chunks else pa.array([], arr.type)
+        de = arr.dictionary_encode()
+        vals = de.dictionary.to_pylist()
+        idx = pc.fill_null(de.indices, len(vals)).to_numpy(zero_copy_only=False)
+        idx_cols.append(idx.astype(np.int64))
+        val_lists.append(vals + [None])
+    code = idx_cols[0].copy()
+    for i in range(1, len(idx_cols)):
+        code = code * len(val_lists[i]) + idx_cols[i]
+    uniq, inv = np.unique(code, return_inverse=True)
+    combos = []
+    for u in uniq:
+        rem = int(u)
+        rev = []
+        for vals in reversed(val_lists[1:]):
+            rem, k = divmod(rem, len(vals))
+            rev.append(vals[k])
+        rev.append(val_lists[0][rem])
+        combos.append(tuple(reversed(rev)))
+    return inv.astype(np.int64), combos
+
+
 def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
                       out_dir: str, codec: Optional[str], job_id: str,
                       num_shards: int, shard_offset: int, eng: str,
@@ -125,20 +158,53 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
                 raise KeyError(f"partition column '{c}' not found")
         data_cols = [c for c in table.column_names if c not in set(partition_by)]
         data_schema = StructType([f for f in schema.fields if f.name in set(data_cols)])
-        # group rows by partition tuple
-        part_vals = [table.column(c).to_pylist() for c in partition_by]
-        groups: Dict[tuple, List[int]] = {}
-        for i, combo in enumerate(zip(*part_vals)):
-            groups.setdefault(combo, []).append(i)
+        codes, combos = _factorize_partitions(table, partition_by)
         stripped = table.select(data_cols)
-        for combo, idxs in sorted(groups.items(), key=lambda kv: str(kv[0])):
-            sub = stripped.take(pa.array(idxs, type=pa.int64()))
-            sub_dir = os.path.join(
-                path, *(f"{c}={_partition_dir_value(v)}"
-                        for c, v in zip(partition_by, combo)))
-            os.makedirs(sub_dir, exist_ok=True)
-            _encode_and_write(sub, data_schema, record_type, sub_dir, codec, job_id,
-                              num_shards, shard_offset, eng, metrics)
+
+        def part_dir(combo) -> str:
+            d = os.path.join(path, *(f"{c}={_partition_dir_value(v)}"
+                                     for c, v in zip(partition_by, combo)))
+            os.makedirs(d, exist_ok=True)
+            return d
+
+        if (eng == "gpu" and record_type != "ByteArray" and num_shards == 1
+                and table.num_rows > 0):
+            # MI355X path: encode every row ONCE on the GPU, then split into
+            # per-partition file images by gathering framed records in HBM
+            # (frames are concatenable — no re-serialization per partition)
+            from ..engine import gpu as gpu_engine
+
+            batch = gpu_engine.batch_to_device(
+                table_to_batch(stripped, data_schema))
+            img, ranges = gpu_engine.encode_partitions_device(
+                batch, codes, len(combos), record_type)
+            for p, lo, hi in ranges:
+                sub_dir = part_dir(combos[p])
+                fname = P.part_file_name(shard_offset, codec, job_id)
+                fpath = os.path.join(sub_dir, fname)
+                if codec is None:
+                    # atomic like write_file_atomic: land bytes in a temp
+                    # file, then rename over the final name
+                    tmp = fpath + ".inprogress"
+                    gpu_engine.device_to_file(img[lo:hi], tmp)
+                    os.replace(tmp, fpath)
+                else:
+                    payload = P.compress_bytes(
+                        gpu_engine.device_to_bytes(img[lo:hi]), codec)
+                    P.write_file_atomic(payload, fpath)
+                if metrics is not None:
+                    metrics.add(nbytes=hi - lo, files=1)
+            if metrics is not None:
+                metrics.add(rows=table.num_rows)
+        else:
+            for p, combo in enumerate(combos):
+                idxs = np.nonzero(codes == p)[0]
+                if idxs.size == 0:
+                    continue
+                sub = stripped.take(pa.array(idxs, type=pa.int64()))
+                _encode_and_write(sub, data_schema, record_type, part_dir(combo),
+                                  codec, job_id, num_shards, shard_offset, eng,
+                                  metrics)
     else:
         _encode_and_write(table, schema, record_type, path, codec, job_id,
                           num_shards, shard_offset, eng, metrics)

@@ -33,15 +33,20 @@ def needs_rebuild() -> bool:
     return any(p.stat().st_mtime > out_mtime for p in SOURCES + HEADERS)
 
 
-def build(force: bool = False, verbose: bool = True) -> Path:
+def build(force: bool = False, verbose: bool = True,
+          sanitize: bool = False) -> Path:
     if not force and not needs_rebuild():
         return OUT
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     objdir = ROOT / "build"
     objdir.mkdir(exist_ok=True)
+    # host-side ASan/UBSan build for the C++ codec + IO pool (SURVEY.md §5
+    # race-detection/sanitizer row); device code is unaffected
+    san_flags = (["-Xarch_host", "-fsanitize=address,undefined",
+                  "-fno-omit-frame-pointer"] if sanitize else [])
     objs = []
     for src in SOURCES:
-        obj = objdir / (src.stem + ".o")
+        obj = objdir / (src.stem + (".san.o" if sanitize else ".o"))
         cmd = [
             "hipcc",
             "--offload-arch=gfx950",
@@ -51,6 +56,7 @@ def build(force: bool = False, verbose: bool = True) -> Path:
             "-DTFREC_WITH_HIP",
             "-Xarch_host",
             "-msse4.2",
+            *san_flags,
             "-x",
             "hip",
             *_include_flags(),
@@ -63,7 +69,9 @@ def build(force: bool = False, verbose: bool = True) -> Path:
             print("[build_native]", " ".join(cmd), flush=True)
         subprocess.run(cmd, check=True, cwd=ROOT)
         objs.append(obj)
-    link = ["hipcc", "-shared", "-fPIC", *map(str, objs), "-o", str(OUT)]
+    link = ["hipcc", "-shared", "-fPIC", *(
+        ["-fsanitize=address,undefined", "-shared-libasan"] if sanitize else []),
+        *map(str, objs), "-o", str(OUT)]
     if verbose:
         print("[build_native]", " ".join(link), flush=True)
     subprocess.run(link, check=True, cwd=ROOT)
@@ -71,5 +79,5 @@ def build(force: bool = False, verbose: bool = True) -> Path:
 
 
 if __name__ == "__main__":
-    build(force="--force" in sys.argv)
+    build(force="--force" in sys.argv, sanitize="--sanitize" in sys.argv)
     print(f"built {OUT}")

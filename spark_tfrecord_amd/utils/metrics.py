@@ -94,6 +94,41 @@ class StageTimer:
         return self
 
     def __exit__(self, *exc):
+        dt = time.perf_counter() - self.t
         if self.m is not None:
-            self.m.stage(self.name, time.perf_counter() - self.t)
+            self.m.stage(self.name, dt)
+        _trace_event(self.name, self.t, dt)
         return False
+
+
+# ---------------------------------------------------------------------------
+# Chrome-trace event log: TFREC_TRACE=/path/trace.json records every stage as
+# a chrome://tracing / Perfetto "X" event. Kernel-level timelines come from
+# rocprofv3; this covers the host orchestration above them.
+# ---------------------------------------------------------------------------
+
+_trace_path = None
+_trace_lock = threading.Lock()
+_trace_t0 = time.perf_counter()
+
+
+def _trace_event(name: str, start: float, dur: float):
+    import os
+
+    global _trace_path
+    path = os.environ.get("TFREC_TRACE")
+    if not path:
+        return
+    ev = {"name": name, "ph": "X", "pid": os.getpid(),
+          "tid": threading.get_ident() % 1_000_000,
+          "ts": (start - _trace_t0) * 1e6, "dur": dur * 1e6}
+    import json
+    with _trace_lock:
+        first = _trace_path != path
+        if first:
+            _trace_path = path
+            with open(path, "w") as f:
+                f.write("[\n" + json.dumps(ev))
+        else:
+            with open(path, "a") as f:
+                f.write(",\n" + json.dumps(ev))

@@ -276,3 +276,18 @@ class TestMetrics:
         m = last_metrics()
         assert m.op == "read" and m.rows == 500
         assert "infer_schema" in m.stages
+
+    def test_chrome_trace_written(self, tmp_sandbox, monkeypatch):
+        import json as _json
+
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+
+        tr = str(tmp_sandbox / "trace.json")
+        monkeypatch.setenv("TFREC_TRACE", tr)
+        out = str(tmp_sandbox / "tr")
+        stf.write_tfrecord({"x": np.arange(10, dtype=np.int64)}, out, engine="cpu")
+        stf.read_tfrecord(out, engine="cpu")
+        events = _json.loads(open(tr).read() + "]")
+        assert any(e["name"] == "infer_schema" for e in events)

@@ -83,6 +83,9 @@ def _factorize_partitions(table: pa.Table, partition_by: Sequence[str]):
         idx = pc.fill_null(de.indices, len(vals)).to_numpy(zero_copy_only=False)
         idx_cols.append(idx.astype(np.int64))
         val_lists.append(vals + [None])
+    if len(idx_cols) == 1:
+        # dictionary codes are already dense [0, n_values): no unique() pass
+        return idx_cols[0], [(v,) for v in val_lists[0]]
     code = idx_cols[0].copy()
     for i in range(1, len(idx_cols)):
         code = code * len(val_lists[i]) + idx_cols[i]

@@ -61,6 +61,7 @@ def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
 
 
 _CHUNK = 64 << 20  # 64 MiB: overlap file IO with PCIe copies chunkwise
+_READ_SLICE = 48 << 20  # H2D slice size for the pipelined read (tests shrink it)
 
 
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:
@@ -603,7 +604,7 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
         off, lens = scan_frames_device(data)
         return decode_device(data, off, lens, schema, record_type, verify_crc)
     data = torch.empty(n, dtype=torch.uint8, device=device)
-    span = 48 << 20
+    span = _READ_SLICE
     S = max(1, (n + span - 1) // span)
     max_cand = n // 16 + 64
     cand_pos = torch.empty(max_cand, dtype=torch.int64, device=device)

@@ -162,6 +162,20 @@ class TestPipelinedFileIO:
             path, batch.schema, "Example", verify_crc=True))
         assert_batches_equal(batch, out)
 
+    def test_pipelined_read_multi_slice(self, tmp_path, monkeypatch):
+        # force several H2D slices so the guard-band handoff between the
+        # sliced frame-candidate launches is exercised on a small file
+        g = _gpu_engine()
+        monkeypatch.setattr(g, "_READ_SLICE", 64 << 10)  # 64 KiB slices
+        batch = make_batch(5000, seed=10)
+        path = str(tmp_path / "s.tfrecord")
+        with open(path, "wb") as f:
+            f.write(cpu_engine.encode_batch(batch, "Example"))
+        assert os.path.getsize(path) > 4 * (64 << 10)
+        out = g.batch_to_host(g.read_file_to_batch_pipelined(
+            path, batch.schema, "Example", verify_crc=True))
+        assert_batches_equal(batch, out)
+
     def test_roundtrip_overwrites_shorter_file(self, tmp_path):
         # in-place mmap reuse must truncate correctly when the file shrinks
         g = _gpu_engine()

@@ -16,6 +16,7 @@ import torch  # noqa: F401  (intentional import order)
 from .api import DataFrame, DataFrameReader, DataFrameWriter, TFRecordSession, session
 from .io.reader import read_tfrecord
 from .io.writer import write_tfrecord
+from .io.validate import validate_tfrecord
 from .schema import (
     ArrayType,
     BinaryType,
@@ -35,7 +36,7 @@ __version__ = "0.1.0"
 
 __all__ = [
     "session", "DataFrame", "DataFrameReader", "DataFrameWriter",
-    "TFRecordSession", "read_tfrecord", "write_tfrecord",
+    "TFRecordSession", "read_tfrecord", "write_tfrecord", "validate_tfrecord",
     "DataType", "NullType", "IntegerType", "LongType", "FloatType",
     "DoubleType", "DecimalType", "StringType", "BinaryType", "ArrayType",
     "StructField", "StructType",

@@ -132,7 +132,16 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
                    partition_by: Optional[Sequence[str]] = None,
                    schema: Optional[StructType] = None, num_shards: int = 1,
                    engine: str = "auto", write_success: bool = True,
-                   shard_offset: int = 0, _apply_mode: bool = True) -> None:
+                   shard_offset: int = 0, job_id: Optional[str] = None,
+                   _apply_mode: bool = True) -> None:
+    """Write DataFrame-shaped `data` as TFRecord files under `path`.
+
+    Fault tolerance: every part file lands via write-to-temp + atomic
+    rename, and `_SUCCESS` is written last — a crashed job never leaves a
+    partial file visible. Pass an explicit `job_id` to make retries
+    idempotent: the rerun produces identical part-file names and atomically
+    replaces whatever the failed attempt left behind (the analog of Spark's
+    task-commit protocol the reference relies on)."""
     if record_type not in RECORD_TYPES:
         raise ValueError(
             f"Unsupported recordType {record_type!r} (expected one of {RECORD_TYPES})")
@@ -152,7 +161,8 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
     if _apply_mode:
         if not P.apply_save_mode(path, mode):
             return
-    job_id = uuid.uuid4().hex[:12]
+    if job_id is None:
+        job_id = uuid.uuid4().hex[:12]
     metrics = IOMetrics("write")
 
     if partition_by:

@@ -212,10 +212,10 @@ def bench_gzip_bytearray(args):
     table = pa.table({"byteArray": pa.array(payloads, type=pa.large_binary())})
     d = _workdir("gzba", rank if world > 1 else 0)
     out = os.path.join(d, "t")
-    # 8 shards, like one Spark task per file in the reference: gzip is
+    # many shards, like one Spark task per file in the reference: gzip is
     # sequential PER file, so shards are the parallelism axis on read
     stf.write_tfrecord(table, out, record_type="ByteArray", codec="gzip",
-                       mode="overwrite", engine="cpu", num_shards=8)
+                       mode="overwrite", engine="cpu", num_shards=32)
     nbytes = sum(os.path.getsize(os.path.join(out, f))
                  for f in os.listdir(out) if not f.startswith("_"))
     eng = "gpu" if use_cuda else "cpu"

@@ -244,104 +244,110 @@ __device__ inline u32 crc8b_reg(u64 w, const uint32_t (*tab)[256]) {
 // length bound check (random u64 <= N with p ~ N/2^64) culls everything but
 // real frame heads before any CRC work.
 //
-// Candidates are compacted through an LDS staging buffer with ONE global
-// atomicAdd per block iteration: naive per-candidate atomics on a single
-// global counter serialize at the owning L2 bank (~1M contended atomics cost
-// ~11 ms, measured — profiles/r01_kernel_stats.txt); the aggregated scheme
-// reduces that to ~8k block-level atomics.
+// Deterministic two-pass compaction: pass 1 (COUNT) writes per-block
+// candidate counts, Python prefix-sums them, pass 2 (EMIT) re-evaluates and
+// writes candidates at exact offsets — output is SORTED BY POSITION by
+// construction (block b covers chunks [c_lo + b*blockDim, ...); thread t its
+// t-th chunk; positions ascend within a chunk), so no device sort and no
+// contended atomics at all. (The first design used per-candidate atomics on
+// one global counter — ~11 ms of L2 serialization at 1M records, see
+// profiles/r01_kernel_stats.txt — then an LDS-aggregated append that still
+// needed a 0.9 ms device sort; this version needs neither.) The count pass
+// runs per arrived slice under the H2D DMA of the pipelined reader.
 constexpr int kPosPerLane = 16;
-constexpr int kLdsCand = 2048;  // 32 KiB LDS; realistic max ≈ blockDim*16/24
 
-// Evaluates candidate positions in [pos_start, pos_end) only — the sliced
-// read pipeline scans each arrived slice while the next one is still in
-// flight on the DMA stream (a position's 32-byte load window must have
-// landed: callers keep a 32-byte guard band at the slice boundary).
-__global__ void frame_candidate_kernel(const u8* __restrict__ data, i64 N,
+// Evaluates the (up to) 16 candidate positions of chunk c, calling
+// f(pos, len) for each one that passes the length cull + header CRC.
+template <typename F>
+__device__ inline void eval_frame_chunk(const u8* __restrict__ data,
+                                        const u64* __restrict__ wdata, i64 N,
+                                        i64 c, i64 pos_start, i64 pos_end,
+                                        const uint32_t (*tab)[256], F&& f) {
+  // p0 = 16*c is 8-byte aligned, so every index below is compile-time
+  // constant after unrolling (runtime-indexed w[] would spill to scratch).
+  i64 p0 = c * kPosPerLane;
+  i64 w0 = p0 >> 3;
+  u64 w[4];
+  if (p0 + 32 <= N) {  // interior chunk: full-word loads
+#pragma unroll
+    for (int j = 0; j < 4; ++j) w[j] = wdata[w0 + j];
+  } else {  // file tail: assemble partial words byte-wise, zero-padded
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      u64 v = 0;
+      i64 boff = (w0 + j) * 8;
+      for (int b = 0; b < 8 && boff + b < N; ++b)
+        v |= (u64)data[boff + b] << (8 * b);
+      w[j] = v;
+    }
+  }
+#pragma unroll
+  for (int k = 0; k < kPosPerLane; ++k) {
+    i64 i = p0 + k;
+    if (i + 16 > N || i >= pos_end) break;
+    if (i < pos_start) continue;
+    constexpr int _ppl = kPosPerLane;
+    static_assert(_ppl == 16, "index math below assumes 16 positions");
+    const int wi = k >> 3;
+    const int sh = (k & 7) * 8;
+    u64 len = (sh == 0) ? w[wi] : (w[wi] >> sh) | (w[wi + 1] << (64 - sh));
+    if (len > (u64)(N - i) - 16) continue;
+    const int wi2 = (k + 8) >> 3;
+    const int sh2 = sh;  // (k+8) & 7 == k & 7
+    u64 hi = (sh2 == 0) ? w[wi2]
+                        : (w[wi2] >> sh2) | (w[wi2 + 1] << (64 - sh2));
+    u32 want = (u32)hi;
+    if (mask_crc(crc8b_reg(len, tab)) != want) continue;
+    f(i, (i64)len);
+  }
+}
+
+// EMIT=false: block_data[block_base + b] = candidate count of block b.
+// EMIT=true:  block_data holds the exclusive scan of those counts (read);
+//             candidates land at cand_pos/cand_len[block offset + rank].
+template <bool EMIT>
+__global__ void frame_scan_pass_kernel(const u8* __restrict__ data, i64 N,
                                        i64 pos_start, i64 pos_end,
+                                       i64 block_base,
+                                       i64* __restrict__ block_data,
                                        i64* __restrict__ cand_pos,
-                                       i64* __restrict__ cand_len,
-                                       unsigned long long* __restrict__ count,
-                                       i64 max_cand) {
+                                       i64* __restrict__ cand_len) {
   __shared__ uint32_t tab[8][256];
-  __shared__ i64 lpos[kLdsCand];
-  __shared__ i64 llen[kLdsCand];
-  __shared__ unsigned int lcount;
-  __shared__ unsigned long long gbase;
+  __shared__ unsigned int cnt[256];
   stage_crc_tables(tab);
-  if (threadIdx.x == 0) lcount = 0;
-  __syncthreads();
   const u64* wdata = reinterpret_cast<const u64*>(data);  // data is 8B-aligned
   i64 c_lo = pos_start / kPosPerLane;
-  i64 nchunks = (pos_end + kPosPerLane - 1) / kPosPerLane - c_lo;
-  // Uniform per-block trip count so every thread reaches the __syncthreads
-  // flush barrier even when its own chunk index runs past nchunks.
-  for (i64 cbase = (i64)blockIdx.x * blockDim.x; cbase < nchunks;
-       cbase += (i64)gridDim.x * blockDim.x) {
-    i64 c = c_lo + cbase + threadIdx.x;
-    if (cbase + threadIdx.x < nchunks) {
-      // p0 = 16*c is 8-byte aligned, so every index below is compile-time
-      // constant after unrolling (runtime-indexed w[] would spill to scratch).
-      i64 p0 = c * kPosPerLane;
-      i64 w0 = p0 >> 3;
-      u64 w[4];
-      if (p0 + 32 <= N) {  // interior chunk: full-word loads
-#pragma unroll
-        for (int j = 0; j < 4; ++j) w[j] = wdata[w0 + j];
-      } else {  // file tail: assemble partial words byte-wise, zero-padded
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          u64 v = 0;
-          i64 boff = (w0 + j) * 8;
-          for (int b = 0; b < 8 && boff + b < N; ++b)
-            v |= (u64)data[boff + b] << (8 * b);
-          w[j] = v;
-        }
-      }
-#pragma unroll
-      for (int k = 0; k < kPosPerLane; ++k) {
-        i64 i = p0 + k;
-        if (i + 16 > N || i >= pos_end) break;
-        if (i < pos_start) continue;
-        constexpr int _ppl = kPosPerLane;
-        static_assert(_ppl == 16, "index math below assumes 16 positions");
-        const int wi = k >> 3;
-        const int sh = (k & 7) * 8;
-        u64 len = (sh == 0) ? w[wi] : (w[wi] >> sh) | (w[wi + 1] << (64 - sh));
-        if (len > (u64)(N - i) - 16) continue;
-        const int wi2 = (k + 8) >> 3;
-        const int sh2 = sh;  // (k+8) & 7 == k & 7
-        u64 hi = (sh2 == 0) ? w[wi2]
-                            : (w[wi2] >> sh2) | (w[wi2 + 1] << (64 - sh2));
-        u32 want = (u32)hi;
-        if (mask_crc(crc8b_reg(len, tab)) != want) continue;
-        unsigned int s = atomicAdd(&lcount, 1u);
-        if (s < kLdsCand) {
-          lpos[s] = i;
-          llen[s] = (i64)len;
-        } else {  // adversarial overflow: spill straight to global (slow, rare)
-          unsigned long long slot = atomicAdd(count, 1ull);
-          if ((i64)slot < max_cand) {
-            cand_pos[slot] = i;
-            cand_len[slot] = (i64)len;
-          }
-        }
-      }
+  i64 c_hi = (pos_end + kPosPerLane - 1) / kPosPerLane;
+  i64 c = c_lo + (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  unsigned int my = 0;
+  if (c < c_hi)
+    eval_frame_chunk(data, wdata, N, c, pos_start, pos_end, tab,
+                     [&](i64, i64) { ++my; });
+  cnt[threadIdx.x] = my;
+  __syncthreads();
+  if (!EMIT) {
+    for (int off = blockDim.x / 2; off; off >>= 1) {  // tree reduction
+      if ((int)threadIdx.x < off) cnt[threadIdx.x] += cnt[threadIdx.x + off];
+      __syncthreads();
     }
+    if (threadIdx.x == 0) block_data[block_base + blockIdx.x] = (i64)cnt[0];
+    return;
+  }
+  // block-local inclusive scan of per-thread counts (Hillis-Steele in LDS)
+  for (int off = 1; off < (int)blockDim.x; off <<= 1) {
+    unsigned int v = (int)threadIdx.x >= off ? cnt[threadIdx.x - off] : 0;
     __syncthreads();
-    unsigned int n = lcount < kLdsCand ? lcount : kLdsCand;
-    if (threadIdx.x == 0 && n) gbase = atomicAdd(count, (unsigned long long)n);
-    __syncthreads();
-    for (unsigned int s = threadIdx.x; s < n; s += blockDim.x) {
-      i64 slot = (i64)gbase + s;
-      if (slot < max_cand) {
-        cand_pos[slot] = lpos[s];
-        cand_len[slot] = llen[s];
-      }
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) lcount = 0;
+    cnt[threadIdx.x] += v;
     __syncthreads();
   }
+  i64 slot = block_data[block_base + blockIdx.x] + (i64)(cnt[threadIdx.x] - my);
+  if (c < c_hi)
+    eval_frame_chunk(data, wdata, N, c, pos_start, pos_end, tab,
+                     [&](i64 i, i64 len) {
+                       cand_pos[slot] = i;
+                       cand_len[slot] = len;
+                       ++slot;
+                     });
 }
 
 // ---------------------------------------------------------------------------
@@ -709,15 +715,37 @@ void gpu_frame_bytes(uintptr_t src, uintptr_t elem_off, uintptr_t frame_off, i64
   HIP_CHECK(hipGetLastError());
 }
 
-void gpu_frame_scan(uintptr_t data, i64 N, i64 pos_start, i64 pos_end,
-                    uintptr_t cand_pos, uintptr_t cand_len,
-                    uintptr_t count, i64 max_cand, uintptr_t stream) {
-  hipLaunchKernelGGL(frame_candidate_kernel,
-                     dim3(grid_for((pos_end - pos_start) / kPosPerLane + 1)),
-                     dim3(kBlock), 0,
-                     (hipStream_t)stream, (const u8*)data, N, pos_start, pos_end,
-                     (i64*)cand_pos, (i64*)cand_len,
-                     (unsigned long long*)count, max_cand);
+// Number of blocks a [pos_start, pos_end) range occupies in the frame-scan
+// pass geometry (Python sizes the per-block count buffer with this).
+i64 gpu_frame_scan_blocks(i64 pos_start, i64 pos_end) {
+  i64 c_lo = pos_start / kPosPerLane;
+  i64 c_hi = (pos_end + kPosPerLane - 1) / kPosPerLane;
+  i64 n = c_hi - c_lo;
+  return (n + kBlock - 1) / kBlock;
+}
+
+void gpu_frame_scan_count(uintptr_t data, i64 N, i64 pos_start, i64 pos_end,
+                          i64 block_base, uintptr_t block_counts,
+                          uintptr_t stream) {
+  i64 B = gpu_frame_scan_blocks(pos_start, pos_end);
+  if (B <= 0) return;
+  hipLaunchKernelGGL(frame_scan_pass_kernel<false>, dim3((uint32_t)B),
+                     dim3(kBlock), 0, (hipStream_t)stream, (const u8*)data, N,
+                     pos_start, pos_end, block_base, (i64*)block_counts,
+                     nullptr, nullptr);
+  HIP_CHECK(hipGetLastError());
+}
+
+void gpu_frame_scan_emit(uintptr_t data, i64 N, i64 pos_start, i64 pos_end,
+                         i64 block_base, uintptr_t block_off,
+                         uintptr_t cand_pos, uintptr_t cand_len,
+                         uintptr_t stream) {
+  i64 B = gpu_frame_scan_blocks(pos_start, pos_end);
+  if (B <= 0) return;
+  hipLaunchKernelGGL(frame_scan_pass_kernel<true>, dim3((uint32_t)B),
+                     dim3(kBlock), 0, (hipStream_t)stream, (const u8*)data, N,
+                     pos_start, pos_end, block_base, (i64*)block_off,
+                     (i64*)cand_pos, (i64*)cand_len);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -900,7 +928,9 @@ void register_gpu(py::module_& m) {
   m.def("gpu_size_records", &gpu_size_records);
   m.def("gpu_emit_records", &gpu_emit_records);
   m.def("gpu_frame_bytes", &gpu_frame_bytes);
-  m.def("gpu_frame_scan", &gpu_frame_scan);
+  m.def("gpu_frame_scan_blocks", &gpu_frame_scan_blocks);
+  m.def("gpu_frame_scan_count", &gpu_frame_scan_count);
+  m.def("gpu_frame_scan_emit", &gpu_frame_scan_emit);
   m.def("gpu_gather_payloads", &gpu_gather_payloads);
   m.def("gpu_infer_codes", &gpu_infer_codes);
   m.def("gpu_scan_temp_bytes", &gpu_scan_temp_bytes);

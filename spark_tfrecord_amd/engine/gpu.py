@@ -242,37 +242,39 @@ def scan_frames_device(data: torch.Tensor):
     """Device file image -> (payload_off, payload_len) device tensors.
 
     Every byte position is CRC-tested as a candidate frame head in parallel
-    (frame_candidate_kernel); the sorted candidates must chain exactly
-    (pos[k+1] == pos[k] + 16 + len[k]) — a false positive (p=2^-32/byte) or a
-    torn file breaks the chain and raises."""
+    (two-pass count/emit — candidates come out position-sorted by
+    construction, no device sort); the chain pos[k+1] == pos[k]+16+len[k]
+    must hold exactly — a false positive (p=2^-32/byte) or a torn file
+    breaks it and raises."""
     check_native()
     N = data.numel()
     device = data.device
     if N == 0:
         z = torch.zeros(0, dtype=torch.int64, device=device)
         return z, z.clone()
-    max_cand = N // 16 + 64
-    cand_pos = torch.empty(max_cand, dtype=torch.int64, device=device)
-    cand_len = torch.empty(max_cand, dtype=torch.int64, device=device)
-    count = torch.zeros(1, dtype=torch.int64, device=device)
-    _native.gpu_frame_scan(data.data_ptr(), N, 0, N, cand_pos.data_ptr(),
-                           cand_len.data_ptr(), count.data_ptr(), max_cand,
-                           _stream())
-    return _chain_candidates(data, cand_pos, cand_len, count, N)
+    B = _native.gpu_frame_scan_blocks(0, N)
+    block_counts = torch.empty(B, dtype=torch.int64, device=device)
+    _native.gpu_frame_scan_count(data.data_ptr(), N, 0, N, 0,
+                                 block_counts.data_ptr(), _stream())
+    return _emit_and_chain(data, [(0, N)], block_counts, N)
 
 
-def _chain_candidates(data, cand_pos, cand_len, count, N):
-    """Sort frame-head candidates and validate the chain
-    pos[k+1] == pos[k] + 16 + len[k]; returns (payload_off, payload_len)."""
+def _emit_and_chain(data, ranges, block_counts, N):
+    """Prefix-sum the per-block candidate counts, emit the (sorted)
+    candidates at exact offsets, then validate the frame chain."""
     device = data.device
-    max_cand = cand_pos.numel()
-    C = int(count.item())
+    block_off = excl_sum(block_counts)
+    C = int(block_off[-1].item())
     if C == 0:
         raise RuntimeError("corrupt TFRecord: no valid frame header found")
-    if C > max_cand:
-        raise RuntimeError("corrupt TFRecord: implausible candidate count")
-    pos, order = torch.sort(cand_pos[:C])
-    lens = cand_len[:C][order]
+    pos = torch.empty(C, dtype=torch.int64, device=device)
+    lens = torch.empty(C, dtype=torch.int64, device=device)
+    base = 0
+    for s, e in ranges:
+        _native.gpu_frame_scan_emit(data.data_ptr(), N, s, e, base,
+                                    block_off.data_ptr(), pos.data_ptr(),
+                                    lens.data_ptr(), _stream())
+        base += _native.gpu_frame_scan_blocks(s, e)
     # chain check fused to ONE device scalar -> one sync
     expect_next = pos + 16 + lens
     ok = bool(((pos[0] == 0) & (expect_next[-1] == N)
@@ -606,13 +608,22 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
     data = torch.empty(n, dtype=torch.uint8, device=device)
     span = _READ_SLICE
     S = max(1, (n + span - 1) // span)
-    max_cand = n // 16 + 64
-    cand_pos = torch.empty(max_cand, dtype=torch.int64, device=device)
-    cand_len = torch.empty(max_cand, dtype=torch.int64, device=device)
-    count = torch.zeros(1, dtype=torch.int64, device=device)
+    # pre-size the per-block count buffer over the slice ranges
+    ranges = []
+    scanned = 0
+    for s in range(S):
+        b1 = min(n, (s + 1) * span)
+        scan_end = b1 - 32 if s < S - 1 else n
+        if scan_end > scanned:
+            ranges.append((scanned, scan_end))
+            scanned = scan_end
+    nblocks = [int(_native.gpu_frame_scan_blocks(a, b)) for a, b in ranges]
+    block_counts = torch.empty(max(sum(nblocks), 1), dtype=torch.int64,
+                               device=device)
     main = torch.cuda.current_stream()
     streams = _dma_streams()
-    scanned = 0
+    ri = 0
+    base = 0
     for s in range(S):
         b0, b1 = s * span, min(n, (s + 1) * span)
         st = streams[s % len(streams)]
@@ -620,12 +631,13 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
                                st.cuda_stream)
         main.wait_stream(st)
         scan_end = b1 - 32 if s < S - 1 else n
-        if scan_end > scanned:
-            _native.gpu_frame_scan(data.data_ptr(), n, scanned, scan_end,
-                                   cand_pos.data_ptr(), cand_len.data_ptr(),
-                                   count.data_ptr(), max_cand, _stream())
-            scanned = scan_end
-    off, lens = _chain_candidates(data, cand_pos, cand_len, count, n)
+        if ri < len(ranges) and ranges[ri][1] == scan_end:
+            _native.gpu_frame_scan_count(data.data_ptr(), n, ranges[ri][0],
+                                         ranges[ri][1], base,
+                                         block_counts.data_ptr(), _stream())
+            base += nblocks[ri]
+            ri += 1
+    off, lens = _emit_and_chain(data, ranges, block_counts, n)
     return decode_device(data, off, lens, schema, record_type, verify_crc)
 
 

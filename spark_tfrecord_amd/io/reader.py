@@ -122,8 +122,9 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     def _needs_host_bytes(fpath: str) -> bool:
         return eng != "gpu" or P.codec_from_path(fpath) is not None
 
-    pool = ThreadPoolExecutor(max_workers=8)
-    window = 8
+    workers = min(32, (os.cpu_count() or 8))
+    pool = ThreadPoolExecutor(max_workers=workers)
+    window = 2 * workers
     futures: dict = {}
 
     def _blob(i: int):

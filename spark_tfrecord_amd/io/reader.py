@@ -132,38 +132,60 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
             futures[i] = pool.submit(_load_file, files[i])
         return futures.pop(i).result()
 
+    # GPU path: consecutive uncompressed files are decoded as ONE pipeline
+    # (images concatenated in HBM — frames are concatenable — scanned and
+    # decoded in a single pass, rows split per file afterwards). Compressed
+    # files and the CPU engine go file-by-file through host bytes.
+    _GROUP_BYTES = 4 << 30
+
+    def _append_with_parts(t, fpaths, row_counts):
+        pvs = [P.partition_values_of(f, base_dir) for f in fpaths]
+        for c in part_cols:
+            vals = np.repeat([pv.get(c) for pv in pvs], row_counts).tolist()
+            t = t.append_column(c, _partition_col_array(vals, t.num_rows))
+        tables.append(t)
+
     tables = []
     try:
-        for i, fpath in enumerate(files):
+        i = 0
+        while i < len(files):
+            fpath = files[i]
+            if eng == "gpu" and not _needs_host_bytes(fpath):
+                from ..engine import gpu as gpu_engine
+
+                group = []
+                gbytes = 0
+                while (i < len(files) and not _needs_host_bytes(files[i])
+                       and gbytes < _GROUP_BYTES):
+                    group.append(files[i])
+                    sz = os.path.getsize(files[i])
+                    gbytes += sz
+                    metrics.add(files=1, nbytes=sz)
+                    i += 1
+                batch, row_counts = gpu_engine.read_files_to_batch(
+                    group, data_schema, record_type, verify_crc=verify_crc)
+                if batch.num_rows == 0 and not part_cols:
+                    continue
+                t = batch_to_table(gpu_engine.batch_to_host(batch))
+                _append_with_parts(t, group, row_counts)
+                continue
             for j in range(i, min(i + window, len(files))):
                 if _needs_host_bytes(files[j]) and j not in futures:
                     futures[j] = pool.submit(_load_file, files[j])
             metrics.add(files=1, nbytes=os.path.getsize(fpath))
+            data = _blob(i)
+            i += 1
+            if data.size == 0:
+                continue
             if eng == "gpu":
                 from ..engine import gpu as gpu_engine
-                if P.codec_from_path(fpath) is None:
-                    if os.path.getsize(fpath) == 0:
-                        continue
-                    batch = gpu_engine.batch_to_host(gpu_engine.read_file_to_batch(
-                        fpath, data_schema, record_type, verify_crc=verify_crc))
-                else:
-                    data = _blob(i)
-                    if data.size == 0:
-                        continue
-                    batch = gpu_engine.decode_buffer_to_cpu(
-                        data, data_schema, record_type, verify_crc=verify_crc)
+                batch = gpu_engine.decode_buffer_to_cpu(
+                    data, data_schema, record_type, verify_crc=verify_crc)
             else:
-                data = _blob(i)
-                if data.size == 0:
-                    continue
                 batch = cpu_engine.decode_buffer(data, data_schema, record_type,
                                                  verify_crc=verify_crc)
             t = batch_to_table(batch)
-            pv = P.partition_values_of(fpath, base_dir)
-            for c in part_cols:
-                vals = [pv.get(c)] * t.num_rows
-                t = t.append_column(c, _partition_col_array(vals, t.num_rows))
-            tables.append(t)
+            _append_with_parts(t, [fpath], [t.num_rows])
     finally:
         pool.shutdown(wait=False)
     if not tables:

@@ -12,6 +12,7 @@ torch ops on the same stream.
 
 from __future__ import annotations
 
+import os
 from typing import List, Optional
 
 import numpy as np
@@ -61,7 +62,9 @@ def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
 
 
 _CHUNK = 64 << 20  # 64 MiB: overlap file IO with PCIe copies chunkwise
-_READ_SLICE = 48 << 20  # H2D slice size for the pipelined read (tests shrink it)
+# engine knobs: overridable via env for tuning
+_READ_SLICE = int(os.environ.get("TFREC_READ_SLICE", 48 << 20))
+_WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 3))
 
 
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:
@@ -528,7 +531,8 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
 
 
 def write_batch_to_file(batch: RecordBatch, path: str,
-                        record_type: str = "Example", slices: int = 3) -> int:
+                        record_type: str = "Example",
+                        slices: Optional[int] = None) -> int:
     """Device batch -> framed TFRecord file, with the emit kernel sliced over
     record ranges so the D2H DMA of slice k streams to the file's mapped
     pages while slice k+1 is still being emitted (SURVEY.md §7 step 3:
@@ -549,7 +553,7 @@ def write_batch_to_file(batch: RecordBatch, path: str,
     _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], R, psize.data_ptr(), _stream())
     frame_off = excl_sum(psize)
-    S = max(1, min(slices, R))
+    S = max(1, min(slices if slices is not None else _WRITE_SLICES, R))
     ridx = [R * s // S for s in range(S + 1)]
     bounds = frame_off[ridx].cpu()  # one sync: slice byte bounds + total
     total = int(bounds[-1])

@@ -64,7 +64,7 @@ def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
 _CHUNK = 64 << 20  # 64 MiB: overlap file IO with PCIe copies chunkwise
 # engine knobs: overridable via env for tuning
 _READ_SLICE = int(os.environ.get("TFREC_READ_SLICE", 48 << 20))
-_WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 3))
+_WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 5))
 
 
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:

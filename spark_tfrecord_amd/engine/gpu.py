@@ -34,7 +34,14 @@ def _stream() -> int:
 def _dev(x, dtype, device="cuda") -> torch.Tensor:
     if isinstance(x, torch.Tensor):
         return x.to(device=device, dtype=dtype).contiguous()
-    return torch.as_tensor(np.ascontiguousarray(x), dtype=dtype).to(device)
+    arr = np.ascontiguousarray(x)
+    # arrow buffers are read-only and torch warns on wrapping them; the
+    # wrapped tensor is only ever read (the .to() below copies to device)
+    import warnings
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore", UserWarning)
+        return torch.as_tensor(arr, dtype=dtype).to(device)
 
 
 def check_native():

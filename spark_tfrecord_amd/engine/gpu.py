@@ -369,14 +369,22 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     # FieldStat[R][F] as int64 [R,F,6]: pos,len,nvals,nbytes,nlists,(kind|err)
     stats = torch.empty((R, F, 6), dtype=torch.int64, device=device)
     err = torch.zeros(1, dtype=torch.int32, device=device)
-    # frame CRC verification is fused into the structure scan (record bytes
-    # are read once, while L2-hot)
-    crc_err = torch.full((1,), -1, dtype=torch.int64, device=device) \
-        if verify_crc else None
+    # frame CRC verification runs CONCURRENTLY with the structure scan on a
+    # side stream — the two kernels read the same (independent) bytes, so
+    # the CRC pass hides entirely under the longer parse
+    main = torch.cuda.current_stream()
+    crc_err = None
+    crc_stream = None
+    if verify_crc:
+        crc_err = torch.full((1,), -1, dtype=torch.int64, device=device)
+        crc_stream = _dma_streams()[1]
+        crc_stream.wait_stream(main)
+        _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(), lens.data_ptr(),
+                               R, crc_err.data_ptr(), crc_stream.cuda_stream)
+        crc_err.record_stream(crc_stream)
     _native.gpu_scan_records(data.data_ptr(), off.data_ptr(), lens.data_ptr(), R,
                              FMT[record_type], blob.data_ptr(), F,
-                             stats.data_ptr(), err.data_ptr(),
-                             crc_err.data_ptr() if verify_crc else 0, _stream())
+                             stats.data_ptr(), err.data_ptr(), 0, _stream())
 
     # Per-field exclusive prefix sums ([F, R+1]): rocprim strided scans read
     # the stat column straight out of the [R, F, 6] buffer — no transpose
@@ -395,6 +403,7 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     totals = torch.stack([val_base[:, -1], byte_base[:, -1], list_base[:, -1]])
     totals_h = totals.cpu()  # one sync for all allocations
     if verify_crc:
+        main.wait_stream(crc_stream)
         bad = int(crc_err.item())
         if bad != -1:
             raise RuntimeError(

@@ -324,6 +324,25 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
         np.cumsum(lens, out=row_off[1:])
         return WireColumn(kind, False, presence, row_off, vals)
 
+    # Fast path: 2-D numeric nested list (SequenceExample FeatureList of
+    # Int64List/FloatList). Offsets compose: values-per-row cumulative is
+    # off2[off1] — no python loop over 1M ragged rows.
+    if seq and kind != KIND_BYTES and \
+            (pa.types.is_list(arr.type) or pa.types.is_large_list(arr.type)):
+        inner = arr.values
+        if (null_count == 0 and arr.offset == 0 and inner.offset == 0
+                and inner.null_count == 0
+                and (pa.types.is_list(inner.type)
+                     or pa.types.is_large_list(inner.type))):
+            off1 = _np_offsets(arr).astype(np.int64)    # [R+1] sublists/row
+            off2 = _np_offsets(inner).astype(np.int64)  # [L+1] values/sublist
+            if off1[0] == 0 and off2[0] == 0:
+                child = inner.values.to_numpy(zero_copy_only=False)
+                L = int(off1[-1])
+                vals = numeric_cast(child[: int(off2[L])])
+                return WireColumn(kind, True, presence, off2[off1].copy(), vals,
+                                  None, off1.copy(), off2[: L + 1].copy())
+
     # General fallback: python objects
     return column_from_values(arr.to_pylist(), dt, nullable, name)
 

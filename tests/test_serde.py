@@ -266,3 +266,41 @@ class TestLattice:
         assert wire_kind_of(stf.LongType()) == KIND_INT64
         assert wire_kind_of(stf.DoubleType()) == KIND_FLOAT
         assert wire_kind_of(stf.ArrayType(stf.StringType())) == KIND_BYTES
+
+
+class TestArrowNestedFastPath:
+    def test_nested_list_fast_path_matches_fallback(self):
+        import numpy as np
+        import pyarrow as pa
+
+        from spark_tfrecord_amd.arrow_interop import arrow_to_wire
+        from spark_tfrecord_amd.columnar import column_from_values
+
+        rng = np.random.default_rng(0)
+        rows = [[[float(v) for v in rng.random(int(rng.integers(0, 4)))]
+                 for _ in range(int(rng.integers(0, 3)))] for _ in range(200)]
+        dt = stf.ArrayType(stf.ArrayType(stf.FloatType()))
+        arr = pa.array(rows, type=pa.large_list(pa.large_list(pa.float32())))
+        fast = arrow_to_wire(arr, dt, True, "rag")
+        slow = column_from_values(rows, dt, True, "rag")
+        np.testing.assert_array_equal(fast.presence, slow.presence)
+        np.testing.assert_array_equal(fast.row_off, slow.row_off)
+        np.testing.assert_array_equal(np.asarray(fast.list_off),
+                                      np.asarray(slow.list_off))
+        np.testing.assert_array_equal(np.asarray(fast.sub_off),
+                                      np.asarray(slow.sub_off))
+        np.testing.assert_array_equal(np.asarray(fast.values),
+                                      np.asarray(slow.values))
+
+    def test_nested_with_nulls_uses_fallback(self):
+        import numpy as np
+        import pyarrow as pa
+
+        from spark_tfrecord_amd.arrow_interop import arrow_to_wire
+
+        rows = [[[1.0]], None, [[2.0, 3.0], []]]
+        dt = stf.ArrayType(stf.ArrayType(stf.FloatType()))
+        arr = pa.array(rows, type=pa.large_list(pa.large_list(pa.float32())))
+        col = arrow_to_wire(arr, dt, True, "rag")
+        np.testing.assert_array_equal(col.presence, [1, 0, 1])
+        np.testing.assert_array_equal(np.asarray(col.row_off), [0, 1, 1, 3])

@@ -144,22 +144,33 @@ def _owner_of(partition_key: Tuple, world: int) -> int:
     return int.from_bytes(h, "little") % world
 
 
-def _all_to_all_bytes(send: List[bytes], dev: torch.device) -> List[bytes]:
-    """Exchange one byte blob per peer. nccl: true all-to-all over xGMI;
-    gloo (CPU tests): emulated with all_gather per destination."""
+def _blob_len(b) -> int:
+    return b.numel() if isinstance(b, torch.Tensor) else len(b)
+
+
+def _all_to_all_blobs(send: List, dev: torch.device) -> List:
+    """Exchange one blob per peer. Blobs may be bytes or torch.uint8
+    tensors. nccl: a true all-to-all of DEVICE tensors over xGMI — encoded
+    partition bytes move GPU-to-GPU and come back as device tensors (the
+    receiving rank DMAs them straight into its partition file, no host
+    round-trip). gloo (CPU tests): emulated with all_gather per
+    destination, bytes in/out."""
     rank, world = _world()
     if dist.get_backend() == "nccl":
-        in_sizes = torch.tensor([len(b) for b in send], dtype=torch.int64,
+        in_sizes = torch.tensor([_blob_len(b) for b in send], dtype=torch.int64,
                                 device=dev)
         out_sizes = torch.empty(world, dtype=torch.int64, device=dev)
         dist.all_to_all_single(out_sizes, in_sizes)
-        send_t = [torch.frombuffer(bytearray(b), dtype=torch.uint8).to(dev)
-                  if len(b) else torch.zeros(0, dtype=torch.uint8, device=dev)
+        send_t = [b.to(dev).contiguous() if isinstance(b, torch.Tensor) else
+                  (torch.frombuffer(bytearray(b), dtype=torch.uint8).to(dev)
+                   if len(b) else torch.zeros(0, dtype=torch.uint8, device=dev))
                   for b in send]
         recv_t = [torch.empty(int(s), dtype=torch.uint8, device=dev)
                   for s in out_sizes]
         dist.all_to_all(recv_t, send_t)
-        return [bytes(t.cpu().numpy().tobytes()) for t in recv_t]
+        return recv_t
+    send = [bytes(b.cpu().numpy().tobytes()) if isinstance(b, torch.Tensor)
+            else b for b in send]
     # gloo fallback: for each destination d, gather everyone's blob-to-d
     out: List[bytes] = [b""] * world
     for d in range(world):
@@ -237,29 +248,56 @@ def write_tfrecord_distributed(data, path: str, record_type: str = "Example",
         P.write_file_atomic(P.compress_bytes(raw, codec),
                             os.path.join(path, fname))
     else:
+        import numpy as np
+
+        from ..io.writer import _factorize_partitions
+
         data_cols = [c for c in table.column_names if c not in set(partition_by)]
         data_schema = StructType([f for f in schema.fields
                                   if f.name in set(data_cols)])
-        part_vals = [table.column(c).to_pylist() for c in partition_by]
-        groups: Dict[tuple, List[int]] = {}
-        for i, combo in enumerate(zip(*part_vals)):
-            groups.setdefault(combo, []).append(i)
+        codes, combos = _factorize_partitions(table, partition_by)
         stripped = table.select(data_cols)
-        # encode each partition's rows once, destined for its owner rank
-        send_parts: List[List[Tuple[tuple, bytes]]] = [[] for _ in range(world)]
-        for combo, idxs in sorted(groups.items(), key=lambda kv: str(kv[0])):
-            sub = stripped.take(pa.array(idxs, type=pa.int64()))
-            blob = encode(sub, data_schema)
-            send_parts[_owner_of(combo, world)].append((combo, blob))
-        # header: repr(list[(combo, length)]) exchanged as objects (tiny)
+        # per-partition wire blobs: on the GPU every row is encoded ONCE and
+        # the partition split is a framed-record gather in HBM — the blobs
+        # stay device tensors all the way through the RCCL all-to-all
+        part_blobs: Dict[int, object] = {}
+        if eng == "gpu" and record_type != "ByteArray" and table.num_rows > 0:
+            from ..engine import gpu as gpu_engine
+
+            batch = gpu_engine.batch_to_device(
+                table_to_batch(stripped, data_schema))
+            img, ranges = gpu_engine.encode_partitions_device(
+                batch, codes, len(combos), record_type)
+            for p, lo, hi in ranges:
+                part_blobs[p] = img[lo:hi]
+        else:
+            for p in range(len(combos)):
+                idxs = np.nonzero(codes == p)[0]
+                if idxs.size == 0:
+                    continue
+                sub = stripped.take(pa.array(idxs, type=pa.int64()))
+                part_blobs[p] = encode(sub, data_schema)
+        send_parts: List[List[Tuple[tuple, object]]] = [[] for _ in range(world)]
+        for p in sorted(part_blobs):
+            send_parts[_owner_of(combos[p], world)].append(
+                (combos[p], part_blobs[p]))
+        # header: list[(combo, length)] exchanged as objects (tiny)
         headers: List[List[Tuple[tuple, int]]] = [
-            [(c, len(b)) for c, b in parts] for parts in send_parts]
+            [(c, _blob_len(b)) for c, b in parts] for parts in send_parts]
         gathered_headers: List[List[List[Tuple[tuple, int]]]] = [None] * world
         dist.all_gather_object(gathered_headers, headers)
-        send_blobs = [b"".join(b for _, b in parts) for parts in send_parts]
-        recv_blobs = _all_to_all_bytes(send_blobs, dev)
-        # stitch: per incoming rank, split by its header for me
-        mine: Dict[tuple, List[bytes]] = {}
+
+        def _concat(parts):
+            blobs = [b for _, b in parts]
+            if blobs and isinstance(blobs[0], torch.Tensor):
+                return (torch.cat(blobs) if len(blobs) > 1
+                        else blobs[0])
+            return b"".join(blobs)
+
+        recv_blobs = _all_to_all_blobs([_concat(p) for p in send_parts], dev)
+        # stitch: per incoming rank, split by its header for me (slicing a
+        # device tensor is a view — no copy)
+        mine: Dict[tuple, List] = {}
         for src in range(world):
             hdr = gathered_headers[src][rank]
             blob = recv_blobs[src]
@@ -271,10 +309,24 @@ def write_tfrecord_distributed(data, path: str, record_type: str = "Example",
             sub_dir = os.path.join(
                 path, *(f"{c}={_partition_dir_value(v)}"
                         for c, v in zip(partition_by, combo)))
-            raw = b"".join(blobs)  # TFRecord frames concatenate losslessly
+            os.makedirs(sub_dir, exist_ok=True)
             fname = P.part_file_name(rank, codec, job_id)
-            P.write_file_atomic(P.compress_bytes(raw, codec),
-                                os.path.join(sub_dir, fname))
+            fpath = os.path.join(sub_dir, fname)
+            if blobs and isinstance(blobs[0], torch.Tensor):
+                from ..engine import gpu as gpu_engine
+
+                raw_t = torch.cat(blobs) if len(blobs) > 1 else blobs[0]
+                if codec is None:
+                    tmp = fpath + ".inprogress"
+                    gpu_engine.device_to_file(raw_t.contiguous(), tmp)
+                    os.replace(tmp, fpath)
+                else:
+                    P.write_file_atomic(P.compress_bytes(
+                        gpu_engine.device_to_bytes(raw_t.contiguous()), codec),
+                        fpath)
+            else:
+                raw = b"".join(blobs)  # TFRecord frames concatenate losslessly
+                P.write_file_atomic(P.compress_bytes(raw, codec), fpath)
 
     dist.barrier()
     if rank == 0:

@@ -830,14 +830,31 @@ def batch_to_device(batch: RecordBatch, device="cuda") -> RecordBatch:
 
 
 def batch_to_host(batch: RecordBatch) -> RecordBatch:
+    """Device batch -> host numpy batch. All D2H copies go into PINNED
+    host tensors (torch's caching host allocator reuses them) issued
+    async back-to-back with ONE sync — pageable `.cpu()` per tensor runs at
+    ~8 GB/s with a hidden staging copy; this runs at link speed. The numpy
+    arrays are zero-copy views of the pinned tensors (kept alive by the
+    view's base)."""
+    pending = []
+
     def h(t):
-        return t.cpu().numpy() if isinstance(t, torch.Tensor) else t
+        if not isinstance(t, torch.Tensor):
+            return t
+        if not t.is_cuda:
+            return t.numpy()
+        host = torch.empty_like(t, device="cpu", pin_memory=True)
+        host.copy_(t, non_blocking=True)
+        pending.append(True)
+        return host.numpy()
 
     cols = [WireColumn(c.kind, c.is_seq, h(c.presence), h(c.row_off), h(c.values),
                        h(c.elem_off) if c.elem_off is not None else None,
                        h(c.list_off) if c.list_off is not None else None,
                        h(c.sub_off) if c.sub_off is not None else None)
             for c in batch.columns]
+    if pending:
+        torch.cuda.current_stream().synchronize()
     return RecordBatch(batch.schema, cols, batch.num_rows)
 
 

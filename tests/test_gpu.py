@@ -239,6 +239,33 @@ class TestGpuInference:
         assert rows[3]["c"] == "s3"
 
 
+class TestGpuRandomized:
+    """Randomized numerics sweep: many random batches, GPU kernels vs the
+    plain host codec (the CPU reference implementation of the same ops)."""
+
+    @pytest.mark.parametrize("seed", range(8))
+    def test_random_batches_encode_decode(self, seed):
+        g = _gpu_engine()
+        rng = np.random.default_rng(1000 + seed)
+        n = int(rng.integers(1, 400))
+        batch = make_batch(n, seed=seed)
+        cpu_img = cpu_engine.encode_batch(batch, "Example")
+        assert g.encode_batch_from_cpu(batch, "Example") == cpu_img
+        data = np.frombuffer(cpu_img, np.uint8)
+        assert_batches_equal(
+            cpu_engine.decode_buffer(data, batch.schema, "Example"),
+            g.decode_buffer_to_cpu(data, batch.schema, "Example"))
+
+    def test_single_record_file(self, tmp_path):
+        g = _gpu_engine()
+        batch = make_batch(1, seed=42)
+        path = str(tmp_path / "one.tfrecord")
+        g.write_batch_to_file(g.batch_to_device(batch), path, "Example")
+        out = g.batch_to_host(g.read_file_to_batch_pipelined(
+            path, batch.schema, "Example"))
+        assert_batches_equal(batch, out)
+
+
 class TestGpuEndToEnd:
     def test_write_read_files_gpu_engine(self, tmp_sandbox):
         out = str(tmp_sandbox / "g")

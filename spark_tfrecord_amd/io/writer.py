@@ -115,7 +115,8 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
             continue
         chunk = table.slice(lo, hi - lo)
         batch = table_to_batch(chunk, schema)
-        if eng == "gpu":
+        # device DevCols holds at most 64 fields; wider schemas encode on host
+        if eng == "gpu" and len(batch.columns) <= 64:
             from ..engine import gpu as gpu_engine
             raw = gpu_engine.encode_batch_from_cpu(batch, record_type)
         else:
@@ -181,7 +182,7 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
             return d
 
         if (eng == "gpu" and record_type != "ByteArray" and num_shards == 1
-                and table.num_rows > 0):
+                and table.num_rows > 0 and len(data_schema.fields) <= 64):
             # MI355X path: encode every row ONCE on the GPU, then split into
             # per-partition file images by gathering framed records in HBM
             # (frames are concatenable — no re-serialization per partition)

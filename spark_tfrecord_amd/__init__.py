@@ -17,6 +17,7 @@ from .api import DataFrame, DataFrameReader, DataFrameWriter, TFRecordSession, s
 from .io.reader import read_tfrecord
 from .io.writer import write_tfrecord
 from .io.validate import validate_tfrecord
+from . import torch_data
 from .schema import (
     ArrayType,
     BinaryType,

@@ -266,6 +266,29 @@ class TestGpuRandomized:
         assert_batches_equal(batch, out)
 
 
+class TestTorchDatasetGpu:
+    def test_device_resident_stream(self, tmp_sandbox):
+        import torch
+
+        from spark_tfrecord_amd.torch_data import TFRecordIterableDataset
+
+        out = str(tmp_sandbox / "tds")
+        rng = np.random.default_rng(0)
+        stf.write_tfrecord({"uid": np.arange(3000, dtype=np.int64),
+                            "v": rng.random(3000).astype(np.float32)},
+                           out, engine="gpu", num_shards=3)
+        ds = TFRecordIterableDataset(out, batch_rows=512, engine="gpu")
+        total = 0
+        seen = []
+        for b in ds:
+            assert b["uid"].is_cuda and b["v"].is_cuda  # stays on device
+            total += int(b["_num_rows"])
+            seen.append(b["uid"].cpu())
+        assert total == 3000
+        got = np.sort(torch.cat(seen).numpy())
+        np.testing.assert_array_equal(got, np.arange(3000))
+
+
 class TestGpuEndToEnd:
     def test_write_read_files_gpu_engine(self, tmp_sandbox):
         out = str(tmp_sandbox / "g")

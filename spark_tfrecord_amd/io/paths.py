@@ -64,14 +64,69 @@ def codec_from_path(path: str) -> Optional[str]:
     return _EXT_TO_CODEC.get(ext)
 
 
+# Gzip output is written as a SINGLE standard gzip member whose deflate
+# stream carries Z_FULL_FLUSH sync points every _GZ_SEGMENT of input (each
+# flush resets the dictionary and byte-aligns with the 00 00 FF FF empty
+# stored block, pigz-style). Any gzip reader — TensorFlow included — decodes
+# the file normally; OUR reader finds the sync markers and inflates the
+# segments in parallel, verifying the member's CRC32 trailer (false-positive
+# markers fall back to sequential inflate). Matches the reference's
+# isSplitable=false model: gzip files still read as whole files, just on
+# more than one core.
+_GZ_SEGMENT = 4 << 20
+_GZ_MARK = b"\x00\x00\xff\xff"
+
+
 def compress_bytes(data: bytes, codec: Optional[str]) -> bytes:
     if codec is None:
         return data
     if codec == "gzip":
-        return gzip.compress(data, compresslevel=6)
+        c = zlib.compressobj(6, zlib.DEFLATED, 16 + 15)  # gzip wrapper
+        out = []
+        for pos in range(0, len(data), _GZ_SEGMENT):
+            out.append(c.compress(data[pos:pos + _GZ_SEGMENT]))
+            if pos + _GZ_SEGMENT < len(data):
+                out.append(c.flush(zlib.Z_FULL_FLUSH))
+        out.append(c.flush())
+        return b"".join(out)
     if codec == "deflate":
         return zlib.compress(data, 6)
     raise ValueError(codec)
+
+
+def _gunzip_parallel(raw: bytes) -> Optional[bytes]:
+    """Parallel segmented inflate of a single-member gzip blob written with
+    full-flush sync points. Returns None when the fast path does not apply
+    (multi-member, extra header fields, marker false positive, CRC mismatch)."""
+    if len(raw) < 20 or raw[:3] != b"\x1f\x8b\x08" or raw[3] != 0:
+        return None  # flags would shift the header; let zlib handle it
+    body = raw[10:-8]
+    crc_want = int.from_bytes(raw[-8:-4], "little")
+    isize = int.from_bytes(raw[-4:], "little")
+    cuts = []
+    p = body.find(_GZ_MARK)
+    while p != -1:
+        cuts.append(p + 4)
+        p = body.find(_GZ_MARK, p + 4)
+    if not cuts:
+        return None
+    bounds = [0] + cuts + [len(body)]
+    segs = [body[bounds[i]:bounds[i + 1]] for i in range(len(bounds) - 1)]
+
+    def inflate(seg):
+        d = zlib.decompressobj(-15)
+        return d.decompress(seg) + d.flush()
+
+    from concurrent.futures import ThreadPoolExecutor
+    try:
+        with ThreadPoolExecutor(max_workers=min(16, os.cpu_count() or 4)) as ex:
+            parts = list(ex.map(inflate, segs))
+    except zlib.error:
+        return None
+    out = b"".join(parts)
+    if len(out) % (1 << 32) != isize or (zlib.crc32(out) & 0xFFFFFFFF) != crc_want:
+        return None
+    return out
 
 
 def decompress_file(path: str) -> bytes:
@@ -81,7 +136,8 @@ def decompress_file(path: str) -> bytes:
     if codec is None:
         return raw
     if codec == "gzip":
-        return gzip.decompress(raw)
+        out = _gunzip_parallel(raw)
+        return out if out is not None else gzip.decompress(raw)
     if codec == "deflate":
         return zlib.decompress(raw)
     raise ValueError(codec)

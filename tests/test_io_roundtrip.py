@@ -328,3 +328,45 @@ class TestValidate:
                            mode="append")
         assert sorted(__import__("os").listdir(out)) == first  # replaced, not duplicated
         assert stf.read_tfrecord(out, engine="cpu").count() == 50
+
+
+class TestParallelGzip:
+    def test_segmented_gzip_interop_and_parallel_path(self):
+        import gzip as _gz
+        import numpy as np
+
+        from spark_tfrecord_amd.io import paths as P
+
+        rng = np.random.default_rng(0)
+        # ~20 MB, mixed compressibility, crosses several 4 MB segments
+        data = (rng.integers(0, 8, 10_000_000, dtype=np.uint8).tobytes()
+                + b"A" * 10_000_000)
+        raw = P.compress_bytes(data, "gzip")
+        # standard gzip readers decode the full-flush stream unchanged
+        assert _gz.decompress(raw) == data
+        # our parallel segmented path reproduces it too
+        out = P._gunzip_parallel(raw)
+        assert out is not None and out == data
+
+    def test_foreign_gzip_falls_back(self):
+        import gzip as _gz
+
+        from spark_tfrecord_amd.io import paths as P
+        import os, tempfile
+
+        data = b"hello world " * 1000
+        fd, path = tempfile.mkstemp(suffix=".tfrecord.gz")
+        with os.fdopen(fd, "wb") as f:
+            f.write(_gz.compress(data))
+        try:
+            assert P.decompress_file(path) == data
+        finally:
+            os.unlink(path)
+
+    def test_empty_gzip(self):
+        import gzip as _gz
+
+        from spark_tfrecord_amd.io import paths as P
+
+        raw = P.compress_bytes(b"", "gzip")
+        assert _gz.decompress(raw) == b""

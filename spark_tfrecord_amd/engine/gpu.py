@@ -266,12 +266,60 @@ def scan_frames_device(data: torch.Tensor):
     block_counts = torch.empty(B, dtype=torch.int64, device=device)
     _native.gpu_frame_scan_count(data.data_ptr(), N, 0, N, 0,
                                  block_counts.data_ptr(), _stream())
-    return _emit_and_chain(data, [(0, N)], block_counts, N)
+    off, lens, _ = _emit_and_chain(data, [(0, N)], block_counts, N)
+    return off, lens
+
+
+def _prescan_prefix(data, pre_ranges, block_counts, nblocks, N, arrived,
+                    schema, record_type):
+    """Emit the candidates of the ALREADY-ARRIVED slice prefix and launch
+    the structure scan for every record that lies entirely within `arrived`
+    bytes — on the main stream, which at this point is NOT yet ordered
+    after the in-flight tail slices, so this work overlaps their DMA.
+    Returns (stats1[K,F,6], err1, K) for decode_device, or None."""
+    device = data.device
+    nb_pre = sum(nblocks[:len(pre_ranges)])
+    if nb_pre == 0:
+        return None
+    block_off = excl_sum(block_counts[:nb_pre])
+    C = int(block_off[-1].item())  # syncs main: prefix counts only
+    if C == 0:
+        return None
+    pos = torch.empty(C, dtype=torch.int64, device=device)
+    lens = torch.empty(C, dtype=torch.int64, device=device)
+    base = 0
+    for s, e in pre_ranges:
+        _native.gpu_frame_scan_emit(data.data_ptr(), N, s, e, base,
+                                    block_off.data_ptr(), pos.data_ptr(),
+                                    lens.data_ptr(), _stream())
+        base += _native.gpu_frame_scan_blocks(s, e)
+    # records fully inside the arrived prefix (candidates are sorted)
+    K = int(torch.searchsorted(pos + 16 + lens, torch.tensor(
+        arrived + 1, device=device)).item())
+    if K == 0:
+        return None
+    fields = wire_fields(schema)
+    F = len(fields)
+    if F == 0:
+        return None
+    blob = torch.frombuffer(bytearray(schema_blob(schema)),
+                            dtype=torch.uint8).to(device)
+    stats1 = torch.empty((K, F, 6), dtype=torch.int64, device=device)
+    err1 = torch.zeros(1, dtype=torch.int32, device=device)
+    off1 = (pos[:K] + 12).contiguous()
+    lens1 = lens[:K].contiguous()
+    _native.gpu_scan_records(data.data_ptr(), off1.data_ptr(),
+                             lens1.data_ptr(), K, FMT[record_type],
+                             blob.data_ptr(), F, stats1.data_ptr(),
+                             err1.data_ptr(), 0, _stream())
+    return stats1, err1, K
 
 
 def _emit_and_chain(data, ranges, block_counts, N):
     """Prefix-sum the per-block candidate counts, emit the (sorted)
-    candidates at exact offsets, then validate the frame chain."""
+    candidates at exact offsets, then validate the frame chain. Returns
+    (payload_off, payload_len, clean) — clean=False when a false-positive
+    candidate had to be stitched out on host."""
     device = data.device
     block_off = excl_sum(block_counts)
     C = int(block_off[-1].item())
@@ -309,7 +357,7 @@ def _emit_and_chain(data, ranges, block_counts, N):
         sel = torch.as_tensor(_np.asarray(keep, _np.int64), device=device)
         pos = pos[sel]
         lens = lens[sel]
-    return pos + 12, lens
+    return pos + 12, lens, ok
 
 
 # ---------------------------------------------------------------------------
@@ -332,8 +380,10 @@ def crc_verify_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor)
 
 def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                   schema: StructType, record_type: str,
-                  verify_crc: bool = True) -> RecordBatch:
-    """Decode framed records already resident in HBM into device wire-form."""
+                  verify_crc: bool = True, prescan=None) -> RecordBatch:
+    """Decode framed records already resident in HBM into device wire-form.
+    `prescan` = (stats1, err1, K) from _prescan_prefix: rows [0, K) were
+    already structure-scanned while the tail of the file was in DMA flight."""
     check_native()
     device = data.device
     R = off.numel()
@@ -382,9 +432,22 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
         _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(), lens.data_ptr(),
                                R, crc_err.data_ptr(), crc_stream.cuda_stream)
         crc_err.record_stream(crc_stream)
-    _native.gpu_scan_records(data.data_ptr(), off.data_ptr(), lens.data_ptr(), R,
-                             FMT[record_type], blob.data_ptr(), F,
-                             stats.data_ptr(), err.data_ptr(), 0, _stream())
+    r0 = 0
+    err1 = None
+    if prescan is not None:
+        stats1, err1, K = prescan
+        if K <= R and stats1.shape[1] == F:
+            stats[:K].copy_(stats1)
+            r0 = K
+        else:
+            err1 = None
+    if R > r0:
+        stride = F * 6 * 8
+        _native.gpu_scan_records(data.data_ptr() , off.data_ptr() + r0 * 8,
+                                 lens.data_ptr() + r0 * 8, R - r0,
+                                 FMT[record_type], blob.data_ptr(), F,
+                                 stats.data_ptr() + r0 * stride,
+                                 err.data_ptr(), 0, _stream())
 
     # Per-field exclusive prefix sums ([F, R+1]): rocprim strided scans read
     # the stat column straight out of the [R, F, 6] buffer — no transpose
@@ -408,9 +471,10 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
         if bad != -1:
             raise RuntimeError(
                 f"corrupt TFRecord: bad CRC in record {bad - 1 if bad > 0 else bad}")
-    if int(err.item()) != 0:
+    rc = int(err.item()) or (int(err1.item()) if err1 is not None else 0)
+    if rc != 0:
         raise RuntimeError(f"TFRecord decode failed (native error "
-                           f"{int(err.item())}; kind mismatch or malformed record)")
+                           f"{rc}; kind mismatch or malformed record)")
 
     metas = []
     outs = []
@@ -642,23 +706,61 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
                                device=device)
     main = torch.cuda.current_stream()
     streams = _dma_streams()
-    ri = 0
-    base = 0
+    # issue ALL H2D slices up front on the side streams; per-slice events
+    # let the main stream gate each count pass on just its own slice
+    events = []
+    for st in streams:
+        st.wait_stream(main)  # `data` allocation ordering
     for s in range(S):
         b0, b1 = s * span, min(n, (s + 1) * span)
         st = streams[s % len(streams)]
         _native.gpu_memcpy_h2d(data.data_ptr() + b0, ptr + b0, b1 - b0,
                                st.cuda_stream)
-        main.wait_stream(st)
-        scan_end = b1 - 32 if s < S - 1 else n
-        if ri < len(ranges) and ranges[ri][1] == scan_end:
-            _native.gpu_frame_scan_count(data.data_ptr(), n, ranges[ri][0],
-                                         ranges[ri][1], base,
-                                         block_counts.data_ptr(), _stream())
-            base += nblocks[ri]
-            ri += 1
-    off, lens = _emit_and_chain(data, ranges, block_counts, n)
-    return decode_device(data, off, lens, schema, record_type, verify_crc)
+        ev = torch.cuda.Event()
+        ev.record(st)
+        events.append(ev)
+    data.record_stream(streams[0])
+    if len(streams) > 1:
+        data.record_stream(streams[1])
+
+    def launch_counts(ri: int, base: int):
+        _native.gpu_frame_scan_count(data.data_ptr(), n, ranges[ri][0],
+                                     ranges[ri][1], base,
+                                     block_counts.data_ptr(), _stream())
+
+    # stage point: once ~2/3 of the slices have landed, scan the records that
+    # are fully inside the arrived prefix WHILE the tail slices are still in
+    # DMA flight (hides most of the structure-scan under the H2D)
+    k0 = (2 * S) // 3 if S >= 3 and record_type != "ByteArray" else 0
+
+    ri = 0
+    base = 0
+
+    def consume_slices(upto: int):
+        nonlocal ri, base
+        for s in range(upto):
+            if events[s] is None:
+                continue
+            main.wait_event(events[s])
+            events[s] = None
+            scan_end = min(n, (s + 1) * span) - 32 if s < S - 1 else n
+            if ri < len(ranges) and ranges[ri][1] == scan_end:
+                launch_counts(ri, base)
+                base += nblocks[ri]
+                ri += 1
+
+    prescan = None
+    if k0:
+        consume_slices(k0)  # main now ordered after the first k0 slices ONLY
+        prescan = _prescan_prefix(data, ranges[:ri], block_counts, nblocks, n,
+                                  arrived=min(n, k0 * span), schema=schema,
+                                  record_type=record_type)
+    consume_slices(S)
+    off, lens, clean = _emit_and_chain(data, ranges, block_counts, n)
+    if not clean:
+        prescan = None  # host-stitched chain: prescanned rows are stale
+    return decode_device(data, off, lens, schema, record_type, verify_crc,
+                         prescan=prescan)
 
 
 def read_files_to_batch(paths, schema: StructType, record_type: str,

@@ -72,6 +72,7 @@ _CHUNK = 64 << 20  # 64 MiB: overlap file IO with PCIe copies chunkwise
 # engine knobs: overridable via env for tuning
 _READ_SLICE = int(os.environ.get("TFREC_READ_SLICE", 48 << 20))
 _WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 5))
+_PRESCAN = os.environ.get("TFREC_PRESCAN", "1") != "0"
 
 
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:
@@ -731,7 +732,8 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
     # stage point: once ~2/3 of the slices have landed, scan the records that
     # are fully inside the arrived prefix WHILE the tail slices are still in
     # DMA flight (hides most of the structure-scan under the H2D)
-    k0 = (2 * S) // 3 if S >= 3 and record_type != "ByteArray" else 0
+    k0 = ((2 * S) // 3 if _PRESCAN and S >= 3 and record_type != "ByteArray"
+          else 0)
 
     ri = 0
     base = 0

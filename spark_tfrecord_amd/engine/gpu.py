@@ -72,7 +72,12 @@ _CHUNK = 64 << 20  # 64 MiB: overlap file IO with PCIe copies chunkwise
 # engine knobs: overridable via env for tuning
 _READ_SLICE = int(os.environ.get("TFREC_READ_SLICE", 48 << 20))
 _WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 5))
-_PRESCAN = os.environ.get("TFREC_PRESCAN", "1") != "0"
+# Prescan (structure-scan the arrived prefix under the tail DMA) measured
+# NET-NEGATIVE on this host: every host<->device copy executes as a blit
+# KERNEL (no SDMA for host-registered or torch-pinned memory — see
+# exp/exp_sdma.py), so "overlapped" compute contends with the copy kernels
+# and the extra syncs cost more than the hidden scan. Off by default.
+_PRESCAN = os.environ.get("TFREC_PRESCAN", "0") == "1"
 
 
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:

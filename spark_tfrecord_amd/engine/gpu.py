@@ -425,19 +425,15 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     # FieldStat[R][F] as int64 [R,F,6]: pos,len,nvals,nbytes,nlists,(kind|err)
     stats = torch.empty((R, F, 6), dtype=torch.int64, device=device)
     err = torch.zeros(1, dtype=torch.int32, device=device)
-    # frame CRC verification runs CONCURRENTLY with the structure scan on a
-    # side stream — the two kernels read the same (independent) bytes, so
-    # the CRC pass hides entirely under the longer parse
+    # frame CRC verification is FUSED into the structure scan: the record
+    # bytes are CRC'd while L2-hot from the parse. (A concurrent side-stream
+    # CRC pass was tried and reverted: every host<->device copy on this
+    # stack is a blit KERNEL, so "free" overlap doesn't exist and the fused
+    # form has ~1 ms less total GPU work per 215 MB.)
     main = torch.cuda.current_stream()
     crc_err = None
-    crc_stream = None
     if verify_crc:
         crc_err = torch.full((1,), -1, dtype=torch.int64, device=device)
-        crc_stream = _dma_streams()[1]
-        crc_stream.wait_stream(main)
-        _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(), lens.data_ptr(),
-                               R, crc_err.data_ptr(), crc_stream.cuda_stream)
-        crc_err.record_stream(crc_stream)
     r0 = 0
     err1 = None
     if prescan is not None:
@@ -449,11 +445,18 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
             err1 = None
     if R > r0:
         stride = F * 6 * 8
-        _native.gpu_scan_records(data.data_ptr() , off.data_ptr() + r0 * 8,
+        _native.gpu_scan_records(data.data_ptr(), off.data_ptr() + r0 * 8,
                                  lens.data_ptr() + r0 * 8, R - r0,
                                  FMT[record_type], blob.data_ptr(), F,
                                  stats.data_ptr() + r0 * stride,
-                                 err.data_ptr(), 0, _stream())
+                                 err.data_ptr(),
+                                 crc_err.data_ptr() if verify_crc else 0,
+                                 _stream())
+    if verify_crc and r0 > 0:
+        # prescanned rows skipped the fused path: verify them separately
+        _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(),
+                               lens.data_ptr(), r0, crc_err.data_ptr(),
+                               _stream())
 
     # Per-field exclusive prefix sums ([F, R+1]): rocprim strided scans read
     # the stat column straight out of the [R, F, 6] buffer — no transpose
@@ -472,7 +475,6 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     totals = torch.stack([val_base[:, -1], byte_base[:, -1], list_base[:, -1]])
     totals_h = totals.cpu()  # one sync for all allocations
     if verify_crc:
-        main.wait_stream(crc_stream)
         bad = int(crc_err.item())
         if bad != -1:
             raise RuntimeError(

@@ -319,3 +319,17 @@ class TestGpuEndToEnd:
             ["part=1", "part=2"]
         df = stf.read_tfrecord(out, engine="gpu").sort("x")
         assert [r["part"] for r in df.collect()] == [1, 1, 2, 2]
+
+    def test_gpu_multi_column_partition(self, tmp_sandbox):
+        out = str(tmp_sandbox / "mc")
+        data = {"a": np.array([1, 1, 2, 2, 1], np.int64),
+                "b": ["x", "y", "x", "x", "y"],
+                "v": np.arange(5, dtype=np.int64)}
+        stf.write_tfrecord(data, out, partition_by=["a", "b"], engine="gpu")
+        dirs = sorted(f"{d}/{s}" for d in os.listdir(out) if d.startswith("a=")
+                      for s in os.listdir(os.path.join(out, d)))
+        assert dirs == ["a=1/b=x", "a=1/b=y", "a=2/b=x"]
+        df = stf.read_tfrecord(out, engine="gpu").sort("v")
+        rows = df.collect()
+        assert [(r["a"], r["b"], r["v"]) for r in rows] == \
+            [(1, "x", 0), (1, "y", 1), (2, "x", 2), (2, "x", 3), (1, "y", 4)]

@@ -142,15 +142,22 @@ def _bytes_arrow(col: WireColumn, dt: DataType, elem_idx: np.ndarray) -> pa.Arra
     """Build a large_utf8/large_binary array for the given string elements."""
     elem_off = np.asarray(col.elem_off)
     data = np.asarray(col.values)
+    t = pa.large_utf8() if isinstance(dt, StringType) else pa.large_binary()
+    n = len(elem_idx)
+    # identity selection (all elements in order): wrap the wire buffers
+    # zero-copy — the common whole-column read path
+    if n == len(elem_off) - 1 and (n == 0 or (
+            elem_idx[0] == 0 and elem_idx[-1] == n - 1)):
+        return pa.Array.from_buffers(
+            t, n, [None, pa.py_buffer(elem_off.astype(np.int64, copy=False)),
+                   pa.py_buffer(data)])
     starts = elem_off[elem_idx]
     lengths = elem_off[elem_idx + 1] - starts
     gathered = data[_ragged_gather_idx(starts, lengths)]
-    new_off = np.zeros(len(elem_idx) + 1, np.int64)
+    new_off = np.zeros(n + 1, np.int64)
     np.cumsum(lengths, out=new_off[1:])
-    t = pa.large_utf8() if isinstance(dt, StringType) else pa.large_binary()
     return pa.Array.from_buffers(
-        t, len(elem_idx),
-        [None, pa.py_buffer(new_off), pa.py_buffer(gathered)])
+        t, n, [None, pa.py_buffer(new_off), pa.py_buffer(gathered)])
 
 
 def wire_to_arrow(col: WireColumn, dt: DataType, nullable: bool, name: str,

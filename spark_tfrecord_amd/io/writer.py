@@ -115,15 +115,27 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
             continue
         chunk = table.slice(lo, hi - lo)
         batch = table_to_batch(chunk, schema)
+        fname = P.part_file_name(shard_offset + s, codec, job_id)
+        fpath = os.path.join(out_dir, fname)
         # device DevCols holds at most 64 fields; wider schemas encode on host
         if eng == "gpu" and len(batch.columns) <= 64:
             from ..engine import gpu as gpu_engine
-            raw = gpu_engine.encode_batch_from_cpu(batch, record_type)
+            dev_batch = gpu_engine.batch_to_device(batch)
+            if codec is None:
+                # encode sliced + DMA straight into the (temp) file mapping
+                tmp = fpath + ".inprogress"
+                nbytes = gpu_engine.write_batch_to_file(dev_batch, tmp,
+                                                        record_type)
+                os.replace(tmp, fpath)
+                if metrics is not None:
+                    metrics.add(rows=hi - lo, nbytes=nbytes, files=1)
+                continue
+            raw = gpu_engine.device_to_bytes(
+                gpu_engine.encode_device(dev_batch, record_type))
         else:
             raw = cpu_engine.encode_batch(batch, record_type)
         payload = P.compress_bytes(raw, codec)
-        fname = P.part_file_name(shard_offset + s, codec, job_id)
-        P.write_file_atomic(payload, os.path.join(out_dir, fname))
+        P.write_file_atomic(payload, fpath)
         if metrics is not None:
             metrics.add(rows=hi - lo, nbytes=len(payload), files=1)
 

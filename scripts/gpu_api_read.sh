@@ -30,3 +30,19 @@ torch.cuda.synchronize()
 t1 = time.perf_counter()
 print(f"device-resident stream: {total/(t1-t0)/1e6:.1f}M rows/s")
 PY
+timeout 300 python - <<'PY' 2>&1 | tail -2
+import time, os
+import numpy as np
+import spark_tfrecord_amd as stf
+from bench import make_batch
+from spark_tfrecord_amd.arrow_interop import batch_to_table
+
+rows = 1_000_000
+table = batch_to_table(make_batch(rows, seed=6))
+out = "/dev/shm/apiwrite/t"
+stf.write_tfrecord(table, out, engine="gpu", mode="overwrite")  # warm
+t0 = time.perf_counter()
+stf.write_tfrecord(table, out, engine="gpu", mode="overwrite")
+t1 = time.perf_counter()
+print(f"API write (flagship schema, arrow in): {rows/(t1-t0)/1e6:.1f}M rows/s")
+PY

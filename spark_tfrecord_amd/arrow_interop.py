@@ -331,6 +331,26 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
         np.cumsum(lens, out=row_off[1:])
         return WireColumn(kind, False, presence, row_off, vals)
 
+    # Fast path: 1-D string/binary list (BytesList arrays like token lists)
+    if not seq and isinstance(dt, ArrayType) and kind == KIND_BYTES and \
+            (pa.types.is_list(arr.type) or pa.types.is_large_list(arr.type)):
+        inner = arr.values
+        if (null_count == 0 and arr.offset == 0 and inner.null_count == 0
+                and (pa.types.is_string(inner.type)
+                     or pa.types.is_large_string(inner.type)
+                     or pa.types.is_binary(inner.type)
+                     or pa.types.is_large_binary(inner.type))):
+            off1 = _np_offsets(arr).astype(np.int64)
+            a = inner.cast(pa.large_binary())
+            if a.offset == 0 and off1[0] == 0:
+                E = int(off1[-1])
+                eoff = np.frombuffer(a.buffers()[1], np.int64, len(a) + 1)
+                db = a.buffers()[2]
+                data = (np.frombuffer(db, np.uint8, len(db)) if db
+                        else np.zeros(0, np.uint8))
+                return WireColumn(kind, False, presence, off1,
+                                  data[:int(eoff[E])], eoff[:E + 1])
+
     # Fast path: 2-D numeric nested list (SequenceExample FeatureList of
     # Int64List/FloatList). Offsets compose: values-per-row cumulative is
     # off2[off1] — no python loop over 1M ragged rows.

@@ -168,21 +168,31 @@ def device_to_file(img: torch.Tensor, path: str):
     if n == 0:
         open(path, "wb").close()
         return
-    ptr, pinned = _native.file_mmap_pinned(path, n, True)
-    if pinned:
-        _multi_dma(ptr, img.data_ptr(), n, _native.gpu_memcpy_d2h,
-                   after_main=False)
-        return
+    # Registered-mapping DMA only pays when the mapping is reused (or the
+    # file is small): hipHostRegister costs ~0.16 ms/MB on fresh files
+    # (exp/exp_thp.py). Fresh big files take the one-shot mmap +
+    # pinned-staged parallel-memcpy path instead.
+    if _native.file_mmap_cached(path, n) or n <= _FRESH_MMAP_MAX:
+        ptr, pinned = _native.file_mmap_pinned(path, n, True)
+        if pinned:
+            _multi_dma(ptr, img.data_ptr(), n, _native.gpu_memcpy_d2h,
+                       after_main=False)
+            return
     _write_file_staged(img, path)
 
 
-def _write_file_staged(img: torch.Tensor, path: str):
-    import os as _os
+_FRESH_MMAP_MAX = int(os.environ.get("TFREC_FRESH_MMAP_MAX", 48 << 20))
 
+
+def _write_file_staged(img: torch.Tensor, path: str):
+    """Fresh-file writer: one-shot mmap of the sized file, D2H chunks into
+    pinned staging overlapped with a multi-threaded memcpy into the mapping
+    (writing through the mapping avoids the per-file inode mutex that
+    serializes parallel pwrite)."""
     n = img.numel()
     buf = pinned_buffer("fwrite", min(n, 2 * _CHUNK) or 1)
     ev = [torch.cuda.Event(), torch.cuda.Event()]
-    fd = _os.open(path, _os.O_RDWR | _os.O_CREAT, 0o644)
+    ptr = _native.mmap_plain(path, n, True)
     try:
         pos = 0
         which = 0
@@ -201,12 +211,11 @@ def _write_file_staged(img: torch.Tensor, path: str):
                     img[nxt:nxt + m2], non_blocking=True)
                 ev[nwhich].record()
             ev[which].synchronize()
-            _native.pwrite_parallel(fd, buf.data_ptr() + which * _CHUNK, m, pos)
+            _native.memcpy_parallel(ptr + pos, buf.data_ptr() + which * _CHUNK, m)
             pos = nxt
             which = nwhich
-        _os.ftruncate(fd, n)
     finally:
-        _os.close(fd)
+        _native.munmap_plain(ptr, n)
 
 
 # ---------------------------------------------------------------------------
@@ -647,7 +656,10 @@ def write_batch_to_file(batch: RecordBatch, path: str,
     total = int(bounds[-1])
     file = torch.empty(total, dtype=torch.uint8, device=device)
     err = torch.zeros(1, dtype=torch.int32, device=device)
-    ptr, pinned = _native.file_mmap_pinned(path, total, True)
+    if _native.file_mmap_cached(path, total) or total <= _FRESH_MMAP_MAX:
+        ptr, pinned = _native.file_mmap_pinned(path, total, True)
+    else:
+        ptr, pinned = 0, False  # fresh big file: skip registration cost
     if not pinned:
         _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                                  FMT[record_type], 0, R, frame_off.data_ptr(),

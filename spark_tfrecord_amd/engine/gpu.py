@@ -168,31 +168,28 @@ def device_to_file(img: torch.Tensor, path: str):
     if n == 0:
         open(path, "wb").close()
         return
-    # Registered-mapping DMA only pays when the mapping is reused (or the
-    # file is small): hipHostRegister costs ~0.16 ms/MB on fresh files
-    # (exp/exp_thp.py). Fresh big files take the one-shot mmap +
-    # pinned-staged parallel-memcpy path instead.
-    if _native.file_mmap_cached(path, n) or n <= _FRESH_MMAP_MAX:
-        ptr, pinned = _native.file_mmap_pinned(path, n, True)
-        if pinned:
-            _multi_dma(ptr, img.data_ptr(), n, _native.gpu_memcpy_d2h,
-                       after_main=False)
-            return
+    # Always map+register: on a FRESH file every write strategy is
+    # page-allocation-bound anyway (~6-7 GB/s measured across register+DMA,
+    # pwrite pool, mmap memcpy and plain write — exp/exp_freshwrite.py),
+    # and the registered mapping makes every REWRITE a pure ~56 GB/s DMA.
+    ptr, pinned = _native.file_mmap_pinned(path, n, True)
+    if pinned:
+        _multi_dma(ptr, img.data_ptr(), n, _native.gpu_memcpy_d2h,
+                   after_main=False)
+        return
     _write_file_staged(img, path)
 
 
-_FRESH_MMAP_MAX = int(os.environ.get("TFREC_FRESH_MMAP_MAX", 48 << 20))
-
-
 def _write_file_staged(img: torch.Tensor, path: str):
-    """Fresh-file writer: one-shot mmap of the sized file, D2H chunks into
-    pinned staging overlapped with a multi-threaded memcpy into the mapping
-    (writing through the mapping avoids the per-file inode mutex that
-    serializes parallel pwrite)."""
+    """Fallback writer when hipHostRegister is unavailable: D2H chunks into
+    pinned staging overlapped with plain write()s (single-threaded write
+    beats a pwrite pool on the per-inode mutex)."""
+    import os as _os
+
     n = img.numel()
     buf = pinned_buffer("fwrite", min(n, 2 * _CHUNK) or 1)
     ev = [torch.cuda.Event(), torch.cuda.Event()]
-    ptr = _native.mmap_plain(path, n, True)
+    fd = _os.open(path, _os.O_RDWR | _os.O_CREAT, 0o644)
     try:
         pos = 0
         which = 0
@@ -211,11 +208,12 @@ def _write_file_staged(img: torch.Tensor, path: str):
                     img[nxt:nxt + m2], non_blocking=True)
                 ev[nwhich].record()
             ev[which].synchronize()
-            _native.memcpy_parallel(ptr + pos, buf.data_ptr() + which * _CHUNK, m)
+            _native.pwrite_parallel(fd, buf.data_ptr() + which * _CHUNK, m, pos)
             pos = nxt
             which = nwhich
+        _os.ftruncate(fd, n)
     finally:
-        _native.munmap_plain(ptr, n)
+        _os.close(fd)
 
 
 # ---------------------------------------------------------------------------
@@ -656,10 +654,7 @@ def write_batch_to_file(batch: RecordBatch, path: str,
     total = int(bounds[-1])
     file = torch.empty(total, dtype=torch.uint8, device=device)
     err = torch.zeros(1, dtype=torch.int32, device=device)
-    if _native.file_mmap_cached(path, total) or total <= _FRESH_MMAP_MAX:
-        ptr, pinned = _native.file_mmap_pinned(path, total, True)
-    else:
-        ptr, pinned = 0, False  # fresh big file: skip registration cost
+    ptr, pinned = _native.file_mmap_pinned(path, total, True)
     if not pinned:
         _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                                  FMT[record_type], 0, R, frame_off.data_ptr(),

@@ -61,13 +61,6 @@ class IOPool {
     });
   }
 
-  void run_memcpy(u8* dst, const u8* src, i64 n) {
-    run_spans(n, [=](i64 o, i64 m) {
-      std::memcpy(dst + o, src + o, (size_t)m);
-      return 0;
-    });
-  }
-
   template <typename Fn>
   int run_spans(i64 n, Fn&& fn) {
     if (n <= 0) return 0;
@@ -173,14 +166,6 @@ void pread_parallel(int fd, uintptr_t ptr, i64 n, i64 file_off) {
     e = IOPool::instance().run(false, fd, (u8*)ptr, n, file_off);
   }
   if (e) throw std::runtime_error("pread failed: " + std::string(strerror(-e)));
-}
-
-// Plain memcpy split across the worker pool — used to fill a fresh file's
-// mmap from pinned staging (parallel pwrite to ONE file serializes on the
-// inode mutex; writing through the mapping doesn't).
-void memcpy_parallel(uintptr_t dst, uintptr_t src, i64 n) {
-  py::gil_scoped_release rel;
-  IOPool::instance().run_memcpy((u8*)dst, (const u8*)src, n);
 }
 
 struct BufView {
@@ -650,8 +635,6 @@ PYBIND11_MODULE(_native, m) {
   m.def("pread_parallel", &pread_parallel, py::arg("fd"), py::arg("ptr"),
         py::arg("n"), py::arg("file_off"),
         "Multi-threaded pread of n bytes at file_off into ptr");
-  m.def("memcpy_parallel", &memcpy_parallel, py::arg("dst"), py::arg("src"),
-        py::arg("n"), "Multi-threaded memcpy (fills fresh-file mmaps)");
   m.def("crc32c", &crc32c_py, "CRC32C (Castagnoli) of a byte buffer");
   m.def("masked_crc32c", &masked_crc32c_py, "TFRecord-masked CRC32C");
   m.def("scan_frames", &scan_frames, py::arg("data"), py::arg("verify_crc") = true,

@@ -904,42 +904,6 @@ void file_mmap_drop(const std::string& path) {
   drop_mapping_locked(path);
 }
 
-// Is there a live registered mapping for (path, n)? Callers use this to
-// decide between the register+DMA path (cache hit: free) and the one-shot
-// mmap + parallel-memcpy path (fresh big file: hipHostRegister costs
-// ~0.16 ms/MB on 4 KiB tmpfs pages — shmem THP is disabled on these boxes,
-// measured in exp/exp_thp.py).
-bool file_mmap_cached(const std::string& path, i64 n) {
-  std::lock_guard<std::mutex> lk(g_mmap_mu);
-  auto it = g_mmap_cache.find(path);
-  if (it == g_mmap_cache.end() || !it->second.pinned ||
-      it->second.n != (size_t)n)
-    return false;
-  struct stat st {};
-  if (stat(path.c_str(), &st) != 0) return false;
-  return it->second.dev == st.st_dev && it->second.ino == st.st_ino;
-}
-
-// One-shot uncached mapping (create/size the file first when writable);
-// caller must munmap_plain it.
-uintptr_t mmap_plain(const std::string& path, i64 n, bool writable) {
-  int fd = ::open(path.c_str(), writable ? (O_RDWR | O_CREAT) : O_RDONLY, 0644);
-  if (fd < 0)
-    throw std::runtime_error("open failed: " + path + ": " + strerror(errno));
-  if (writable && ftruncate(fd, (off_t)n) != 0) {
-    int e = errno;
-    ::close(fd);
-    throw std::runtime_error("ftruncate failed: " + path + ": " + strerror(e));
-  }
-  int prot = PROT_READ | (writable ? PROT_WRITE : 0);
-  void* p = mmap(nullptr, (size_t)n, prot, MAP_SHARED, fd, 0);
-  ::close(fd);
-  if (p == MAP_FAILED)
-    throw std::runtime_error("mmap failed: " + path + ": " + strerror(errno));
-  return (uintptr_t)p;
-}
-
-void munmap_plain(uintptr_t p, i64 n) { munmap((void*)p, (size_t)n); }
 
 void gpu_memcpy_d2h(uintptr_t dst, uintptr_t src, i64 n, uintptr_t stream) {
   HIP_CHECK(hipMemcpyAsync((void*)dst, (const void*)src, (size_t)n,
@@ -975,9 +939,6 @@ void register_gpu(py::module_& m) {
   m.def("file_mmap_pinned", &file_mmap_pinned, py::arg("path"), py::arg("n"),
         py::arg("writable"));
   m.def("file_mmap_drop", &file_mmap_drop);
-  m.def("file_mmap_cached", &file_mmap_cached);
-  m.def("mmap_plain", &mmap_plain);
-  m.def("munmap_plain", &munmap_plain);
   m.def("gpu_memcpy_d2h", &gpu_memcpy_d2h);
   m.def("gpu_memcpy_h2d", &gpu_memcpy_h2d);
   m.def("gpu_devcols_bytes", &gpu_devcols_bytes);

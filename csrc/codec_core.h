@@ -478,6 +478,366 @@ TFR_HOSTDEV inline int32_t scan_record(const u8* data, i64 abs_off, i64 rec_len,
 }
 
 // ---------------------------------------------------------------------------
+// Fused single-pass structure scan + payload CRC32C (cursor form).
+// The split form loads every payload byte at least twice: once through the
+// varint parser and once through the CRC loop. The cursor walks the record
+// ONCE in 8-byte windows; each window feeds the parser (funnel-shifted
+// peeks) and, when fully consumed, one slicing-by-8 CRC step. Key bytes are
+// additionally read directly for the schema-name compare (short).
+// Records whose map entries put the value before the key (legal protobuf,
+// never produced by TF or this library) return ERR_RETRY_UNFUSED and the
+// caller re-scans them with the two-pass form.
+// ---------------------------------------------------------------------------
+
+constexpr int32_t ERR_RETRY_UNFUSED = -100;
+
+TFR_HOSTDEV inline u32 crc32c_step8(u32 crc, u64 w, const uint32_t (*tab)[256]) {
+  w ^= crc;
+  return tab[7][w & 0xFF] ^ tab[6][(w >> 8) & 0xFF] ^ tab[5][(w >> 16) & 0xFF] ^
+         tab[4][(w >> 24) & 0xFF] ^ tab[3][(w >> 32) & 0xFF] ^
+         tab[2][(w >> 40) & 0xFF] ^ tab[1][(w >> 48) & 0xFF] ^
+         tab[0][(w >> 56) & 0xFF];
+}
+
+struct ScanCur {
+  const u8* base;  // record payload start
+  i64 len;         // payload length
+  i64 pos;         // parse position (monotonic, 0..len)
+  i64 wpos;        // start of window w0 (multiple of 8)
+  u64 w0, w1;      // windows [wpos, wpos+8) and [wpos+8, wpos+16)
+  u32 crc;         // running ~crc state
+  const uint32_t (*tab)[256];
+};
+
+TFR_HOSTDEV inline u64 cur_load(const ScanCur& c, i64 at) {
+  if (at + 8 <= c.len) {
+    u64 w;
+    __builtin_memcpy(&w, c.base + at, 8);
+    return w;
+  }
+  u64 w = 0;
+  for (i64 b = at; b < c.len; ++b) w |= (u64)c.base[b] << (8 * (b - at));
+  return w;
+}
+
+TFR_HOSTDEV inline void cur_init(ScanCur& c, const u8* base, i64 len,
+                                 const uint32_t (*tab)[256]) {
+  c.base = base;
+  c.len = len;
+  c.pos = 0;
+  c.wpos = 0;
+  c.tab = tab;
+  c.crc = 0xFFFFFFFFu;
+  c.w0 = cur_load(c, 0);
+  c.w1 = cur_load(c, 8);
+}
+
+// Invariant: wpos <= pos <= wpos + 8 (so [pos, pos+8) is inside w0/w1).
+TFR_HOSTDEV inline u64 cur_peek(const ScanCur& c) {
+  int sh = (int)(c.pos - c.wpos) * 8;
+  if (sh == 0) return c.w0;
+  if (sh == 64) return c.w1;
+  return (c.w0 >> sh) | (c.w1 << (64 - sh));
+}
+
+TFR_HOSTDEV inline void cur_advance(ScanCur& c, i64 k) {
+  c.pos += k;
+  while (c.pos > c.wpos + 8) {
+    if (c.wpos + 8 <= c.len) c.crc = crc32c_step8(c.crc, c.w0, c.tab);
+    c.w0 = c.w1;
+    c.wpos += 8;
+    c.w1 = cur_load(c, c.wpos + 8);
+  }
+}
+
+// CRC of any bytes not yet folded (tail window + sub-8 remainder) -> final.
+TFR_HOSTDEV inline u32 cur_finish_crc(ScanCur& c) {
+  while (c.wpos + 8 <= c.len) {
+    c.crc = crc32c_step8(c.crc, c.w0, c.tab);
+    c.w0 = c.w1;
+    c.wpos += 8;
+    c.w1 = cur_load(c, c.wpos + 8);
+  }
+  u32 crc = c.crc;
+  for (i64 b = c.wpos; b < c.len; ++b)
+    crc = c.tab[0][(crc ^ c.base[b]) & 0xFF] ^ (crc >> 8);
+  return ~crc;
+}
+
+TFR_HOSTDEV inline bool cur_varint(ScanCur& c, i64 end, u64* out) {
+  i64 avail = end - c.pos;
+  if (avail <= 0) return false;
+  u64 w = cur_peek(c);
+  u64 stops = ~w & kMsbMask;
+  if (stops) {
+    int nb = (ctz64(stops) >> 3) + 1;
+    if (nb > avail) return false;  // terminator past the region end
+    u64 v = 0;
+    for (int i = 0; i < nb; ++i) v |= ((w >> (8 * i)) & 0x7F) << (7 * i);
+    *out = v;
+    cur_advance(c, nb);
+    return true;
+  }
+  if (avail < 9) return false;
+  u64 v = 0;
+  for (int i = 0; i < 8; ++i) v |= ((w >> (8 * i)) & 0x7F) << (7 * i);
+  cur_advance(c, 8);
+  for (int shift = 56; shift < 64 && c.pos < end; shift += 7) {
+    u8 b = (u8)cur_peek(c);
+    v |= (u64)(b & 0x7F) << shift;
+    cur_advance(c, 1);
+    if (!(b & 0x80)) {
+      *out = v;
+      return true;
+    }
+  }
+  return false;
+}
+
+TFR_HOSTDEV inline int32_t cur_skip_field(ScanCur& c, i64 end, u32 wt) {
+  u64 tmp;
+  switch (wt) {
+    case 0:
+      return cur_varint(c, end, &tmp) ? ERR_OK : ERR_BAD_VARINT;
+    case 1:
+      if (end - c.pos < 8) return ERR_TRUNCATED;
+      cur_advance(c, 8);
+      return ERR_OK;
+    case 2:
+      if (!cur_varint(c, end, &tmp)) return ERR_BAD_VARINT;
+      if ((u64)(end - c.pos) < tmp) return ERR_TRUNCATED;
+      cur_advance(c, (i64)tmp);
+      return ERR_OK;
+    case 5:
+      if (end - c.pos < 4) return ERR_TRUNCATED;
+      cur_advance(c, 4);
+      return ERR_OK;
+    default:
+      return ERR_BAD_WIRETYPE;
+  }
+}
+
+// Count varint terminators (MSB clear) in the next n bytes, window-wise.
+TFR_HOSTDEV inline i64 cur_count_terms(ScanCur& c, i64 n) {
+  i64 cnt = 0;
+  while (n >= 8) {
+    cnt += popcount64(~cur_peek(c) & kMsbMask);
+    cur_advance(c, 8);
+    n -= 8;
+  }
+  if (n > 0) {
+    u64 mask = kMsbMask >> (8 * (8 - n));
+    cnt += popcount64(~cur_peek(c) & mask);
+    cur_advance(c, n);
+  }
+  return cnt;
+}
+
+TFR_HOSTDEV inline int32_t cur_count_list_body(ScanCur& c, i64 end, int32_t kind,
+                                               i64* nvals, i64* nbytes) {
+  while (c.pos < end) {
+    u64 tag;
+    if (!cur_varint(c, end, &tag)) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    if (fieldno != 1) {
+      int32_t rc = cur_skip_field(c, end, wt);
+      if (rc != ERR_OK) return rc;
+      continue;
+    }
+    if (kind == KIND_BYTES) {
+      if (wt != 2) return ERR_BAD_WIRETYPE;
+      u64 blen;
+      if (!cur_varint(c, end, &blen)) return ERR_BAD_VARINT;
+      if ((u64)(end - c.pos) < blen) return ERR_TRUNCATED;
+      *nvals += 1;
+      *nbytes += (i64)blen;
+      cur_advance(c, (i64)blen);
+    } else if (kind == KIND_FLOAT) {
+      if (wt == 2) {
+        u64 blen;
+        if (!cur_varint(c, end, &blen)) return ERR_BAD_VARINT;
+        if ((u64)(end - c.pos) < blen) return ERR_TRUNCATED;
+        *nvals += (i64)(blen / 4);
+        cur_advance(c, (i64)blen);
+      } else if (wt == 5) {
+        if (end - c.pos < 4) return ERR_TRUNCATED;
+        *nvals += 1;
+        cur_advance(c, 4);
+      } else {
+        return ERR_BAD_WIRETYPE;
+      }
+    } else {  // KIND_INT64
+      if (wt == 2) {
+        u64 blen;
+        if (!cur_varint(c, end, &blen)) return ERR_BAD_VARINT;
+        if ((u64)(end - c.pos) < blen) return ERR_TRUNCATED;
+        if (blen && (c.base[c.pos + (i64)blen - 1] & 0x80))
+          return ERR_BAD_VARINT;  // packed run ends mid-varint
+        *nvals += cur_count_terms(c, (i64)blen);
+      } else if (wt == 0) {
+        u64 v;
+        if (!cur_varint(c, end, &v)) return ERR_BAD_VARINT;
+        *nvals += 1;
+      } else {
+        return ERR_BAD_WIRETYPE;
+      }
+    }
+  }
+  return ERR_OK;
+}
+
+TFR_HOSTDEV inline int32_t cur_scan_feature_body(ScanCur& c, i64 end,
+                                                 int32_t expect_kind,
+                                                 int32_t* kind_found, i64* nvals,
+                                                 i64* nbytes) {
+  while (c.pos < end) {
+    u64 tag;
+    if (!cur_varint(c, end, &tag)) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    if (fieldno >= 1 && fieldno <= 3 && wt == 2) {
+      u64 blen;
+      if (!cur_varint(c, end, &blen)) return ERR_BAD_VARINT;
+      if ((u64)(end - c.pos) < blen) return ERR_TRUNCATED;
+      *kind_found = (int32_t)fieldno;
+      if (expect_kind >= 0 && (int32_t)fieldno != expect_kind)
+        return ERR_KIND_MISMATCH;
+      int32_t rc = cur_count_list_body(c, c.pos + (i64)blen, (int32_t)fieldno,
+                                       nvals, nbytes);
+      if (rc != ERR_OK) return rc;
+    } else {
+      int32_t rc = cur_skip_field(c, end, wt);
+      if (rc != ERR_OK) return rc;
+    }
+  }
+  return ERR_OK;
+}
+
+TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_rel,
+                                                  const SchemaView& schema,
+                                                  int want_seq, FieldStat* stats) {
+  while (c.pos < end) {
+    u64 tag;
+    if (!cur_varint(c, end, &tag)) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    if (fieldno != 1 || wt != 2) {
+      int32_t rc = cur_skip_field(c, end, wt);
+      if (rc != ERR_OK) return rc;
+      continue;
+    }
+    u64 entry_len;
+    if (!cur_varint(c, end, &entry_len)) return ERR_BAD_VARINT;
+    if ((u64)(end - c.pos) < entry_len) return ERR_TRUNCATED;
+    i64 ee = c.pos + (i64)entry_len;
+    int f = -1;
+    bool key_seen = false;
+    while (c.pos < ee) {
+      u64 etag;
+      if (!cur_varint(c, ee, &etag)) return ERR_BAD_VARINT;
+      u32 efn = (u32)(etag >> 3), ewt = (u32)(etag & 7);
+      if (efn == 1 && ewt == 2) {  // key
+        u64 klen;
+        if (!cur_varint(c, ee, &klen)) return ERR_BAD_VARINT;
+        if ((u64)(ee - c.pos) < klen) return ERR_TRUNCATED;
+        f = schema_find(schema, c.base + c.pos, (int)klen, want_seq);
+        key_seen = true;
+        cur_advance(c, (i64)klen);
+      } else if (efn == 2 && ewt == 2) {  // value (Feature / FeatureList)
+        if (!key_seen) return ERR_RETRY_UNFUSED;  // value before key: bail
+        u64 vlen;
+        if (!cur_varint(c, ee, &vlen)) return ERR_BAD_VARINT;
+        if ((u64)(ee - c.pos) < vlen) return ERR_TRUNCATED;
+        i64 ve = c.pos + (i64)vlen;
+        if (f < 0) {  // unknown feature: consume body (CRC still covers it)
+          cur_advance(c, (i64)vlen);
+          continue;
+        }
+        FieldStat* st = &stats[f];
+        st->pos = data_rel + c.pos;
+        st->len = (i64)vlen;
+        if (!want_seq) {
+          int32_t rc = cur_scan_feature_body(c, ve, schema.fields[f].kind,
+                                             &st->kind_found, &st->nvals,
+                                             &st->nbytes);
+          if (rc == ERR_RETRY_UNFUSED) return rc;
+          if (rc != ERR_OK) st->err = rc;  // mirror two-pass: record and go on
+        } else {
+          while (c.pos < ve && st->err == ERR_OK) {
+            // FeatureList: repeated Feature (field 1)
+            u64 ltag;
+            if (!cur_varint(c, ve, &ltag)) {
+              st->err = ERR_BAD_VARINT;
+              break;
+            }
+            u32 lfn = (u32)(ltag >> 3), lwt = (u32)(ltag & 7);
+            if (lfn == 1 && lwt == 2) {
+              u64 flen;
+              if (!cur_varint(c, ve, &flen) || (u64)(ve - c.pos) < flen) {
+                st->err = ERR_TRUNCATED;
+                break;
+              }
+              st->nlists += 1;
+              int32_t kf = 0;
+              int32_t rc = cur_scan_feature_body(c, c.pos + (i64)flen,
+                                                 schema.fields[f].kind, &kf,
+                                                 &st->nvals, &st->nbytes);
+              if (rc == ERR_RETRY_UNFUSED) return rc;
+              if (rc != ERR_OK) {
+                st->err = rc;
+                break;
+              }
+              if (kf) st->kind_found = kf;
+            } else {
+              int32_t rc = cur_skip_field(c, ve, lwt);
+              if (rc != ERR_OK) {
+                st->err = ERR_TRUNCATED;
+                break;
+              }
+            }
+          }
+        }
+        if (c.pos < ve) cur_advance(c, ve - c.pos);  // resync after stat error
+      } else {
+        int32_t rc = cur_skip_field(c, ee, ewt);
+        if (rc != ERR_OK) return rc;
+      }
+    }
+  }
+  return ERR_OK;
+}
+
+// Fused pass-A entry: fills stats AND returns the payload CRC32C through
+// *crc_out (valid only when the return code is ERR_OK). abs_off as in
+// scan_record. ERR_RETRY_UNFUSED => caller must redo with the two-pass form.
+TFR_HOSTDEV inline int32_t scan_record_fused(const u8* data, i64 abs_off,
+                                             i64 rec_len, int32_t fmt,
+                                             const SchemaView& schema,
+                                             FieldStat* stats, u32* crc_out,
+                                             const uint32_t (*tab)[256]) {
+  ScanCur c;
+  cur_init(c, data + abs_off, rec_len, tab);
+  while (c.pos < rec_len) {
+    u64 tag;
+    if (!cur_varint(c, rec_len, &tag)) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    bool is_features = fieldno == 1;
+    bool is_fl = (fmt == FMT_SEQUENCE && fieldno == 2);
+    if ((is_features || is_fl) && wt == 2) {
+      u64 blen;
+      if (!cur_varint(c, rec_len, &blen)) return ERR_BAD_VARINT;
+      if ((u64)(rec_len - c.pos) < blen) return ERR_TRUNCATED;
+      int32_t rc = cur_scan_features_body(c, c.pos + (i64)blen, abs_off, schema,
+                                          is_fl ? 1 : 0, stats);
+      if (rc != ERR_OK) return rc;
+    } else {
+      int32_t rc = cur_skip_field(c, rec_len, wt);
+      if (rc != ERR_OK) return rc;
+    }
+  }
+  *crc_out = cur_finish_crc(c);
+  return ERR_OK;
+}
+
+// ---------------------------------------------------------------------------
 // Decode pass B: value extraction for one (record, field), given the body
 // extent from pass A and destination offsets from the prefix sums.
 // ---------------------------------------------------------------------------

@@ -611,6 +611,51 @@ py::dict infer_schema_codes(py::buffer data, py::array_t<i64> rec_off,
   return out;
 }
 
+// TEST-ONLY: run pass-A on host in either form and return the raw stats +
+// per-record payload CRCs, so the fused cursor scan can be differentially
+// tested against the two-pass reference on arbitrary inputs.
+py::dict scan_stats_debug(py::buffer data, py::array_t<i64> rec_off,
+                          py::array_t<i64> rec_len, int32_t fmt,
+                          py::bytes schema_blob, bool fused) {
+  py::buffer_info info;
+  BufView buf = as_bytes(data, info);
+  std::string blob = schema_blob;
+  SchemaView schema = schema_view(reinterpret_cast<const u8*>(blob.data()));
+  const int F = schema.nfields;
+  auto off = rec_off.unchecked<1>();
+  auto len = rec_len.unchecked<1>();
+  i64 R = rec_off.size();
+  auto stats_arr = py::array_t<i64>({R, (i64)F, (i64)6});
+  auto crc_arr = py::array_t<i64>(R);
+  auto rc_arr = py::array_t<i64>(R);
+  auto* st_all = reinterpret_cast<FieldStat*>(stats_arr.mutable_data());
+  for (i64 r = 0; r < R; ++r) {
+    FieldStat* st = st_all + r * F;
+    for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+    int32_t rc;
+    u32 crc = 0;
+    if (fused) {
+      rc = scan_record_fused(buf.data, off(r), len(r), fmt, schema, st, &crc,
+                             kCrcTables.t);
+      if (rc == ERR_RETRY_UNFUSED) {
+        for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+        rc = scan_record(buf.data, off(r), len(r), fmt, schema, st);
+        crc = crc32c(buf.data + off(r), (size_t)len(r));
+      }
+    } else {
+      rc = scan_record(buf.data, off(r), len(r), fmt, schema, st);
+      crc = crc32c(buf.data + off(r), (size_t)len(r));
+    }
+    crc_arr.mutable_at(r) = rc == ERR_OK ? (i64)crc : -1;
+    rc_arr.mutable_at(r) = rc;
+  }
+  py::dict out;
+  out["stats"] = stats_arr;
+  out["crc"] = crc_arr;
+  out["rc"] = rc_arr;
+  return out;
+}
+
 u32 crc32c_py(py::buffer data) {
   py::buffer_info info;
   BufView buf = as_bytes(data, info);
@@ -635,6 +680,10 @@ PYBIND11_MODULE(_native, m) {
   m.def("pread_parallel", &pread_parallel, py::arg("fd"), py::arg("ptr"),
         py::arg("n"), py::arg("file_off"),
         "Multi-threaded pread of n bytes at file_off into ptr");
+  m.def("scan_stats_debug", &scan_stats_debug, py::arg("data"),
+        py::arg("rec_off"), py::arg("rec_len"), py::arg("fmt"),
+        py::arg("schema_blob"), py::arg("fused"),
+        "TEST-ONLY: pass-A stats + payload CRCs (fused cursor vs two-pass)");
   m.def("crc32c", &crc32c_py, "CRC32C (Castagnoli) of a byte buffer");
   m.def("masked_crc32c", &masked_crc32c_py, "TFRecord-masked CRC32C");
   m.def("scan_frames", &scan_frames, py::arg("data"), py::arg("verify_crc") = true,

@@ -104,18 +104,32 @@ __global__ void scan_records_kernel(const u8* __restrict__ data,
   SchemaView schema = schema_view(schema_blob);
   for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
        r += (i64)gridDim.x * blockDim.x) {
+    FieldStat* st = stats + r * F;
+    for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+    int32_t rc;
     if (crc_err) {
+      // fused cursor pass: each payload window is loaded ONCE for both the
+      // structure scan and the CRC (codec_core.h ScanCur)
+      u32 payload_crc = 0;
+      rc = scan_record_fused(data, off[r], len[r], fmt, schema, st,
+                             &payload_crc, tab);
+      if (rc == ERR_RETRY_UNFUSED) {  // value-before-key map entry: redo
+        for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+        rc = scan_record(data, off[r], len[r], fmt, schema, st);
+        payload_crc = crc32c_sw(data + off[r], (size_t)len[r], 0, tab);
+      }
+      if (rc != ERR_OK)  // parse bailed mid-record: recompute CRC plainly
+        payload_crc = crc32c_sw(data + off[r], (size_t)len[r], 0, tab);
       const u8* h = data + off[r] - 12;
       u32 len_crc, data_crc;
       __builtin_memcpy(&len_crc, h + 8, 4);
       __builtin_memcpy(&data_crc, h + 12 + len[r], 4);
       bool ok = mask_crc(crc32c_sw(h, 8, 0, tab)) == len_crc &&
-                mask_crc(crc32c_sw(h + 12, (size_t)len[r], 0, tab)) == data_crc;
+                mask_crc(payload_crc) == data_crc;
       if (!ok) atomicMin(crc_err, (unsigned long long)(r + 1));
+    } else {
+      rc = scan_record(data, off[r], len[r], fmt, schema, st);
     }
-    FieldStat* st = stats + r * F;
-    for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
-    int32_t rc = scan_record(data, off[r], len[r], fmt, schema, st);
     if (rc != ERR_OK) err[0] = rc;
     for (int f = 0; f < F; ++f)
       if (st[f].err != ERR_OK) err[0] = st[f].err;

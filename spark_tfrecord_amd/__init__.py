@@ -14,7 +14,7 @@ of a JVM + Spark. See SURVEY.md for the structural map of the reference.
 import torch  # noqa: F401  (intentional import order)
 
 from .api import DataFrame, DataFrameReader, DataFrameWriter, TFRecordSession, session
-from .io.reader import read_tfrecord
+from .io.reader import count_tfrecord, read_tfrecord
 from .io.writer import write_tfrecord
 from .io.validate import validate_tfrecord
 from . import torch_data
@@ -38,6 +38,7 @@ __version__ = "0.1.0"
 __all__ = [
     "session", "DataFrame", "DataFrameReader", "DataFrameWriter",
     "TFRecordSession", "read_tfrecord", "write_tfrecord", "validate_tfrecord",
+    "count_tfrecord",
     "DataType", "NullType", "IntegerType", "LongType", "FloatType",
     "DoubleType", "DecimalType", "StringType", "BinaryType", "ArrayType",
     "StructField", "StructType",

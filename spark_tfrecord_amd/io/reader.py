@@ -90,9 +90,41 @@ def _partition_col_array(values: List[str], n: int):
             type=pa.large_utf8())
 
 
+def count_tfrecord(path: str, engine: str = "auto") -> int:
+    """Total record count without decoding any payloads (frame scan only —
+    the GPU path discovers and chains frame boundaries in parallel)."""
+    files = P.list_data_files(path)
+    if not files:
+        raise FileNotFoundError(f"No TFRecord files found under {path}")
+    eng = engine_mod.resolve_engine(engine)
+    total = 0
+    for f in files:
+        if eng == "gpu" and P.codec_from_path(f) is None:
+            if os.path.getsize(f) == 0:
+                continue
+            from ..engine import gpu as gpu_engine
+
+            off, _ = gpu_engine.scan_frames_device(
+                gpu_engine.read_file_to_device(f))
+            total += int(off.numel())
+        else:
+            data = _load_file(f)
+            if data.size == 0:
+                continue
+            off, _ = _native.scan_frames(data, False)
+            total += len(off)
+    return total
+
+
 def read_tfrecord(path: str, schema: Optional[StructType] = None,
                   record_type: str = "Example", engine: str = "auto",
-                  verify_crc: bool = True, base_dir: Optional[str] = None):
+                  verify_crc: bool = True, base_dir: Optional[str] = None,
+                  columns: Optional[List[str]] = None):
+    """Read a TFRecord dataset as a DataFrame.
+
+    `columns` projects the read: only the named fields are scanned and
+    extracted (the schema-driven kernels skip other features entirely, like
+    the reference's deserializer ignores unknown ones)."""
     from ..api import DataFrame
 
     if record_type not in ("Example", "SequenceExample", "ByteArray"):
@@ -110,6 +142,13 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
         with StageTimer(metrics, "infer_schema"):
             schema = (byte_array_schema() if record_type == "ByteArray"
                       else infer_schema_of_paths(files, record_type, eng))
+    if columns is not None:
+        missing = [c for c in columns if c not in [f.name for f in schema.fields]
+                   and c not in part_cols]
+        if missing:
+            raise KeyError(f"columns not in schema: {missing}")
+        schema = StructType([f for f in schema.fields if f.name in set(columns)])
+        part_cols = [c for c in part_cols if c in set(columns)]
     data_schema = StructType([f for f in schema.fields if f.name not in part_cols])
 
     # Files needing host bytes (compressed, or CPU engine) are loaded and

@@ -239,6 +239,49 @@ class TestGpuInference:
         assert rows[3]["c"] == "s3"
 
 
+class TestGpuRobustness:
+    """Hostile inputs must raise cleanly — never crash a kernel or hang."""
+
+    def test_random_garbage_raises(self, tmp_path):
+        g = _gpu_engine()
+        rng = np.random.default_rng(0)
+        path = str(tmp_path / "garbage.tfrecord")
+        with open(path, "wb") as f:
+            f.write(rng.bytes(1 << 20))
+        schema = stf.StructType([stf.StructField("x", stf.LongType(), True)])
+        with pytest.raises(RuntimeError, match="corrupt TFRecord"):
+            g.read_file_to_batch_pipelined(path, schema, "Example")
+
+    def test_truncated_file_raises(self, tmp_path):
+        g = _gpu_engine()
+        batch = make_batch(500, seed=1)
+        img = cpu_engine.encode_batch(batch, "Example")
+        path = str(tmp_path / "trunc.tfrecord")
+        with open(path, "wb") as f:
+            f.write(img[:-7])  # chop mid-frame
+        with pytest.raises(RuntimeError, match="corrupt TFRecord"):
+            g.read_file_to_batch_pipelined(path, batch.schema, "Example")
+
+    def test_wrong_schema_kind_raises(self, tmp_path):
+        g = _gpu_engine()
+        batch = make_batch(100, seed=2)
+        path = str(tmp_path / "k.tfrecord")
+        with open(path, "wb") as f:
+            f.write(cpu_engine.encode_batch(batch, "Example"))
+        wrong = stf.StructType([stf.StructField("id", stf.StringType(), True)])
+        with pytest.raises(RuntimeError):
+            g.read_file_to_batch_pipelined(path, wrong, "Example")
+
+    def test_count_and_projection_gpu(self, tmp_sandbox):
+        out = str(tmp_sandbox / "cp")
+        stf.write_tfrecord({"a": np.arange(2000, dtype=np.int64),
+                            "b": np.arange(2000, dtype=np.float32)},
+                           out, engine="gpu", num_shards=2)
+        assert stf.count_tfrecord(out, engine="gpu") == 2000
+        df = stf.read_tfrecord(out, engine="gpu", columns=["b"])
+        assert df.columns == ["b"] and df.count() == 2000
+
+
 class TestGpuRandomized:
     """Randomized numerics sweep: many random batches, GPU kernels vs the
     plain host codec (the CPU reference implementation of the same ops)."""

@@ -370,3 +370,45 @@ class TestParallelGzip:
 
         raw = P.compress_bytes(b"", "gzip")
         assert _gz.decompress(raw) == b""
+
+
+class TestProjectionAndCount:
+    def test_column_projection(self, tmp_sandbox):
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+
+        out = str(tmp_sandbox / "proj")
+        stf.write_tfrecord({"a": np.arange(50, dtype=np.int64),
+                            "b": [f"s{i}" for i in range(50)],
+                            "c": np.arange(50, dtype=np.float32)},
+                           out, engine="cpu")
+        df = stf.read_tfrecord(out, engine="cpu", columns=["a", "c"])
+        assert sorted(df.columns) == ["a", "c"]
+        assert df.count() == 50
+        with __import__("pytest").raises(KeyError):
+            stf.read_tfrecord(out, engine="cpu", columns=["nope"])
+
+    def test_projection_keeps_partition_cols(self, tmp_sandbox):
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+
+        out = str(tmp_sandbox / "projp")
+        stf.write_tfrecord({"p": np.array([1, 2] * 10, np.int64),
+                            "x": np.arange(20, dtype=np.int64)},
+                           out, engine="cpu", partition_by=["p"])
+        df = stf.read_tfrecord(out, engine="cpu", columns=["x", "p"])
+        assert sorted(df.columns) == ["p", "x"]
+        df2 = stf.read_tfrecord(out, engine="cpu", columns=["x"])
+        assert df2.columns == ["x"]
+
+    def test_count(self, tmp_sandbox):
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+
+        out = str(tmp_sandbox / "cnt")
+        stf.write_tfrecord({"x": np.arange(123, dtype=np.int64)}, out,
+                           engine="cpu", num_shards=3)
+        assert stf.count_tfrecord(out, engine="cpu") == 123

@@ -258,9 +258,13 @@ py::list decode_records(py::buffer data, py::array_t<i64> rec_off,
   auto off = rec_off.unchecked<1>();
   auto len = rec_len.unchecked<1>();
 
-  // Pass A: structure scan -> per (record, field) stats.
+  // Pass A: structure scan -> per (record, field) stats. Pure C++ on raw
+  // pointers: the GIL is released so multi-file readers can decode on
+  // worker threads in parallel.
   std::vector<FieldStat> stats(static_cast<size_t>(R) * F);
-  for (i64 r = 0; r < R; ++r) {
+  {
+    py::gil_scoped_release rel;
+    for (i64 r = 0; r < R; ++r) {
     FieldStat* st = stats.data() + r * F;
     for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
     int32_t rc = scan_record(buf.data, off(r), len(r), fmt, schema, st);
@@ -278,6 +282,7 @@ py::list decode_records(py::buffer data, py::array_t<i64> rec_off,
         throw std::runtime_error(std::string("TFRecord decode failed in record ") +
                                  std::to_string(r) + ": " + err_name(st[f].err));
       }
+    }
   }
 
   // Prefix sums per field + allocation + pass B.
@@ -408,26 +413,31 @@ py::bytes encode_records(py::bytes schema_blob, int32_t fmt, py::list col_dicts,
       throw std::invalid_argument("column missing presence/row_off");
   }
 
-  // Pass A: payload sizes -> frame offsets.
-  std::vector<i64> frame_off(R + 1);
-  std::vector<i64> psize(R);
-  frame_off[0] = 0;
-  for (i64 r = 0; r < R; ++r) {
-    psize[r] = record_payload_size(cols.data(), schema, fmt, r);
-    frame_off[r + 1] = frame_off[r] + psize[r] + kFrameOverhead;
-  }
-  i64 total = frame_off[R];
+  // Both passes are pure C++ over raw pointers: run them GIL-free so shard
+  // encodes / decodes parallelize across reader/writer worker threads.
+  std::string out;
+  {
+    py::gil_scoped_release rel;
+    std::vector<i64> frame_off(R + 1);
+    std::vector<i64> psize(R);
+    frame_off[0] = 0;
+    for (i64 r = 0; r < R; ++r) {
+      psize[r] = record_payload_size(cols.data(), schema, fmt, r);
+      frame_off[r + 1] = frame_off[r] + psize[r] + kFrameOverhead;
+    }
+    i64 total = frame_off[R];
 
-  // Pass B: emit payloads + frames.
-  std::string out(static_cast<size_t>(total), '\0');
-  u8* file = reinterpret_cast<u8*>(out.data());
-  for (i64 r = 0; r < R; ++r) {
-    u8* o = file + frame_off[r] + 12;
-    u8* oe = emit_record_payload(o, cols.data(), schema, fmt, r);
-    if (oe - o != psize[r])
-      throw std::runtime_error("internal error: emit size mismatch (record " +
-                               std::to_string(r) + ")");
-    write_frame_header_footer(file, frame_off[r], psize[r], kCrcTables.t);
+    // Pass B: emit payloads + frames.
+    out.resize(static_cast<size_t>(total), '\0');
+    u8* file = reinterpret_cast<u8*>(out.data());
+    for (i64 r = 0; r < R; ++r) {
+      u8* o = file + frame_off[r] + 12;
+      u8* oe = emit_record_payload(o, cols.data(), schema, fmt, r);
+      if (oe - o != psize[r])
+        throw std::runtime_error("internal error: emit size mismatch (record " +
+                                 std::to_string(r) + ")");
+      write_frame_header_footer(file, frame_off[r], psize[r], kCrcTables.t);
+    }
   }
   return py::bytes(out);
 }

@@ -166,9 +166,17 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     window = 2 * workers
     futures: dict = {}
 
+    def _host_task(fpath: str):
+        data = _load_file(fpath)
+        if data.size == 0 or eng == "gpu":
+            return data  # GPU engine decodes on the device stream
+        # the native codec releases the GIL: files decode concurrently
+        return cpu_engine.decode_buffer(data, data_schema, record_type,
+                                        verify_crc=verify_crc)
+
     def _blob(i: int):
         if i not in futures:
-            futures[i] = pool.submit(_load_file, files[i])
+            futures[i] = pool.submit(_host_task, files[i])
         return futures.pop(i).result()
 
     # GPU path: consecutive uncompressed files are decoded as ONE pipeline
@@ -210,19 +218,20 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
                 continue
             for j in range(i, min(i + window, len(files))):
                 if _needs_host_bytes(files[j]) and j not in futures:
-                    futures[j] = pool.submit(_load_file, files[j])
+                    futures[j] = pool.submit(_host_task, files[j])
             metrics.add(files=1, nbytes=os.path.getsize(fpath))
-            data = _blob(i)
+            got = _blob(i)
             i += 1
-            if data.size == 0:
-                continue
             if eng == "gpu":
+                if got.size == 0:
+                    continue
                 from ..engine import gpu as gpu_engine
                 batch = gpu_engine.decode_buffer_to_cpu(
-                    data, data_schema, record_type, verify_crc=verify_crc)
+                    got, data_schema, record_type, verify_crc=verify_crc)
             else:
-                batch = cpu_engine.decode_buffer(data, data_schema, record_type,
-                                                 verify_crc=verify_crc)
+                if isinstance(got, np.ndarray):  # empty file sentinel
+                    continue
+                batch = got
             t = batch_to_table(batch)
             _append_with_parts(t, [fpath], [t.num_rows])
     finally:

@@ -106,38 +106,50 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
                       out_dir: str, codec: Optional[str], job_id: str,
                       num_shards: int, shard_offset: int, eng: str,
                       metrics: Optional[IOMetrics] = None):
-    """Encode `table` into `num_shards` part files under out_dir."""
+    """Encode `table` into `num_shards` part files under out_dir. The
+    compress+write of shard k runs on a worker thread while shard k+1
+    encodes (both the native encoder and zlib release the GIL)."""
+    from concurrent.futures import ThreadPoolExecutor
+
     R = table.num_rows
     bounds = np.linspace(0, R, num_shards + 1).astype(np.int64)
-    for s in range(num_shards):
-        lo, hi = int(bounds[s]), int(bounds[s + 1])
-        if num_shards > 1 and hi == lo:
-            continue
-        chunk = table.slice(lo, hi - lo)
-        batch = table_to_batch(chunk, schema)
-        fname = P.part_file_name(shard_offset + s, codec, job_id)
-        fpath = os.path.join(out_dir, fname)
-        # device DevCols holds at most 64 fields; wider schemas encode on host
-        if eng == "gpu" and len(batch.columns) <= 64:
-            from ..engine import gpu as gpu_engine
-            dev_batch = gpu_engine.batch_to_device(batch)
-            if codec is None:
-                # encode sliced + DMA straight into the (temp) file mapping
-                tmp = fpath + ".inprogress"
-                nbytes = gpu_engine.write_batch_to_file(dev_batch, tmp,
-                                                        record_type)
-                os.replace(tmp, fpath)
-                if metrics is not None:
-                    metrics.add(rows=hi - lo, nbytes=nbytes, files=1)
-                continue
-            raw = gpu_engine.device_to_bytes(
-                gpu_engine.encode_device(dev_batch, record_type))
-        else:
-            raw = cpu_engine.encode_batch(batch, record_type)
+
+    def _compress_write(raw: bytes, fpath: str, rows: int):
         payload = P.compress_bytes(raw, codec)
         P.write_file_atomic(payload, fpath)
         if metrics is not None:
-            metrics.add(rows=hi - lo, nbytes=len(payload), files=1)
+            metrics.add(rows=rows, nbytes=len(payload), files=1)
+
+    with ThreadPoolExecutor(max_workers=min(8, max(num_shards, 1))) as pool:
+        futs = []
+        for s in range(num_shards):
+            lo, hi = int(bounds[s]), int(bounds[s + 1])
+            if num_shards > 1 and hi == lo:
+                continue
+            chunk = table.slice(lo, hi - lo)
+            batch = table_to_batch(chunk, schema)
+            fname = P.part_file_name(shard_offset + s, codec, job_id)
+            fpath = os.path.join(out_dir, fname)
+            # device DevCols holds at most 64 fields; wider schemas encode on host
+            if eng == "gpu" and len(batch.columns) <= 64:
+                from ..engine import gpu as gpu_engine
+                dev_batch = gpu_engine.batch_to_device(batch)
+                if codec is None:
+                    # encode sliced + DMA straight into the (temp) file mapping
+                    tmp = fpath + ".inprogress"
+                    nbytes = gpu_engine.write_batch_to_file(dev_batch, tmp,
+                                                            record_type)
+                    os.replace(tmp, fpath)
+                    if metrics is not None:
+                        metrics.add(rows=hi - lo, nbytes=nbytes, files=1)
+                    continue
+                raw = gpu_engine.device_to_bytes(
+                    gpu_engine.encode_device(dev_batch, record_type))
+            else:
+                raw = cpu_engine.encode_batch(batch, record_type)
+            futs.append(pool.submit(_compress_write, raw, fpath, hi - lo))
+        for f in futs:
+            f.result()
 
 
 def write_tfrecord(data, path: str, record_type: str = "Example",

@@ -856,12 +856,33 @@ void drop_mapping_locked(const std::string& key) {
   g_mmap_cache.erase(it);
 }
 
+// Drop cache entries whose path no longer resolves to the mapped inode
+// (overwritten/renamed/deleted files): their pinned pages keep the DEAD
+// tmpfs file alive until eviction otherwise. One stat per entry, entries
+// are few.
+void sweep_stale_locked() {
+  for (auto it = g_mmap_cache.begin(); it != g_mmap_cache.end();) {
+    struct stat st {};
+    bool live = stat(it->first.c_str(), &st) == 0 &&
+                st.st_dev == it->second.dev && st.st_ino == it->second.ino &&
+                (size_t)st.st_size >= it->second.n;
+    if (live) {
+      ++it;
+    } else {
+      auto key = it->first;
+      ++it;
+      drop_mapping_locked(key);
+    }
+  }
+}
+
 // Returns (ptr, pinned). With writable=true the file is created/resized to n
 // first. pinned=false means hipHostRegister failed (fall back to the staged
 // pread/pwrite path); ptr is 0 iff n == 0.
 py::tuple file_mmap_pinned(const std::string& path, i64 n, bool writable) {
   if (n <= 0) return py::make_tuple((uintptr_t)0, true);
   std::lock_guard<std::mutex> lk(g_mmap_mu);
+  sweep_stale_locked();
   int flags = writable ? (O_RDWR | O_CREAT) : O_RDWR;
   int fd = ::open(path.c_str(), flags, 0644);
   bool rdonly = false;

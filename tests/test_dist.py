@@ -14,11 +14,12 @@ import torch.multiprocessing as mp
 WORLD = 2
 
 
-def _run(fn, tmp, extra=None):
+def _run(fn, tmp, extra=None, world=WORLD):
     ctx = mp.get_context("spawn")
     port = 29600 + (os.getpid() % 500)
-    procs = [ctx.Process(target=_entry, args=(fn.__name__, r, port, str(tmp), extra))
-             for r in range(WORLD)]
+    procs = [ctx.Process(target=_entry,
+                         args=(fn.__name__, r, port, str(tmp), extra, world))
+             for r in range(world)]
     for p in procs:
         p.start()
     for p in procs:
@@ -27,18 +28,18 @@ def _run(fn, tmp, extra=None):
     for p in procs:
         if p.is_alive():
             p.terminate()
-    assert codes == [0, 0], f"worker exit codes: {codes}"
+    assert codes == [0] * world, f"worker exit codes: {codes}"
 
 
-def _entry(fn_name, rank, port, tmp, extra):
+def _entry(fn_name, rank, port, tmp, extra, world=WORLD):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
-    os.environ["WORLD_SIZE"] = str(WORLD)
+    os.environ["WORLD_SIZE"] = str(world)
     os.environ["RANK"] = str(rank)
     os.environ["TFREC_FORCE_CPU"] = "1"
     import torch.distributed as dist
 
-    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
         globals()[fn_name](rank, tmp, extra)
         dist.barrier()
@@ -158,3 +159,34 @@ def test_distributed_schema_inference(tmp_path):
 @pytest.mark.timeout(240)
 def test_distributed_save_mode_ignore(tmp_path):
     _run(_w_save_mode_ignore, tmp_path)
+
+
+def _w_partitioned_any_world(rank, tmp, extra):
+    import torch.distributed as dist
+
+    from spark_tfrecord_amd.parallel import write_tfrecord_distributed
+
+    world = dist.get_world_size()
+    out = os.path.join(tmp, "parts_any")
+    data = {"date": [f"d{i % 3}" for i in range(6)],
+            "x": np.arange(6, dtype=np.int64) + 100 * rank}
+    write_tfrecord_distributed(data, out, partition_by=["date"],
+                               mode="overwrite")
+    dist.barrier()
+    if rank == 0:
+        import spark_tfrecord_amd as stf
+
+        assert sorted(d for d in os.listdir(out) if d.startswith("date=")) == \
+            ["date=d0", "date=d1", "date=d2"]
+        for d in ["date=d0", "date=d1", "date=d2"]:
+            # exactly one owner rank wrote each partition directory
+            assert len(os.listdir(os.path.join(out, d))) == 1
+        df = stf.read_tfrecord(out)
+        assert df.count() == 6 * world
+        xs = sorted(r["x"] for r in df.collect())
+        want = sorted(i + 100 * r for r in range(world) for i in range(6))
+        assert xs == want
+
+
+def test_distributed_partitioned_write_world4(tmp_path):
+    _run(_w_partitioned_any_world, tmp_path, world=4)

@@ -842,7 +842,7 @@ struct MappedFile {
 std::map<std::string, MappedFile> g_mmap_cache;
 std::mutex g_mmap_mu;
 size_t g_mmap_bytes = 0;
-constexpr size_t kMmapCacheCap = 8ull << 30;  // 8 GiB of cached mappings
+constexpr size_t kMmapCacheCap = 4ull << 30;  // 4 GiB of cached mappings (8 ranks share a host)
 
 void drop_mapping_locked(const std::string& key) {
   auto it = g_mmap_cache.find(key);

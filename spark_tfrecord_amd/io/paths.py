@@ -152,9 +152,14 @@ def _is_data_file(name: str) -> bool:
     return not (base.startswith("_") or base.startswith("."))
 
 
-def list_data_files(path: str) -> List[str]:
-    """Resolve a path/glob/dir into a sorted list of data files, recursing
-    into partition directories."""
+def list_data_files(path) -> List[str]:
+    """Resolve a path/glob/dir — or a list of them — into a sorted list of
+    data files, recursing into partition directories."""
+    if isinstance(path, (list, tuple)):
+        out: List[str] = []
+        for p in path:
+            out.extend(list_data_files(p))
+        return sorted(out)
     paths: List[str] = []
     candidates = _glob.glob(path) if _glob.has_magic(path) else [path]
     if _glob.has_magic(path) and not candidates:

@@ -133,7 +133,11 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     if not files:
         raise FileNotFoundError(f"No TFRecord files found under {path}")
     if base_dir is None:
-        base_dir = path if os.path.isdir(path) else os.path.dirname(path)
+        if isinstance(path, (list, tuple)):
+            base_dir = (os.path.commonpath([os.path.dirname(f) for f in files])
+                        if files else "")
+        else:
+            base_dir = path if os.path.isdir(path) else os.path.dirname(path)
     part_cols = _partition_schema(files, base_dir) if base_dir else []
     eng = engine_mod.resolve_engine(engine)
 

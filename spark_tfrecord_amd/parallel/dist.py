@@ -359,11 +359,9 @@ def read_tfrecord_distributed(path: str, schema: Optional[StructType] = None,
                          schema=schema_to_arrow(schema))
         return DataFrame(empty, schema)
     root = path if os.path.isdir(path) else os.path.dirname(path)
+    # ONE call over the rank's whole file list: the GPU reader batches
+    # consecutive uncompressed files into a single scan+decode pipeline;
     # partition-column discovery stays relative to the dataset root
-    frames = [read_tfrecord(f, schema=schema, record_type=record_type,
-                            engine=engine, verify_crc=verify_crc, base_dir=root)
-              for f in my_files]
-    import pyarrow as pa
-    table = pa.concat_tables([d.to_arrow_table() for d in frames],
-                             promote_options="permissive")
-    return DataFrame(table, frames[0].schema)
+    df = read_tfrecord(my_files, schema=schema, record_type=record_type,
+                       engine=engine, verify_crc=verify_crc, base_dir=root)
+    return df

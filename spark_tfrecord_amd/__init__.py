@@ -16,6 +16,7 @@ import torch  # noqa: F401  (intentional import order)
 from .api import DataFrame, DataFrameReader, DataFrameWriter, TFRecordSession, session
 from .io.reader import count_tfrecord, read_tfrecord
 from .io.writer import write_tfrecord
+from .io.stream_writer import ShardWriter
 from .io.validate import validate_tfrecord
 from . import torch_data
 from .schema import (
@@ -38,7 +39,7 @@ __version__ = "0.1.0"
 __all__ = [
     "session", "DataFrame", "DataFrameReader", "DataFrameWriter",
     "TFRecordSession", "read_tfrecord", "write_tfrecord", "validate_tfrecord",
-    "count_tfrecord",
+    "count_tfrecord", "ShardWriter",
     "DataType", "NullType", "IntegerType", "LongType", "FloatType",
     "DoubleType", "DecimalType", "StringType", "BinaryType", "ArrayType",
     "StructField", "StructType",

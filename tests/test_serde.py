@@ -304,3 +304,43 @@ class TestArrowNestedFastPath:
         col = arrow_to_wire(arr, dt, True, "rag")
         np.testing.assert_array_equal(col.presence, [1, 0, 1])
         np.testing.assert_array_equal(np.asarray(col.row_off), [0, 1, 1, 3])
+
+
+class TestFullTypeMatrix:
+    """Every type combination the reference's README advertises (scalars,
+    arrays, arrays-of-arrays of Integer/Long/Float/Double/Decimal/String/
+    Binary) round-trips. Spark ML VectorType has no non-JVM analog; its
+    ArrayType(DoubleType) representation is what this covers."""
+
+    def test_all_advertised_combinations(self):
+        from decimal import Decimal
+
+        import numpy as np
+
+        from spark_tfrecord_amd.columnar import RecordBatch, column_from_values
+        from spark_tfrecord_amd.engine import cpu as cpu_engine
+
+        cases = [
+            (stf.ArrayType(stf.IntegerType()), [[1, 2], [], None, [5]]),
+            (stf.ArrayType(stf.DoubleType()), [[1.5, 2.5], None, [0.25]]),
+            (stf.ArrayType(stf.DecimalType()),
+             [[Decimal("1.5")], [Decimal("2.25"), Decimal("0.5")]]),
+            (stf.ArrayType(stf.BinaryType()), [[b"ab", b""], [b"c"]]),
+            (stf.ArrayType(stf.ArrayType(stf.IntegerType())), [[[1], [2, 3]], []]),
+            (stf.ArrayType(stf.ArrayType(stf.DoubleType())),
+             [[[1.5]], [[2.5, 3.5], []]]),
+            (stf.ArrayType(stf.ArrayType(stf.StringType())),
+             [[["a", "bb"]], [["c"], []]]),
+            (stf.ArrayType(stf.ArrayType(stf.BinaryType())),
+             [[[b"x"]], [[b"yy", b"z"]]]),
+        ]
+        for dt, vals in cases:
+            schema = stf.StructType([stf.StructField("c", dt, True)])
+            col = column_from_values(vals, dt, True, "c")
+            batch = RecordBatch(schema, [col], len(vals))
+            rt = ("SequenceExample"
+                  if isinstance(dt.elementType, stf.ArrayType) else "Example")
+            img = cpu_engine.encode_batch(batch, rt)
+            out = cpu_engine.decode_buffer(np.frombuffer(img, np.uint8), schema, rt)
+            np.testing.assert_array_equal(np.asarray(col.values),
+                                          np.asarray(out.columns[0].values))

@@ -437,7 +437,6 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     # CRC pass was tried and reverted: every host<->device copy on this
     # stack is a blit KERNEL, so "free" overlap doesn't exist and the fused
     # form has ~1 ms less total GPU work per 215 MB.)
-    main = torch.cuda.current_stream()
     crc_err = None
     if verify_crc:
         crc_err = torch.full((1,), -1, dtype=torch.int64, device=device)

@@ -1221,6 +1221,213 @@ TFR_HOSTDEV inline u8* emit_record_payload(u8* o, const FieldColumn* cols,
 // at an absolute offset. Payload must already be in place at frame_off + 12.
 // ---------------------------------------------------------------------------
 
+// ---------------------------------------------------------------------------
+// Fused emit + CRC (write cursor): the mirror image of ScanCur. Output bytes
+// accumulate in an 8-byte register window; each full window is stored ONCE
+// (one unaligned u64 store instead of eight byte stores) and folded into the
+// running payload CRC32C — the separate frame-CRC step used to re-read the
+// whole emitted payload. Byte-exactness vs the plain emitters is guaranteed
+// by construction (same byte sequence, different batching) and covered by
+// the golden-vector / protobuf-interop / GPU-vs-host equality tests.
+// ---------------------------------------------------------------------------
+
+struct WriteCur {
+  u8* base;   // payload destination
+  i64 pos;    // bytes emitted
+  i64 wpos;   // bytes stored + CRC'd (multiple of 8)
+  u64 win;    // pending low (pos - wpos) bytes
+  u32 crc;    // running ~crc
+  const uint32_t (*tab)[256];
+};
+
+TFR_HOSTDEV inline void wcur_init(WriteCur& w, u8* base,
+                                  const uint32_t (*tab)[256]) {
+  w.base = base;
+  w.pos = 0;
+  w.wpos = 0;
+  w.win = 0;
+  w.crc = 0xFFFFFFFFu;
+  w.tab = tab;
+}
+
+TFR_HOSTDEV inline void wcur_flush8(WriteCur& w) {
+  __builtin_memcpy(w.base + w.wpos, &w.win, 8);  // little-endian everywhere
+  w.crc = crc32c_step8(w.crc, w.win, w.tab);
+  w.wpos += 8;
+  w.win = 0;
+}
+
+TFR_HOSTDEV inline void wcur_put(WriteCur& w, u8 b) {
+  w.win |= (u64)b << (8 * (w.pos - w.wpos));
+  if (++w.pos - w.wpos == 8) wcur_flush8(w);
+}
+
+TFR_HOSTDEV inline void wcur_varint(WriteCur& w, u64 v) {
+  while (v >= 0x80) {
+    wcur_put(w, (u8)(v | 0x80));
+    v >>= 7;
+  }
+  wcur_put(w, (u8)v);
+}
+
+TFR_HOSTDEV inline void wcur_bytes(WriteCur& w, const u8* src, i64 n) {
+  while (n > 0 && (w.pos - w.wpos) != 0) {  // fill to the window boundary
+    wcur_put(w, *src++);
+    --n;
+  }
+  while (n >= 8) {  // window empty: stream whole words through it
+    __builtin_memcpy(&w.win, src, 8);
+    w.pos += 8;
+    wcur_flush8(w);
+    src += 8;
+    n -= 8;
+  }
+  while (n-- > 0) wcur_put(w, *src++);
+}
+
+// Stores the partial tail window and returns the final payload CRC32C.
+TFR_HOSTDEV inline u32 wcur_finish(WriteCur& w) {
+  int rem = (int)(w.pos - w.wpos);
+  if (!rem) return ~w.crc;
+  __builtin_memcpy(w.base + w.wpos, &w.win, (size_t)rem);
+  u32 crc = w.crc;
+  for (int i = 0; i < rem; ++i)
+    crc = w.tab[0][(crc ^ (u8)(w.win >> (8 * i))) & 0xFF] ^ (crc >> 8);
+  return ~crc;
+}
+
+TFR_HOSTDEV inline void wcur_list_body(WriteCur& w, const FieldColumn& c,
+                                       i64 v0, i64 v1) {
+  if (c.kind == KIND_FLOAT) {
+    i64 n = v1 - v0;
+    if (n) {
+      wcur_put(w, 0x0A);  // field 1, wiretype 2 (packed)
+      wcur_varint(w, (u64)(4 * n));
+      wcur_bytes(w, (const u8*)(c.f32_vals + v0), 4 * n);
+    }
+  } else if (c.kind == KIND_INT64) {
+    i64 packed = 0;
+    for (i64 v = v0; v < v1; ++v)
+      packed += varint_size((u64)c.i64_vals[v]);
+    if (packed) {
+      wcur_put(w, 0x0A);
+      wcur_varint(w, (u64)packed);
+      for (i64 v = v0; v < v1; ++v) wcur_varint(w, (u64)c.i64_vals[v]);
+    }
+  } else {
+    for (i64 v = v0; v < v1; ++v) {
+      i64 b0 = c.elem_off[v];
+      i64 blen = c.elem_off[v + 1] - b0;
+      wcur_put(w, 0x0A);
+      wcur_varint(w, (u64)blen);
+      wcur_bytes(w, c.bytes_data + b0, blen);
+    }
+  }
+}
+
+TFR_HOSTDEV inline void wcur_feature_body(WriteCur& w, const FieldColumn& c,
+                                          i64 v0, i64 v1) {
+  i64 body = list_body_size(c, v0, v1);
+  wcur_put(w, (u8)((c.kind << 3) | 2));
+  wcur_varint(w, (u64)body);
+  wcur_list_body(w, c, v0, v1);
+}
+
+TFR_HOSTDEV inline void wcur_features_entry(WriteCur& w, const FieldColumn& c,
+                                            const SchemaView& s, int f, i64 v0,
+                                            i64 v1) {
+  i64 fb = feature_body_size(c, v0, v1);
+  i64 klen = s.name_len(f);
+  wcur_put(w, 0x0A);  // key
+  wcur_varint(w, (u64)klen);
+  wcur_bytes(w, s.name(f), klen);
+  wcur_put(w, 0x12);  // value (Feature)
+  wcur_varint(w, (u64)fb);
+  wcur_feature_body(w, c, v0, v1);
+}
+
+TFR_HOSTDEV inline void wcur_feature_list_body(WriteCur& w,
+                                               const FieldColumn& c, i64 r) {
+  for (i64 j = c.list_off[r]; j < c.list_off[r + 1]; ++j) {
+    i64 fb = feature_body_size(c, c.sub_off[j], c.sub_off[j + 1]);
+    wcur_put(w, 0x0A);  // FeatureList.feature
+    wcur_varint(w, (u64)fb);
+    wcur_feature_body(w, c, c.sub_off[j], c.sub_off[j + 1]);
+  }
+}
+
+TFR_HOSTDEV inline void wcur_feature_lists_entry(WriteCur& w,
+                                                 const FieldColumn& c,
+                                                 const SchemaView& s, int f,
+                                                 i64 r) {
+  i64 flb = feature_list_body_size(c, r);
+  i64 klen = s.name_len(f);
+  wcur_put(w, 0x0A);
+  wcur_varint(w, (u64)klen);
+  wcur_bytes(w, s.name(f), klen);
+  wcur_put(w, 0x12);
+  wcur_varint(w, (u64)flb);
+  wcur_feature_list_body(w, c, r);
+}
+
+// Emits one record payload at `o`, returning its length; *crc_out gets the
+// payload CRC32C accumulated from the same register windows the stores used.
+TFR_HOSTDEV inline i64 emit_record_payload_fused(u8* o, const FieldColumn* cols,
+                                                 const SchemaView& s, int32_t fmt,
+                                                 i64 r, u32* crc_out,
+                                                 const uint32_t (*tab)[256]) {
+  i64 ctx_body = 0, fl_body = 0;
+  for (int f = 0; f < s.nfields; ++f) {
+    const FieldColumn& c = cols[f];
+    if (!c.presence[r]) continue;
+    if (!c.is_seq) {
+      i64 e = features_entry_size(c, s, f, c.row_off[r], c.row_off[r + 1]);
+      ctx_body += 1 + varint_size((u64)e) + e;
+    } else {
+      i64 e = feature_lists_entry_size(c, s, f, r);
+      fl_body += 1 + varint_size((u64)e) + e;
+    }
+  }
+  WriteCur w;
+  wcur_init(w, o, tab);
+  auto emit_ctx = [&]() {
+    for (int f = 0; f < s.nfields; ++f) {
+      const FieldColumn& c = cols[f];
+      if (!c.presence[r] || c.is_seq) continue;
+      i64 e = features_entry_size(c, s, f, c.row_off[r], c.row_off[r + 1]);
+      wcur_put(w, 0x0A);  // Features.feature map entry
+      wcur_varint(w, (u64)e);
+      wcur_features_entry(w, c, s, f, c.row_off[r], c.row_off[r + 1]);
+    }
+  };
+  auto emit_fl = [&]() {
+    for (int f = 0; f < s.nfields; ++f) {
+      const FieldColumn& c = cols[f];
+      if (!c.presence[r] || !c.is_seq) continue;
+      i64 e = feature_lists_entry_size(c, s, f, r);
+      wcur_put(w, 0x0A);  // FeatureLists.feature_list map entry
+      wcur_varint(w, (u64)e);
+      wcur_feature_lists_entry(w, c, s, f, r);
+    }
+  };
+  if (fmt == FMT_EXAMPLE) {
+    wcur_put(w, 0x0A);  // Example.features
+    wcur_varint(w, (u64)ctx_body);
+    emit_ctx();
+  } else {
+    if (ctx_body) {
+      wcur_put(w, 0x0A);  // SequenceExample.context
+      wcur_varint(w, (u64)ctx_body);
+      emit_ctx();
+    }
+    wcur_put(w, 0x12);  // SequenceExample.feature_lists
+    wcur_varint(w, (u64)fl_body);
+    emit_fl();
+  }
+  *crc_out = wcur_finish(w);
+  return w.pos;
+}
+
 constexpr i64 kFrameOverhead = 16;  // 8 len + 4 crc + 4 crc
 
 TFR_HOSTDEV inline void write_frame_header_footer(u8* file, i64 frame_off,
@@ -1232,6 +1439,21 @@ TFR_HOSTDEV inline void write_frame_header_footer(u8* file, i64 frame_off,
   u32 lc = mask_crc(crc32c_sw(h, 8, 0, tab));
   __builtin_memcpy(h + 8, &lc, 4);
   u32 dc = mask_crc(crc32c_sw(h + 12, static_cast<size_t>(payload_len), 0, tab));
+  __builtin_memcpy(h + 12 + payload_len, &dc, 4);
+}
+
+// Variant for the fused emitter: the payload CRC is already known, so only
+// the 8 header bytes are (re)read for their CRC — no payload re-read.
+TFR_HOSTDEV inline void write_frame_header_footer_crc(u8* file, i64 frame_off,
+                                                      i64 payload_len,
+                                                      u32 payload_crc,
+                                                      const u32 (*tab)[256]) {
+  u8* h = file + frame_off;
+  u64 len_le = static_cast<u64>(payload_len);
+  __builtin_memcpy(h, &len_le, 8);
+  u32 lc = mask_crc(crc32c_sw(h, 8, 0, tab));
+  __builtin_memcpy(h + 8, &lc, 4);
+  u32 dc = mask_crc(payload_crc);
   __builtin_memcpy(h + 12 + payload_len, &dc, 4);
 }
 

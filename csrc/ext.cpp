@@ -427,16 +427,20 @@ py::bytes encode_records(py::bytes schema_blob, int32_t fmt, py::list col_dicts,
     }
     i64 total = frame_off[R];
 
-    // Pass B: emit payloads + frames.
+    // Pass B: fused emit (each output window stored once, CRC from the same
+    // registers — codec_core.h WriteCur; identical bytes to the plain form).
     out.resize(static_cast<size_t>(total), '\0');
     u8* file = reinterpret_cast<u8*>(out.data());
     for (i64 r = 0; r < R; ++r) {
       u8* o = file + frame_off[r] + 12;
-      u8* oe = emit_record_payload(o, cols.data(), schema, fmt, r);
-      if (oe - o != psize[r])
+      u32 crc = 0;
+      i64 emitted = emit_record_payload_fused(o, cols.data(), schema, fmt, r,
+                                              &crc, kCrcTables.t);
+      if (emitted != psize[r])
         throw std::runtime_error("internal error: emit size mismatch (record " +
                                  std::to_string(r) + ")");
-      write_frame_header_footer(file, frame_off[r], psize[r], kCrcTables.t);
+      write_frame_header_footer_crc(file, frame_off[r], psize[r], crc,
+                                    kCrcTables.t);
     }
   }
   return py::bytes(out);

@@ -212,9 +212,11 @@ __global__ void emit_records_kernel(const DevCols* __restrict__ cols,
        r += (i64)gridDim.x * blockDim.x) {
     i64 payload = (frame_off[r + 1] - frame_off[r]) - kFrameOverhead;
     u8* o = file + frame_off[r] + 12;
-    u8* oe = emit_record_payload(o, cols->c, schema, fmt, r);
-    if (oe - o != payload) err[0] = ERR_OVERFLOW;
-    write_frame_header_footer(file, frame_off[r], payload, tab);
+    u32 crc = 0;
+    i64 emitted = emit_record_payload_fused(o, cols->c, schema, fmt, r, &crc,
+                                            tab);
+    if (emitted != payload) err[0] = ERR_OVERFLOW;
+    write_frame_header_footer_crc(file, frame_off[r], payload, crc, tab);
   }
 }
 

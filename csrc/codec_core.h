@@ -715,6 +715,7 @@ TFR_HOSTDEV inline int32_t cur_scan_feature_body(ScanCur& c, i64 end,
 TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_rel,
                                                   const SchemaView& schema,
                                                   int want_seq, FieldStat* stats) {
+  i64 body_start = c.pos;
   while (c.pos < end) {
     u64 tag;
     if (!cur_varint(c, end, &tag)) return ERR_BAD_VARINT;
@@ -730,6 +731,7 @@ TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_
     i64 ee = c.pos + (i64)entry_len;
     int f = -1;
     bool key_seen = false;
+    bool val_seen = false;
     while (c.pos < ee) {
       u64 etag;
       if (!cur_varint(c, ee, &etag)) return ERR_BAD_VARINT;
@@ -743,6 +745,7 @@ TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_
         cur_advance(c, (i64)klen);
       } else if (efn == 2 && ewt == 2) {  // value (Feature / FeatureList)
         if (!key_seen) return ERR_RETRY_UNFUSED;  // value before key: bail
+        val_seen = true;
         u64 vlen;
         if (!cur_varint(c, ee, &vlen)) return ERR_BAD_VARINT;
         if ((u64)(ee - c.pos) < vlen) return ERR_TRUNCATED;
@@ -800,6 +803,14 @@ TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_
         int32_t rc = cur_skip_field(c, ee, ewt);
         if (rc != ERR_OK) return rc;
       }
+    }
+    if (f >= 0 && !val_seen) {
+      // key with no value field: protobuf map semantics give the default
+      // (empty) Feature — the feature IS present (two-pass form behavior:
+      // scan_features_body sets pos = features-body start, len 0)
+      FieldStat* st = &stats[f];
+      st->pos = data_rel + body_start;
+      st->len = 0;
     }
   }
   return ERR_OK;

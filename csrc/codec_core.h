@@ -737,6 +737,9 @@ TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_
       if (!cur_varint(c, ee, &etag)) return ERR_BAD_VARINT;
       u32 efn = (u32)(etag >> 3), ewt = (u32)(etag & 7);
       if (efn == 1 && ewt == 2) {  // key
+        // a key AFTER the value would override it (protobuf last-field-wins)
+        // but the value was already consumed forward-only: redo two-pass
+        if (val_seen) return ERR_RETRY_UNFUSED;
         u64 klen;
         if (!cur_varint(c, ee, &klen)) return ERR_BAD_VARINT;
         if ((u64)(ee - c.pos) < klen) return ERR_TRUNCATED;

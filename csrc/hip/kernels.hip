@@ -1,16 +1,16 @@
 // gfx950 (CDNA4) kernels for the TFRecord codec + pybind11 launch wrappers.
 //
-// Design (SURVEY.md §2b native inventory, §7 step 3):
+// Design (SURVEY.md §2b native inventory, §7 step 3; docs/KERNELS.md):
 //  - Records are the parallel axis: one record per lane, grid-stride loops
 //    sized so a full file saturates 256 CUs (wave = 64 lanes; blocks of 256).
-//  - Frame boundaries come from a cheap sequential host header scan
-//    (csrc/ext.cpp scan_frame_headers); the GPU does everything per-record:
-//    masked-CRC32C verification, protobuf structure scan, value extraction,
-//    size/emit encode, and frame construction.
-//  - CRC32C slicing-by-8 tables (8 KiB) are staged in LDS per workgroup;
-//    byte-level parse state stays in registers.
-//  - Prefix sums between passes run as torch.cumsum on the same stream
-//    (Python orchestration in spark_tfrecord_amd/engine/gpu.py).
+//  - Frame boundaries are discovered ON DEVICE: every byte position is
+//    CRC-tested as a candidate head by a two-pass (count/emit) kernel whose
+//    output is position-sorted by construction.
+//  - The structure scan and the emit both FUSE the payload CRC32C into the
+//    same register windows their loads/stores use (codec_core.h ScanCur /
+//    WriteCur); CRC32C slicing-by-8 tables (8 KiB) are staged in LDS.
+//  - Prefix sums between passes are rocprim decoupled-lookback scans with
+//    strided input iterators (Python orchestration in engine/gpu.py).
 //
 // All parsing/emit logic is shared with the host via csrc/codec_core.h — the
 // kernels only add the parallel decomposition and memory staging.

@@ -1,13 +1,14 @@
 """GPU engine: gfx950 kernel pipeline orchestration.
 
-Decode: host header scan -> H2D -> [crc_verify | scan_records] -> torch
+Decode: sliced H2D (file-mapping DMA) overlapped with the two-pass frame
+scan -> chain validation -> fused structure-scan+CRC -> rocprim strided
 prefix sums -> extract_fields -> device wire-form columns.
-Encode: device wire-form -> size_records -> prefix sum -> emit + frame CRC
--> device file image (D2H only at the file boundary).
+Encode: device wire-form -> size_records -> prefix sum -> record-range
+sliced fused emit+CRC overlapped with the D2H DMA into the file mapping.
 
-All launches go on the torch current stream, so the cumsum/cast glue and the
-kernels form one in-order pipeline; prefix sums over per-record counts are
-torch ops on the same stream.
+All kernels launch on the torch current stream, forming one in-order
+pipeline; file DMA runs on two side streams gated by events (SURVEY.md §7
+step 3). See docs/KERNELS.md for per-kernel design notes and measurements.
 """
 
 from __future__ import annotations

@@ -63,13 +63,26 @@ class IOPool {
 
   template <typename Fn>
   int run_spans(i64 n, Fn&& fn) {
+    return run_parts(n, 64 << 10, std::forward<Fn>(fn));
+  }
+
+  // Splits [0, n) into per-worker ranges of at least min_span and runs
+  // fn(offset, count) on the pool; fn returns 0 or an error code (first
+  // nonzero wins). Used for byte IO spans AND record ranges.
+  template <typename Fn>
+  int run_parts(i64 n, i64 min_span, Fn&& fn) {
     if (n <= 0) return 0;
     int nspan = (int)nthreads_;
     i64 span = (n + nspan - 1) / nspan;
-    if (span < (64 << 10)) {  // don't split tiny transfers
-      nspan = 1;
-      span = n;
+    if (span < min_span) {
+      nspan = (int)((n + min_span - 1) / min_span);
+      if (nspan < 1) nspan = 1;
+      span = (n + nspan - 1) / nspan;
     }
+    if (nspan == 1) return fn((i64)0, n);  // run inline, skip the pool
+    if (getenv("TFREC_POOL_DEBUG"))
+      fprintf(stderr, "[pool] n=%lld nspan=%d span=%lld\n", (long long)n,
+              nspan, (long long)span);
     std::atomic<int> err{0};
     std::atomic<int> left{nspan};
     {
@@ -263,25 +276,43 @@ py::list decode_records(py::buffer data, py::array_t<i64> rec_off,
   // worker threads in parallel.
   std::vector<FieldStat> stats(static_cast<size_t>(R) * F);
   {
-    py::gil_scoped_release rel;
-    for (i64 r = 0; r < R; ++r) {
-    FieldStat* st = stats.data() + r * F;
-    for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
-    int32_t rc = scan_record(buf.data, off(r), len(r), fmt, schema, st);
-    if (rc != ERR_OK)
-      throw std::runtime_error(std::string("TFRecord decode failed in record ") +
-                               std::to_string(r) + ": " + err_name(rc));
-    for (int f = 0; f < F; ++f)
-      if (st[f].err == ERR_KIND_MISMATCH) {
+    std::atomic<i64> bad_rec{-1};
+    std::atomic<int32_t> bad_rc{0};
+    std::atomic<int> bad_field{-1};
+    {
+      py::gil_scoped_release rel;
+      IOPool::instance().run_parts(R, 4096, [&](i64 o, i64 m) {
+        for (i64 r = o; r < o + m; ++r) {
+          FieldStat* st = stats.data() + r * F;
+          for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+          int32_t rc = scan_record(buf.data, off(r), len(r), fmt, schema, st);
+          for (int f = 0; rc == ERR_OK && f < F; ++f)
+            if (st[f].err != ERR_OK) {
+              rc = st[f].err;
+              bad_field.store(f);
+            }
+          if (rc != ERR_OK) {
+            bad_rec.store(r);
+            bad_rc.store(rc);
+            return 1;
+          }
+        }
+        return 0;
+      });
+    }
+    if (bad_rec.load() >= 0) {
+      i64 r = bad_rec.load();
+      int32_t rc = bad_rc.load();
+      if (rc == ERR_KIND_MISMATCH && bad_field.load() >= 0) {
+        int f = bad_field.load();
         std::string nm(reinterpret_cast<const char*>(schema.name(f)),
                        schema.name_len(f));
         throw std::runtime_error("Feature '" + nm +
                                  "' kind does not match requested data type (record " +
                                  std::to_string(r) + ")");
-      } else if (st[f].err != ERR_OK) {
-        throw std::runtime_error(std::string("TFRecord decode failed in record ") +
-                                 std::to_string(r) + ": " + err_name(st[f].err));
       }
+      throw std::runtime_error(std::string("TFRecord decode failed in record ") +
+                               std::to_string(r) + ": " + err_name(rc));
     }
   }
 
@@ -331,17 +362,42 @@ py::list decode_records(py::buffer data, py::array_t<i64> rec_off,
       dst.sub_count = sub_count.mutable_data();
     }
 
-    i64 val_base = 0, byte_base = 0, list_base = 0;
-    for (i64 r = 0; r < R; ++r) {
-      const FieldStat& st = stats[r * F + f];
-      int32_t rc = extract_field(buf.data, st.pos, st.len, fd.kind, fd.is_seq, dst,
-                                 val_base, byte_base, list_base);
-      if (rc != ERR_OK)
-        throw std::runtime_error(std::string("TFRecord decode failed in record ") +
-                                 std::to_string(r) + ": " + err_name(rc));
-      val_base += st.nvals;
-      byte_base += st.nbytes;
-      list_base += st.nlists;
+    // Extraction is record-independent once the destination bases are known:
+    // prefix the byte/list bases (row_off already carries the value bases),
+    // release the GIL and fan the record ranges across the worker pool.
+    {
+      std::vector<i64> byte_base_v(R);
+      std::vector<i64> list_base_v(fd.is_seq ? R : 0);
+      i64 bb = 0, lb = 0;
+      for (i64 r = 0; r < R; ++r) {
+        const FieldStat& st = stats[r * F + f];
+        byte_base_v[r] = bb;
+        bb += st.nbytes;
+        if (fd.is_seq) {
+          list_base_v[r] = lb;
+          lb += st.nlists;
+        }
+      }
+      const FieldStat* stp = stats.data();
+      const u8* datap = buf.data;
+      int rc;
+      {
+        py::gil_scoped_release rel;
+        rc = IOPool::instance().run_parts(R, 4096, [&](i64 o, i64 m) {
+          for (i64 r = o; r < o + m; ++r) {
+            const FieldStat& st = stp[r * F + f];
+            int32_t e = extract_field(datap, st.pos, st.len, fd.kind,
+                                      fd.is_seq, dst, row_off_p[r],
+                                      byte_base_v[r],
+                                      fd.is_seq ? list_base_v[r] : 0);
+            if (e != ERR_OK) return (int)-e;
+          }
+          return 0;
+        });
+      }
+      if (rc != 0)
+        throw std::runtime_error(std::string("TFRecord decode failed: ") +
+                                 err_name((int32_t)-rc));
     }
 
     py::dict d;
@@ -413,36 +469,46 @@ py::bytes encode_records(py::bytes schema_blob, int32_t fmt, py::list col_dicts,
       throw std::invalid_argument("column missing presence/row_off");
   }
 
-  // Both passes are pure C++ over raw pointers: run them GIL-free so shard
-  // encodes / decodes parallelize across reader/writer worker threads.
+  // Both passes are pure C++ over raw pointers: the GIL is released and the
+  // records parallelize across the worker pool (sizes and emits are
+  // record-independent; only the frame-offset prefix sum is sequential).
   std::string out;
+  bool emit_mismatch = false;
   {
     py::gil_scoped_release rel;
     std::vector<i64> frame_off(R + 1);
     std::vector<i64> psize(R);
+    const FieldColumn* cp = cols.data();
+    IOPool::instance().run_parts(R, 4096, [&](i64 o, i64 m) {
+      for (i64 r = o; r < o + m; ++r)
+        psize[r] = record_payload_size(cp, schema, fmt, r);
+      return 0;
+    });
     frame_off[0] = 0;
-    for (i64 r = 0; r < R; ++r) {
-      psize[r] = record_payload_size(cols.data(), schema, fmt, r);
+    for (i64 r = 0; r < R; ++r)
       frame_off[r + 1] = frame_off[r] + psize[r] + kFrameOverhead;
-    }
     i64 total = frame_off[R];
 
     // Pass B: fused emit (each output window stored once, CRC from the same
     // registers — codec_core.h WriteCur; identical bytes to the plain form).
     out.resize(static_cast<size_t>(total), '\0');
     u8* file = reinterpret_cast<u8*>(out.data());
-    for (i64 r = 0; r < R; ++r) {
-      u8* o = file + frame_off[r] + 12;
-      u32 crc = 0;
-      i64 emitted = emit_record_payload_fused(o, cols.data(), schema, fmt, r,
-                                              &crc, kCrcTables.t);
-      if (emitted != psize[r])
-        throw std::runtime_error("internal error: emit size mismatch (record " +
-                                 std::to_string(r) + ")");
-      write_frame_header_footer_crc(file, frame_off[r], psize[r], crc,
-                                    kCrcTables.t);
-    }
+    int rc = IOPool::instance().run_parts(R, 4096, [&](i64 o, i64 m) {
+      for (i64 r = o; r < o + m; ++r) {
+        u8* dst = file + frame_off[r] + 12;
+        u32 crc = 0;
+        i64 emitted = emit_record_payload_fused(dst, cp, schema, fmt, r, &crc,
+                                                kCrcTables.t);
+        if (emitted != psize[r]) return 1;
+        write_frame_header_footer_crc(file, frame_off[r], psize[r], crc,
+                                      kCrcTables.t);
+      }
+      return 0;
+    });
+    emit_mismatch = rc != 0;
   }
+  if (emit_mismatch)
+    throw std::runtime_error("internal error: emit size mismatch");
   return py::bytes(out);
 }
 

@@ -81,7 +81,8 @@ class TFRecordIterableDataset(torch.utils.data.IterableDataset):
                  record_type: str = "Example", batch_rows: int = 65536,
                  columns: Optional[Sequence[str]] = None,
                  engine: str = "auto", verify_crc: bool = True,
-                 shuffle_files: bool = False, seed: int = 0):
+                 shuffle_files: bool = False, seed: int = 0,
+                 prefetch: int = 1):
         super().__init__()
         self.path = path
         self.record_type = record_type
@@ -91,6 +92,7 @@ class TFRecordIterableDataset(torch.utils.data.IterableDataset):
         self.verify_crc = verify_crc
         self.shuffle_files = shuffle_files
         self.seed = seed
+        self.prefetch = int(prefetch)  # shards decoded ahead (0 = sync)
         files = P.list_data_files(path)
         if not files:
             raise FileNotFoundError(f"No TFRecord files found under {path}")
@@ -123,28 +125,67 @@ class TFRecordIterableDataset(torch.utils.data.IterableDataset):
             files = files[info.id::info.num_workers]
         return files
 
+    def _decode_one(self, fpath: str, eng: str, sub_schema):
+        if os.path.getsize(fpath) == 0:
+            return None
+        if eng == "gpu" and P.codec_from_path(fpath) is None:
+            from .engine import gpu as gpu_engine
+            batch = gpu_engine.read_file_to_batch(
+                fpath, sub_schema, self.record_type, self.verify_crc)
+            torch.cuda.current_stream().synchronize()  # hand off across threads
+            return batch
+        from .engine import cpu as cpu_engine
+        data = np.frombuffer(P.decompress_file(fpath), np.uint8)
+        if data.size == 0:
+            return None
+        return cpu_engine.decode_buffer(data, sub_schema, self.record_type,
+                                        self.verify_crc)
+
     def __iter__(self) -> Iterator[Dict[str, torch.Tensor]]:
         eng = engine_mod.resolve_engine(self.engine)
         fields = [f for f in self.schema.fields
                   if self.columns is None or f.name in self.columns]
         sub_schema = StructType(fields)
-        for fpath in self._my_files():
-            if os.path.getsize(fpath) == 0:
-                continue
-            if eng == "gpu" and P.codec_from_path(fpath) is None:
-                from .engine import gpu as gpu_engine
-                batch = gpu_engine.read_file_to_batch(
-                    fpath, sub_schema, self.record_type, self.verify_crc)
-                device = None  # already device tensors
-            else:
-                from .engine import cpu as cpu_engine
-                data = np.frombuffer(P.decompress_file(fpath), np.uint8)
-                if data.size == 0:
-                    continue
-                batch = cpu_engine.decode_buffer(
-                    data, sub_schema, self.record_type, self.verify_crc)
-                device = None
-            yield from self._emit(batch, fields, device)
+        files = self._my_files()
+        if self.prefetch <= 0:
+            for fpath in files:
+                batch = self._decode_one(fpath, eng, sub_schema)
+                if batch is not None:
+                    yield from self._emit(batch, fields, None)
+            return
+        # background prefetch: a worker thread decodes the next shard(s) on
+        # its own CUDA stream while the consumer drains the current one
+        import queue
+        import threading
+
+        q: "queue.Queue" = queue.Queue(maxsize=self.prefetch)
+        SENTINEL = object()
+
+        def worker():
+            try:
+                stream = (torch.cuda.Stream()
+                          if eng == "gpu" and torch.cuda.is_available() else None)
+                for fpath in files:
+                    if stream is not None:
+                        with torch.cuda.stream(stream):
+                            b = self._decode_one(fpath, eng, sub_schema)
+                    else:
+                        b = self._decode_one(fpath, eng, sub_schema)
+                    if b is not None:
+                        q.put(b)
+                q.put(SENTINEL)
+            except BaseException as e:  # noqa: BLE001 — surface in consumer
+                q.put(e)
+
+        t = threading.Thread(target=worker, daemon=True)
+        t.start()
+        while True:
+            item = q.get()
+            if item is SENTINEL:
+                break
+            if isinstance(item, BaseException):
+                raise item
+            yield from self._emit(item, fields, None)
 
     def _emit(self, batch: RecordBatch, fields, device):
         R = batch.num_rows

@@ -60,3 +60,29 @@ class TestTorchDataset:
         dl = torch.utils.data.DataLoader(ds, batch_size=None, num_workers=2)
         total = sum(int(b["_num_rows"]) for b in dl)
         assert total == 4000
+
+    def test_prefetch_zero_and_default_agree(self, tmp_sandbox):
+        out, data = _write_dataset(tmp_sandbox, rows=2000, shards=5)
+        import spark_tfrecord_amd.torch_data as td
+        a = [int(b["_num_rows"]) for b in td.TFRecordIterableDataset(
+            out, batch_rows=999, engine="cpu", prefetch=0)]
+        b = [int(x["_num_rows"]) for x in td.TFRecordIterableDataset(
+            out, batch_rows=999, engine="cpu", prefetch=2)]
+        assert sum(a) == sum(b) == 2000
+        assert a == b  # same shard/batch order
+
+    def test_prefetch_propagates_errors(self, tmp_sandbox):
+        import pytest
+
+        out, _ = _write_dataset(tmp_sandbox, rows=100, shards=1)
+        # corrupt the shard
+        import os as _os
+        f = [p for p in _os.listdir(out) if p.startswith("part-")][0]
+        path = _os.path.join(out, f)
+        blob = bytearray(open(path, "rb").read())
+        blob[20] ^= 0xFF
+        open(path, "wb").write(bytes(blob))
+        ds = TFRecordIterableDataset(out, engine="cpu", prefetch=2,
+                                     schema=None)
+        with pytest.raises(RuntimeError):
+            list(ds)

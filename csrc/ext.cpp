@@ -682,11 +682,19 @@ py::dict infer_schema_codes(py::buffer data, py::array_t<i64> rec_off,
       }
     }
   }
+  // Feature names decode with replacement characters on invalid UTF-8,
+  // like the JVM's new String(bytes, UTF_8) in the reference — corrupt
+  // names surface as mangled fields (and CRC failures downstream), not
+  // as a decode exception here.
+  auto safe_str = [](const std::string& v) {
+    return py::reinterpret_steal<py::str>(
+        PyUnicode_DecodeUTF8(v.data(), (Py_ssize_t)v.size(), "replace"));
+  };
   py::dict out;
-  for (auto& kv : ctx_codes) out[py::str(kv.first)] = kv.second;
+  for (auto& kv : ctx_codes) out[safe_str(kv.first)] = kv.second;
   for (auto& kv : seq_codes) {
     // A name can only be context or sequence within one record type read.
-    out[py::str(kv.first)] = kv.second;
+    out[safe_str(kv.first)] = kv.second;
   }
   return out;
 }

@@ -913,11 +913,11 @@ def infer_codes_device(data: torch.Tensor, off: torch.Tensor,
     if hi - lo <= (4 << 20):
         blob = data[lo:hi].cpu().numpy().tobytes()
         for o, ln, code in zip(offs, nlens, tab[used, 2]):
-            name = blob[o - lo:o - lo + ln].decode("utf-8")
+            name = blob[o - lo:o - lo + ln].decode("utf-8", errors="replace")
             codes[name] = max(codes.get(name, 0), int(code))
     else:
         for o, ln, code in zip(offs, nlens, tab[used, 2]):
-            name = data[o:o + ln].cpu().numpy().tobytes().decode("utf-8")
+            name = data[o:o + ln].cpu().numpy().tobytes().decode("utf-8", errors="replace")
             codes[name] = max(codes.get(name, 0), int(code))
     return codes
 

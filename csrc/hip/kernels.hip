@@ -221,6 +221,8 @@ __global__ void emit_records_kernel(const DevCols* __restrict__ cols,
 }
 
 // ByteArray framing: payload r occupies [elem_off[r], elem_off[r+1]) of src.
+// The copy streams through a WriteCur so the frame CRC comes from the same
+// register windows the stores use (no payload re-read).
 __global__ void frame_bytes_kernel(const u8* __restrict__ src,
                                    const i64* __restrict__ elem_off,
                                    const i64* __restrict__ frame_off, i64 R,
@@ -230,10 +232,11 @@ __global__ void frame_bytes_kernel(const u8* __restrict__ src,
   for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
        r += (i64)gridDim.x * blockDim.x) {
     i64 n = elem_off[r + 1] - elem_off[r];
-    const u8* s = src + elem_off[r];
-    u8* d = file + frame_off[r] + 12;
-    for (i64 i = 0; i < n; ++i) d[i] = s[i];
-    write_frame_header_footer(file, frame_off[r], n, tab);
+    WriteCur w;
+    wcur_init(w, file + frame_off[r] + 12, tab);
+    wcur_bytes(w, src + elem_off[r], n);
+    u32 crc = wcur_finish(w);
+    write_frame_header_footer_crc(file, frame_off[r], n, crc, tab);
   }
 }
 
@@ -604,7 +607,9 @@ __global__ void infer_codes_kernel(const u8* __restrict__ data,
   }
 }
 
-// Raw-payload gather for the ByteArray read path (payload extents -> packed).
+// Raw-payload gather (ByteArray reads, partitionBy framed-record gather):
+// word-wise copies — gfx950 handles unaligned u64 loads/stores, so a byte
+// loop would cost 8x the memory instructions.
 __global__ void gather_payloads_kernel(const u8* __restrict__ data,
                                        const i64* __restrict__ off,
                                        const i64* __restrict__ len,
@@ -614,7 +619,14 @@ __global__ void gather_payloads_kernel(const u8* __restrict__ data,
        r += (i64)gridDim.x * blockDim.x) {
     const u8* s = data + off[r];
     u8* d = out + dst_off[r];
-    for (i64 i = 0; i < len[r]; ++i) d[i] = s[i];
+    i64 n = len[r];
+    i64 i = 0;
+    for (; i + 8 <= n; i += 8) {
+      u64 w;
+      __builtin_memcpy(&w, s + i, 8);
+      __builtin_memcpy(d + i, &w, 8);
+    }
+    for (; i < n; ++i) d[i] = s[i];
   }
 }
 

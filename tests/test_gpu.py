@@ -376,3 +376,15 @@ class TestGpuEndToEnd:
         rows = df.collect()
         assert [(r["a"], r["b"], r["v"]) for r in rows] == \
             [(1, "x", 0), (1, "y", 1), (2, "x", 2), (2, "x", 3), (1, "y", 4)]
+
+
+class TestShardWriterGpu:
+    def test_stream_writer_gpu_engine(self, tmp_sandbox):
+        p = str(tmp_sandbox / "swg" / "part-00000.tfrecord")
+        with stf.ShardWriter(p, engine="gpu") as w:
+            for k in range(4):
+                w.write({"x": np.arange(k * 100, k * 100 + 100,
+                                        dtype=np.int64)})
+        df = stf.read_tfrecord(p, engine="gpu").sort("x")
+        assert df.count() == 400
+        assert [r["x"] for r in df.collect()[:3]] == [0, 1, 2]

@@ -412,3 +412,42 @@ class TestProjectionAndCount:
         stf.write_tfrecord({"x": np.arange(123, dtype=np.int64)}, out,
                            engine="cpu", num_shards=3)
         assert stf.count_tfrecord(out, engine="cpu") == 123
+
+
+class TestGlobAndGzValidate:
+    def test_glob_pattern_read(self, tmp_sandbox):
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+
+        out = str(tmp_sandbox / "glob")
+        stf.write_tfrecord({"x": np.arange(30, dtype=np.int64)}, out,
+                           engine="cpu", num_shards=3)
+        df = stf.read_tfrecord(out + "/part-*.tfrecord", engine="cpu")
+        assert df.count() == 30
+
+    def test_validate_gzip_dataset(self, tmp_sandbox):
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+
+        out = str(tmp_sandbox / "vgz")
+        stf.write_tfrecord({"x": np.arange(40, dtype=np.int64)}, out,
+                           engine="cpu", codec="gzip", num_shards=2)
+        rep = stf.validate_tfrecord(out, engine="cpu")
+        assert rep.ok and rep.records == 40
+
+    def test_read_list_of_paths(self, tmp_sandbox):
+        import os
+
+        import numpy as np
+
+        import spark_tfrecord_amd as stf
+
+        out = str(tmp_sandbox / "lst")
+        stf.write_tfrecord({"x": np.arange(20, dtype=np.int64)}, out,
+                           engine="cpu", num_shards=4)
+        files = sorted(os.path.join(out, f) for f in os.listdir(out)
+                       if f.startswith("part-"))
+        df = stf.read_tfrecord(files[:2], engine="cpu")
+        assert df.count() == 10

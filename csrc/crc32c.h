@@ -108,6 +108,73 @@ TFR_HOSTDEV inline uint32_t mask_crc(uint32_t crc) {
   return ((crc >> 15) | (crc << 17)) + kMaskDelta;
 }
 
+// ---------------------------------------------------------------------------
+// CRC32C combination (zlib crc32_combine adapted to the Castagnoli
+// polynomial): crc(A||B) from crc(A), crc(B) and len(B). CRC shifting by
+// len(B) zero bytes is a linear operator over GF(2); it is applied with
+// 32x32 bit-matrices built by squaring. Enables wavefront-cooperative CRC:
+// 64 lanes each CRC a contiguous chunk, then 6 shuffle-combine levels.
+// ---------------------------------------------------------------------------
+
+struct CrcMat {
+  uint32_t m[32];  // column i = operator applied to the unit vector 1<<i
+};
+
+TFR_HOSTDEV inline uint32_t crcmat_times(const CrcMat& a, uint32_t vec) {
+  uint32_t sum = 0;
+  int i = 0;
+  while (vec) {
+    if (vec & 1u) sum ^= a.m[i];
+    vec >>= 1;
+    ++i;
+  }
+  return sum;
+}
+
+TFR_HOSTDEV inline void crcmat_square(CrcMat& out, const CrcMat& a) {
+  for (int i = 0; i < 32; ++i) out.m[i] = crcmat_times(a, a.m[i]);
+}
+
+// Builds the operator that advances a finalized CRC32C past `len` zero
+// bytes (i.e. M8^len where M8 shifts one byte). Fold uses: combined =
+// op * crc_left ^ crc_right. Building costs ~log2(len) matrix squarings —
+// callers folding many equal-length segments build it ONCE.
+TFR_HOSTDEV inline void crc32c_shift_op(CrcMat& out, uint64_t len) {
+  CrcMat even, odd;
+  odd.m[0] = kCrc32cPoly;  // operator for one zero BIT
+  for (int i = 1; i < 32; ++i) odd.m[i] = 1u << (i - 1);
+  crcmat_square(even, odd);  // 2 bits
+  crcmat_square(odd, even);  // 4 bits
+  crcmat_square(even, odd);  // 8 bits == one zero byte
+  // out = identity
+  for (int i = 0; i < 32; ++i) out.m[i] = 1u << i;
+  CrcMat* cur = &even;
+  CrcMat* nxt = &odd;
+  uint64_t n = len;
+  while (n) {
+    if (n & 1u) {
+      CrcMat tmp;
+      for (int i = 0; i < 32; ++i) tmp.m[i] = crcmat_times(*cur, out.m[i]);
+      out = tmp;
+    }
+    n >>= 1;
+    if (!n) break;
+    crcmat_square(*nxt, *cur);
+    CrcMat* t = cur;
+    cur = nxt;
+    nxt = t;
+  }
+}
+
+// crc(A||B) given crc of A (unmasked, finalized), crc of B, and len(B).
+TFR_HOSTDEV inline uint32_t crc32c_combine(uint32_t crc1, uint32_t crc2,
+                                           uint64_t len2) {
+  if (len2 == 0) return crc1;
+  CrcMat op;
+  crc32c_shift_op(op, len2);
+  return crcmat_times(op, crc1) ^ crc2;
+}
+
 TFR_HOSTDEV inline uint32_t unmask_crc(uint32_t masked) {
   uint32_t rot = masked - kMaskDelta;
   return (rot << 15) | (rot >> 17);

@@ -774,6 +774,9 @@ PYBIND11_MODULE(_native, m) {
         "TEST-ONLY: pass-A stats + payload CRCs (fused cursor vs two-pass)");
   m.def("crc32c", &crc32c_py, "CRC32C (Castagnoli) of a byte buffer");
   m.def("masked_crc32c", &masked_crc32c_py, "TFRecord-masked CRC32C");
+  m.def("crc32c_combine",
+        [](u32 c1, u32 c2, u64 len2) { return crc32c_combine(c1, c2, len2); },
+        "crc(A||B) from crc(A), crc(B), len(B) (GF(2) shift operator)");
   m.def("scan_frames", &scan_frames, py::arg("data"), py::arg("verify_crc") = true,
         "Scan TFRecord frames -> (payload offsets, payload lengths)");
   m.def("scan_frame_headers", &scan_frame_headers, py::arg("data"),

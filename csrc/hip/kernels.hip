@@ -220,6 +220,110 @@ __global__ void emit_records_kernel(const DevCols* __restrict__ cols,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Wavefront-cooperative large records. One-lane-per-record starves the chip
+// when records are big and few (16k x 16 KB records measured 13 GB/s vs
+// 21 GB/s for small records): a whole wave owns one record instead — copies
+// are lane-strided u64s, and the CRC splits into 64 contiguous chunks whose
+// finalized CRCs fold left-to-right with a precomputed GF(2) shift operator
+// (crc32c_shift_op; one ~log2(chunk) matrix build per record on lane 0).
+// ---------------------------------------------------------------------------
+
+constexpr i64 kWaveRecordBytes = 8 << 10;  // avg record size to switch modes
+
+__device__ inline u32 crc32c_wave(const u8* p, i64 n,
+                                  const uint32_t (*tab)[256]) {
+  int lane = threadIdx.x & 63;
+  i64 chunk = (n + 63) / 64;
+  if (chunk < 64) chunk = 64;  // tiny records: lane 0 does it all
+  i64 s = (i64)lane * chunk;
+  i64 e = s + chunk < n ? s + chunk : n;
+  u32 my = (s < e) ? crc32c_sw(p + s, (size_t)(e - s), 0, tab) : 0;
+  u32 acc = __shfl(my, 0);
+  if (chunk < n) {
+    // every lane folds redundantly in lockstep (uniform control flow, no
+    // divergence around the cross-lane reads); acc ends identical wave-wide
+    CrcMat op;
+    crc32c_shift_op(op, (u64)chunk);
+    for (int i = 1; i < 64; ++i) {
+      i64 si = (i64)i * chunk;
+      if (si >= n) break;
+      u32 ci = __shfl(my, i);
+      i64 li = (si + chunk < n ? chunk : n - si);
+      acc = (li == chunk) ? (crcmat_times(op, acc) ^ ci)
+                          : crc32c_combine(acc, ci, (u64)li);
+    }
+  }
+  return acc;
+}
+
+__device__ inline void copy_wave(u8* d, const u8* s, i64 n) {
+  int lane = threadIdx.x & 63;
+  i64 i = (i64)lane * 8;
+  for (; i + 8 <= n; i += 64 * 8) {
+    u64 w;
+    __builtin_memcpy(&w, s + i, 8);
+    __builtin_memcpy(d + i, &w, 8);
+  }
+  i64 tail = n & ~((i64)7);
+  for (i64 b = tail + lane; b < n; b += 64) d[b] = s[b];
+}
+
+// One record per WAVE: verify header + payload CRCs of large records.
+__global__ void crc_verify_wave_kernel(const u8* __restrict__ data,
+                                       const i64* __restrict__ off,
+                                       const i64* __restrict__ len, i64 R,
+                                       unsigned long long* err_out) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  i64 wave = (blockIdx.x * (i64)blockDim.x + threadIdx.x) / 64;
+  i64 nwaves = ((i64)gridDim.x * blockDim.x) / 64;
+  for (i64 r = wave; r < R; r += nwaves) {
+    const u8* h = data + off[r] - 12;
+    u32 payload = crc32c_wave(h + 12, len[r], tab);
+    if ((threadIdx.x & 63) == 0) {
+      u32 len_crc, data_crc;
+      __builtin_memcpy(&len_crc, h + 8, 4);
+      __builtin_memcpy(&data_crc, h + 12 + len[r], 4);
+      bool ok = mask_crc(crc32c_sw(h, 8, 0, tab)) == len_crc &&
+                mask_crc(payload) == data_crc;
+      if (!ok) atomicMin(err_out, (unsigned long long)(r + 1));
+    }
+  }
+}
+
+// One record per WAVE: ByteArray framing of large payloads.
+__global__ void frame_bytes_wave_kernel(const u8* __restrict__ src,
+                                        const i64* __restrict__ elem_off,
+                                        const i64* __restrict__ frame_off, i64 R,
+                                        u8* __restrict__ file) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  i64 wave = (blockIdx.x * (i64)blockDim.x + threadIdx.x) / 64;
+  i64 nwaves = ((i64)gridDim.x * blockDim.x) / 64;
+  for (i64 r = wave; r < R; r += nwaves) {
+    i64 n = elem_off[r + 1] - elem_off[r];
+    const u8* s = src + elem_off[r];
+    u8* d = file + frame_off[r] + 12;
+    copy_wave(d, s, n);
+    u32 crc = crc32c_wave(s, n, tab);
+    if ((threadIdx.x & 63) == 0)
+      write_frame_header_footer_crc(file, frame_off[r], n, crc, tab);
+  }
+}
+
+// One record per WAVE: payload gather of large extents.
+__global__ void gather_payloads_wave_kernel(const u8* __restrict__ data,
+                                            const i64* __restrict__ off,
+                                            const i64* __restrict__ len,
+                                            const i64* __restrict__ dst_off,
+                                            i64 R, u8* __restrict__ out) {
+  i64 wave = (blockIdx.x * (i64)blockDim.x + threadIdx.x) / 64;
+  i64 nwaves = ((i64)gridDim.x * blockDim.x) / 64;
+  for (i64 r = wave; r < R; r += nwaves)
+    copy_wave(out + dst_off[r], data + off[r], len[r]);
+}
+
 // ByteArray framing: payload r occupies [elem_off[r], elem_off[r+1]) of src.
 // The copy streams through a WriteCur so the frame CRC comes from the same
 // register windows the stores use (no payload re-read).
@@ -644,10 +748,17 @@ inline int grid_for(i64 n) {
 }
 
 void gpu_crc_verify(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
-                    uintptr_t err_out, uintptr_t stream) {
-  hipLaunchKernelGGL(crc_verify_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const u8*)data, (const i64*)off,
-                     (const i64*)len, R, (unsigned long long*)err_out);
+                    uintptr_t err_out, uintptr_t stream, i64 avg_bytes) {
+  if (avg_bytes > kWaveRecordBytes) {  // few big records: one per wave
+    hipLaunchKernelGGL(crc_verify_wave_kernel, dim3(grid_for(R * 64)),
+                       dim3(kBlock), 0, (hipStream_t)stream, (const u8*)data,
+                       (const i64*)off, (const i64*)len, R,
+                       (unsigned long long*)err_out);
+  } else {
+    hipLaunchKernelGGL(crc_verify_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                       (hipStream_t)stream, (const u8*)data, (const i64*)off,
+                       (const i64*)len, R, (unsigned long long*)err_out);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -736,10 +847,17 @@ void gpu_emit_records(uintptr_t cols_dev, uintptr_t schema_blob, int32_t fmt,
 }
 
 void gpu_frame_bytes(uintptr_t src, uintptr_t elem_off, uintptr_t frame_off, i64 R,
-                     uintptr_t file, uintptr_t stream) {
-  hipLaunchKernelGGL(frame_bytes_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const u8*)src, (const i64*)elem_off,
-                     (const i64*)frame_off, R, (u8*)file);
+                     uintptr_t file, uintptr_t stream, i64 avg_bytes) {
+  if (avg_bytes > kWaveRecordBytes) {
+    hipLaunchKernelGGL(frame_bytes_wave_kernel, dim3(grid_for(R * 64)),
+                       dim3(kBlock), 0, (hipStream_t)stream, (const u8*)src,
+                       (const i64*)elem_off, (const i64*)frame_off, R,
+                       (u8*)file);
+  } else {
+    hipLaunchKernelGGL(frame_bytes_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                       (hipStream_t)stream, (const u8*)src, (const i64*)elem_off,
+                       (const i64*)frame_off, R, (u8*)file);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -788,10 +906,19 @@ void gpu_infer_codes(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
 }
 
 void gpu_gather_payloads(uintptr_t data, uintptr_t off, uintptr_t len,
-                         uintptr_t dst_off, i64 R, uintptr_t out, uintptr_t stream) {
-  hipLaunchKernelGGL(gather_payloads_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const u8*)data, (const i64*)off,
-                     (const i64*)len, (const i64*)dst_off, R, (u8*)out);
+                         uintptr_t dst_off, i64 R, uintptr_t out,
+                         uintptr_t stream, i64 avg_bytes) {
+  if (avg_bytes > kWaveRecordBytes) {
+    hipLaunchKernelGGL(gather_payloads_wave_kernel, dim3(grid_for(R * 64)),
+                       dim3(kBlock), 0, (hipStream_t)stream, (const u8*)data,
+                       (const i64*)off, (const i64*)len, (const i64*)dst_off,
+                       R, (u8*)out);
+  } else {
+    hipLaunchKernelGGL(gather_payloads_kernel, dim3(grid_for(R)), dim3(kBlock),
+                       0, (hipStream_t)stream, (const u8*)data,
+                       (const i64*)off, (const i64*)len, (const i64*)dst_off,
+                       R, (u8*)out);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -972,16 +1099,22 @@ void register_gpu(py::module_& m) {
     hipError_t e = hipGetDeviceCount(&n);
     return (e == hipSuccess) ? n : -static_cast<int>(e);
   });
-  m.def("gpu_crc_verify", &gpu_crc_verify);
+  m.def("gpu_crc_verify", &gpu_crc_verify, py::arg("data"), py::arg("off"),
+        py::arg("len"), py::arg("R"), py::arg("err_out"), py::arg("stream"),
+        py::arg("avg_bytes") = 0);
   m.def("gpu_scan_records", &gpu_scan_records);
   m.def("gpu_extract_fields", &gpu_extract_fields);
   m.def("gpu_size_records", &gpu_size_records);
   m.def("gpu_emit_records", &gpu_emit_records);
-  m.def("gpu_frame_bytes", &gpu_frame_bytes);
+  m.def("gpu_frame_bytes", &gpu_frame_bytes, py::arg("src"),
+        py::arg("elem_off"), py::arg("frame_off"), py::arg("R"),
+        py::arg("file"), py::arg("stream"), py::arg("avg_bytes") = 0);
   m.def("gpu_frame_scan_blocks", &gpu_frame_scan_blocks);
   m.def("gpu_frame_scan_count", &gpu_frame_scan_count);
   m.def("gpu_frame_scan_emit", &gpu_frame_scan_emit);
-  m.def("gpu_gather_payloads", &gpu_gather_payloads);
+  m.def("gpu_gather_payloads", &gpu_gather_payloads, py::arg("data"),
+        py::arg("off"), py::arg("len"), py::arg("dst_off"), py::arg("R"),
+        py::arg("out"), py::arg("stream"), py::arg("avg_bytes") = 0);
   m.def("gpu_infer_codes", &gpu_infer_codes);
   m.def("gpu_scan_temp_bytes", &gpu_scan_temp_bytes);
   m.def("gpu_excl_sum_strided", &gpu_excl_sum_strided);

@@ -378,14 +378,18 @@ def _emit_and_chain(data, ranges, block_counts, N):
 # Decode
 # ---------------------------------------------------------------------------
 
-def crc_verify_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor):
-    """Raise on any bad frame CRC (parallel over records on the GPU)."""
+def crc_verify_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
+                      avg_bytes: int = 0):
+    """Raise on any bad frame CRC (parallel over records; records larger
+    than ~8 KB verify one-per-wavefront with GF(2)-combined chunk CRCs)."""
     R = off.numel()
     if R == 0:
         return
+    if not avg_bytes:
+        avg_bytes = max(0, (data.numel() // R) - 16)
     err = torch.full((1,), -1, dtype=torch.int64, device=data.device)  # all-ones
     _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(), lens.data_ptr(), R,
-                           err.data_ptr(), _stream())
+                           err.data_ptr(), _stream(), avg_bytes)
     bad = int(err.item())  # syncs
     if bad != -1:
         raise RuntimeError(
@@ -413,7 +417,7 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
         if R:
             _native.gpu_gather_payloads(data.data_ptr(), off.data_ptr(),
                                         lens.data_ptr(), dst_off.data_ptr(), R,
-                                        out.data_ptr(), _stream())
+                                        out.data_ptr(), _stream(), total // R)
         col = WireColumn(kind=KIND_BYTES, is_seq=False,
                          presence=torch.ones(R, dtype=torch.uint8, device=device),
                          row_off=torch.arange(R + 1, dtype=torch.int64, device=device),
@@ -439,7 +443,14 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     # stack is a blit KERNEL, so "free" overlap doesn't exist and the fused
     # form has ~1 ms less total GPU work per 215 MB.)
     crc_err = None
-    if verify_crc:
+    fuse_crc = verify_crc
+    avg = data.numel() // max(R, 1)
+    if verify_crc and avg > (8 << 10):
+        # huge records starve one-lane-per-record CRC inside the fused scan:
+        # verify them one-per-wavefront instead (GF(2)-combined chunk CRCs)
+        crc_verify_device(data, off, lens, avg)
+        fuse_crc = False
+    if fuse_crc:
         crc_err = torch.full((1,), -1, dtype=torch.int64, device=device)
     r0 = 0
     err1 = None
@@ -457,13 +468,13 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                                  FMT[record_type], blob.data_ptr(), F,
                                  stats.data_ptr() + r0 * stride,
                                  err.data_ptr(),
-                                 crc_err.data_ptr() if verify_crc else 0,
+                                 crc_err.data_ptr() if fuse_crc else 0,
                                  _stream())
-    if verify_crc and r0 > 0:
+    if fuse_crc and r0 > 0:
         # prescanned rows skipped the fused path: verify them separately
         _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(),
                                lens.data_ptr(), r0, crc_err.data_ptr(),
-                               _stream())
+                               _stream(), 0)
 
     # Per-field exclusive prefix sums ([F, R+1]): rocprim strided scans read
     # the stat column straight out of the [R, F, 6] buffer — no transpose
@@ -481,7 +492,7 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     list_base = excl_scan(4)
     totals = torch.stack([val_base[:, -1], byte_base[:, -1], list_base[:, -1]])
     totals_h = totals.cpu()  # one sync for all allocations
-    if verify_crc:
+    if fuse_crc:
         bad = int(crc_err.item())
         if bad != -1:
             raise RuntimeError(
@@ -602,7 +613,7 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
         if R:
             _native.gpu_frame_bytes(col.values.data_ptr(), col.elem_off.data_ptr(),
                                     frame_off.data_ptr(), R, file.data_ptr(),
-                                    _stream())
+                                    _stream(), (total // R) - 16)
         return file
 
     blob = torch.frombuffer(bytearray(schema_blob(batch.schema)),
@@ -860,7 +871,7 @@ def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,
     out = torch.empty(total, dtype=torch.uint8, device=device)
     _native.gpu_gather_payloads(file.data_ptr(), src_off.data_ptr(),
                                 sizes.data_ptr(), dst_off.data_ptr(), R,
-                                out.data_ptr(), _stream())
+                                out.data_ptr(), _stream(), total // max(R, 1))
     # partition boundaries in the ordered space
     counts = torch.bincount(codes, minlength=num_parts)
     row_bound = torch.nn.functional.pad(torch.cumsum(counts, 0), (1, 0))

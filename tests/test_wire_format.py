@@ -131,3 +131,20 @@ class TestProtoInterop:
         rows = df.sort("x").collect()
         assert [r["x"] for r in rows] == list(range(5))
         assert rows[2]["s"] == "v2"
+
+
+class TestCrcCombine:
+    def test_combine_matches_concatenation(self):
+        import numpy as np
+
+        from spark_tfrecord_amd import _native
+
+        rng = np.random.default_rng(0)
+        for _ in range(60):
+            a = rng.bytes(int(rng.integers(0, 3000)))
+            b = rng.bytes(int(rng.integers(0, 3000)))
+            want = _native.crc32c(np.frombuffer(a + b, np.uint8))
+            got = _native.crc32c_combine(
+                _native.crc32c(np.frombuffer(a, np.uint8)),
+                _native.crc32c(np.frombuffer(b, np.uint8)), len(b))
+            assert got == want

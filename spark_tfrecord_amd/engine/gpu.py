@@ -442,14 +442,12 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     # CRC pass was tried and reverted: every host<->device copy on this
     # stack is a blit KERNEL, so "free" overlap doesn't exist and the fused
     # form has ~1 ms less total GPU work per 215 MB.)
+    # NOTE: for Example/SequenceExample the CRC stays FUSED into the scan at
+    # any record size — the serial protobuf parse dominates huge records, so
+    # a separate wave-CRC pass only adds a second full read (measured).
+    # The wave CRC/copy kernels serve the parse-free ByteArray paths.
     crc_err = None
     fuse_crc = verify_crc
-    avg = data.numel() // max(R, 1)
-    if verify_crc and avg > (8 << 10):
-        # huge records starve one-lane-per-record CRC inside the fused scan:
-        # verify them one-per-wavefront instead (GF(2)-combined chunk CRCs)
-        crc_verify_device(data, off, lens, avg)
-        fuse_crc = False
     if fuse_crc:
         crc_err = torch.full((1,), -1, dtype=torch.int64, device=device)
     r0 = 0

@@ -43,14 +43,15 @@ def decode_buffer(data: np.ndarray, schema: StructType, record_type: str,
     off, lens = _native.scan_frames(data, verify_crc)
     R = len(off)
     if record_type == "ByteArray":
-        # payload extents ARE the binary column (elem per row)
-        starts = off
-        sizes = lens
+        # payload extents ARE the binary column (elem per row); vectorized
+        # ragged gather instead of a python loop over records
+        from ..arrow_interop import _ragged_gather_idx
+
+        starts = np.asarray(off, np.int64)
+        sizes = np.asarray(lens, np.int64)
         elem_off = np.zeros(R + 1, np.int64)
         np.cumsum(sizes, out=elem_off[1:])
-        out = np.empty(int(elem_off[-1]), np.uint8)
-        for r in range(R):
-            out[elem_off[r]:elem_off[r + 1]] = data[starts[r]:starts[r] + sizes[r]]
+        out = data[_ragged_gather_idx(starts, sizes)]
         col = WireColumn(kind=1, is_seq=False,
                          presence=np.ones(R, np.uint8),
                          row_off=np.arange(R + 1, dtype=np.int64),

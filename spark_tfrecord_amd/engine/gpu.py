@@ -72,7 +72,7 @@ def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
 _CHUNK = 64 << 20  # 64 MiB: overlap file IO with PCIe copies chunkwise
 # engine knobs: overridable via env for tuning
 _READ_SLICE = int(os.environ.get("TFREC_READ_SLICE", 48 << 20))
-_WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 5))
+_WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 8))
 # Prescan (structure-scan the arrived prefix under the tail DMA) measured
 # NET-NEGATIVE on this host: every host<->device copy executes as a blit
 # KERNEL (no SDMA for host-registered or torch-pinned memory — see

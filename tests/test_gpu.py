@@ -200,7 +200,7 @@ class TestGpuInference:
         data = np.frombuffer(img, np.uint8)
         off, lens = _native.scan_frames(data, False)
         cpu_codes = infer_codes_from_buffer(data, off, lens, record_type)
-        dev = torch.as_tensor(np.ascontiguousarray(data)).cuda()
+        dev = torch.as_tensor(np.ascontiguousarray(data).copy()).cuda()
         gpu_codes = g.infer_codes_device(
             dev, torch.as_tensor(off).cuda(), torch.as_tensor(lens).cuda(),
             record_type)

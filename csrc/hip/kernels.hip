@@ -945,6 +945,109 @@ size_t gpu_scan_temp_bytes(i64 n) {
   return bytes;
 }
 
+// ---------------------------------------------------------------------------
+// All per-field stat prefix sums in ONE launch: the decode needs exclusive
+// scans of nvals/nbytes/nlists for EVERY field ([F][R+1] each). Issuing 3F
+// rocprim scans costs 6F kernel launches; instead a single head-flag
+// SEGMENTED scan runs over the virtual [F*R] sequence of (nvals, nbytes,
+// nlists) triples (segments reset at field boundaries), scattering results
+// into the three [F][R+1] output planes through a custom output iterator.
+// ---------------------------------------------------------------------------
+
+struct Seg3 {
+  i64 v0, v1, v2;
+  u32 flag;
+};
+
+struct Seg3Combine {
+  __host__ __device__ Seg3 operator()(const Seg3& a, const Seg3& b) const {
+    if (b.flag) return b;
+    return Seg3{a.v0 + b.v0, a.v1 + b.v1, a.v2 + b.v2, a.flag};
+  }
+};
+
+struct Seg3Load {
+  const i64* stats;  // [R][F][6]
+  i64 R;
+  int F;
+  __host__ __device__ Seg3 operator()(i64 idx) const {
+    i64 f = idx / R;
+    i64 r = idx - f * R;
+    const i64* st = stats + (r * F + f) * 6;
+    return Seg3{st[2], st[3], st[4], r == 0 ? 1u : 0u};
+  }
+};
+
+// Writable proxy iterator: element idx scatters (v0,v1,v2) into the three
+// output planes at [f][r+1] (out[f][0] is pre-zeroed by the caller).
+struct Seg3OutRef {
+  i64* o0;
+  i64* o1;
+  i64* o2;
+  __host__ __device__ Seg3OutRef& operator=(const Seg3& v) {
+    *o0 = v.v0;
+    *o1 = v.v1;
+    *o2 = v.v2;
+    return *this;
+  }
+};
+
+struct Seg3OutIt {
+  using value_type = Seg3;
+  using reference = Seg3OutRef;
+  using pointer = Seg3OutRef*;
+  using difference_type = i64;
+  using iterator_category = std::random_access_iterator_tag;
+  i64* b0;
+  i64* b1;
+  i64* b2;  // [F][R+1] planes
+  i64 R;
+  i64 pos;
+  __host__ __device__ Seg3OutRef operator*() const { return (*this)[0]; }
+  __host__ __device__ Seg3OutRef operator[](i64 k) const {
+    i64 idx = pos + k;
+    i64 f = idx / R;
+    i64 r = idx - f * R;
+    i64 at = f * (R + 1) + r + 1;
+    return Seg3OutRef{b0 + at, b1 + at, b2 + at};
+  }
+  __host__ __device__ Seg3OutIt operator+(i64 k) const {
+    return Seg3OutIt{b0, b1, b2, R, pos + k};
+  }
+  __host__ __device__ Seg3OutIt& operator+=(i64 k) {
+    pos += k;
+    return *this;
+  }
+  __host__ __device__ Seg3OutIt& operator++() {
+    ++pos;
+    return *this;
+  }
+};
+
+size_t gpu_stat_scan_temp_bytes(i64 R, int F) {
+  size_t bytes = 0;
+  auto in = rocprim::make_transform_iterator(
+      rocprim::make_counting_iterator<i64>(0), Seg3Load{nullptr, R, F});
+  Seg3OutIt out{nullptr, nullptr, nullptr, R, 0};
+  (void)rocprim::inclusive_scan(nullptr, bytes, in, out, (size_t)(R * F),
+                                Seg3Combine());
+  return bytes;
+}
+
+void gpu_stat_scans(uintptr_t temp, size_t temp_bytes, uintptr_t stats, i64 R,
+                    int F, uintptr_t val_base, uintptr_t byte_base,
+                    uintptr_t list_base, uintptr_t stream) {
+  auto s = (hipStream_t)stream;
+  if (R <= 0 || F <= 0) return;
+  auto in = rocprim::make_transform_iterator(
+      rocprim::make_counting_iterator<i64>(0),
+      Seg3Load{(const i64*)stats, R, F});
+  Seg3OutIt out{(i64*)val_base, (i64*)byte_base, (i64*)list_base, R, 0};
+  size_t tb = temp_bytes;
+  HIP_CHECK(rocprim::inclusive_scan((void*)temp, tb, in, out,
+                                    (size_t)(R * F), Seg3Combine(), s));
+}
+
 // Writes out[0] = 0 and out[1..n] = inclusive scan of in[0], in[stride], ...
 // i.e. out is the (n+1)-long exclusive scan with the total at out[n].
 void gpu_excl_sum_strided(uintptr_t temp, size_t temp_bytes, uintptr_t in,
@@ -1117,6 +1220,8 @@ void register_gpu(py::module_& m) {
         py::arg("out"), py::arg("stream"), py::arg("avg_bytes") = 0);
   m.def("gpu_infer_codes", &gpu_infer_codes);
   m.def("gpu_scan_temp_bytes", &gpu_scan_temp_bytes);
+  m.def("gpu_stat_scan_temp_bytes", &gpu_stat_scan_temp_bytes);
+  m.def("gpu_stat_scans", &gpu_stat_scans);
   m.def("gpu_excl_sum_strided", &gpu_excl_sum_strided);
   m.def("file_mmap_pinned", &file_mmap_pinned, py::arg("path"), py::arg("n"),
         py::arg("writable"));

@@ -474,20 +474,24 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                                lens.data_ptr(), r0, crc_err.data_ptr(),
                                _stream(), 0)
 
-    # Per-field exclusive prefix sums ([F, R+1]): rocprim strided scans read
-    # the stat column straight out of the [R, F, 6] buffer — no transpose
-    # materialization, no torch cumsum (see profiles/r01_kernel_stats.txt).
-    def excl_scan(stat_col: int) -> torch.Tensor:
-        out = torch.empty((F, R + 1), dtype=torch.int64, device=device)
-        for i in range(F):
-            _excl_sum_into(out.data_ptr() + i * (R + 1) * 8,
-                           stats.data_ptr() + (i * 6 + stat_col) * 8,
-                           F * 6, R, device)
-        return out
-
-    val_base = excl_scan(2)
-    byte_base = excl_scan(3)
-    list_base = excl_scan(4)
+    # Per-field exclusive prefix sums ([F, R+1] x 3 planes) in ONE rocprim
+    # launch: a head-flag segmented scan over the [F*R] triple sequence
+    # scatters into all planes (3F separate scans cost 6F launches).
+    val_base = torch.empty((F, R + 1), dtype=torch.int64, device=device)
+    byte_base = torch.empty((F, R + 1), dtype=torch.int64, device=device)
+    list_base = torch.empty((F, R + 1), dtype=torch.int64, device=device)
+    val_base[:, 0] = 0
+    byte_base[:, 0] = 0
+    list_base[:, 0] = 0
+    key = ("stat", device.index if hasattr(device, "index") else 0)
+    need = _native.gpu_stat_scan_temp_bytes(R, F)
+    ws = _scan_ws.get(key)
+    if ws is None or ws.numel() < need:
+        ws = torch.empty(int(need * 2) or 1, dtype=torch.uint8, device=device)
+        _scan_ws[key] = ws
+    _native.gpu_stat_scans(ws.data_ptr(), ws.numel(), stats.data_ptr(), R, F,
+                           val_base.data_ptr(), byte_base.data_ptr(),
+                           list_base.data_ptr(), _stream())
     totals = torch.stack([val_base[:, -1], byte_base[:, -1], list_base[:, -1]])
     totals_h = totals.cpu()  # one sync for all allocations
     if fuse_crc:

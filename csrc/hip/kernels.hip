@@ -179,20 +179,16 @@ __global__ void extract_fields_kernel(const u8* __restrict__ data, i64 R, int F,
 // Encode: size pass (thread per record), emit pass, frame+CRC pass.
 // ---------------------------------------------------------------------------
 
-constexpr int kMaxFieldsEnc = 64;
+// The device column table is a variable-length FieldColumn array sized by
+// the schema (no fixed field cap; schema.nfields drives every loop).
 
-struct DevCols {
-  int32_t nfields;
-  FieldColumn c[kMaxFieldsEnc];
-};
-
-__global__ void size_records_kernel(const DevCols* __restrict__ cols,
+__global__ void size_records_kernel(const FieldColumn* __restrict__ cols,
                                     const u8* __restrict__ schema_blob, int32_t fmt,
                                     i64 R, i64* __restrict__ psize) {
   SchemaView schema = schema_view(schema_blob);
   for (i64 r = blockIdx.x * (i64)blockDim.x + threadIdx.x; r < R;
        r += (i64)gridDim.x * blockDim.x) {
-    psize[r] = record_payload_size(cols->c, schema, fmt, r) + kFrameOverhead;
+    psize[r] = record_payload_size(cols, schema, fmt, r) + kFrameOverhead;
   }
 }
 
@@ -200,7 +196,7 @@ __global__ void size_records_kernel(const DevCols* __restrict__ cols,
 // The frame header/footer CRC is computed HERE, right after the payload is
 // emitted, while the bytes are still L2-hot — a separate CRC pass re-reads
 // the whole file image from HBM (~0.5 ms / 215 MB, r01 profile).
-__global__ void emit_records_kernel(const DevCols* __restrict__ cols,
+__global__ void emit_records_kernel(const FieldColumn* __restrict__ cols,
                                     const u8* __restrict__ schema_blob, int32_t fmt,
                                     i64 r0, i64 R, const i64* __restrict__ frame_off,
                                     u8* __restrict__ file,
@@ -213,7 +209,7 @@ __global__ void emit_records_kernel(const DevCols* __restrict__ cols,
     i64 payload = (frame_off[r + 1] - frame_off[r]) - kFrameOverhead;
     u8* o = file + frame_off[r] + 12;
     u32 crc = 0;
-    i64 emitted = emit_record_payload_fused(o, cols->c, schema, fmt, r, &crc,
+    i64 emitted = emit_record_payload_fused(o, cols, schema, fmt, r, &crc,
                                             tab);
     if (emitted != payload) err[0] = ERR_OVERFLOW;
     write_frame_header_footer_crc(file, frame_off[r], payload, crc, tab);
@@ -801,16 +797,14 @@ void gpu_extract_fields(uintptr_t data, i64 R, int F, uintptr_t stats,
   HIP_CHECK(hipGetLastError());
 }
 
-size_t gpu_devcols_bytes() { return sizeof(DevCols); }
+size_t gpu_devcols_bytes(int F) { return sizeof(FieldColumn) * (size_t)F; }
 
-void fill_devcols(py::list cols, DevCols* out) {
+void fill_devcols(py::list cols, std::vector<FieldColumn>& out) {
   int F = (int)cols.size();
-  if (F > kMaxFieldsEnc)
-    throw std::invalid_argument("GPU encode supports at most 64 fields");
-  out->nfields = F;
+  out.resize(F);
   for (int f = 0; f < F; ++f) {
     py::dict d = cols[f].cast<py::dict>();
-    FieldColumn& c = out->c[f];
+    FieldColumn& c = out[f];
     c.kind = d["kind"].cast<int32_t>();
     c.is_seq = d["is_seq"].cast<int32_t>();
     c.presence = (const u8*)d["presence"].cast<uintptr_t>();
@@ -826,12 +820,15 @@ void fill_devcols(py::list cols, DevCols* out) {
 
 void gpu_size_records(py::list cols, uintptr_t cols_dev, uintptr_t schema_blob,
                       int32_t fmt, i64 R, uintptr_t psize, uintptr_t stream) {
-  DevCols host_cols{};
-  fill_devcols(cols, &host_cols);
-  HIP_CHECK(hipMemcpyAsync((void*)cols_dev, &host_cols, sizeof(DevCols),
+  // thread_local so the buffer outlives the async H2D enqueue (ROCm stages
+  // pageable-source copies synchronously, but don't dangle a stack buffer)
+  static thread_local std::vector<FieldColumn> host_cols;
+  fill_devcols(cols, host_cols);
+  HIP_CHECK(hipMemcpyAsync((void*)cols_dev, host_cols.data(),
+                           sizeof(FieldColumn) * host_cols.size(),
                            hipMemcpyHostToDevice, (hipStream_t)stream));
   hipLaunchKernelGGL(size_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const DevCols*)cols_dev,
+                     (hipStream_t)stream, (const FieldColumn*)cols_dev,
                      (const u8*)schema_blob, fmt, R, (i64*)psize);
   HIP_CHECK(hipGetLastError());
 }
@@ -840,7 +837,7 @@ void gpu_emit_records(uintptr_t cols_dev, uintptr_t schema_blob, int32_t fmt,
                       i64 r0, i64 R, uintptr_t frame_off, uintptr_t file,
                       uintptr_t err, uintptr_t stream) {
   hipLaunchKernelGGL(emit_records_kernel, dim3(grid_for(R - r0)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const DevCols*)cols_dev,
+                     (hipStream_t)stream, (const FieldColumn*)cols_dev,
                      (const u8*)schema_blob, fmt, r0, R, (const i64*)frame_off,
                      (u8*)file, (int32_t*)err);
   HIP_CHECK(hipGetLastError());

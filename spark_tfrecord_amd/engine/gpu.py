@@ -621,8 +621,8 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
     blob = torch.frombuffer(bytearray(schema_blob(batch.schema)),
                             dtype=torch.uint8).to(device)
     col_dicts = [_col_ptrs(c) for c in batch.columns]
-    cols_dev = torch.empty(_native.gpu_devcols_bytes(), dtype=torch.uint8,
-                           device=device)
+    cols_dev = torch.empty(_native.gpu_devcols_bytes(len(col_dicts)),
+                           dtype=torch.uint8, device=device)
     psize = torch.empty(R, dtype=torch.int64, device=device)
     _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], R, psize.data_ptr(), _stream())
@@ -655,8 +655,8 @@ def write_batch_to_file(batch: RecordBatch, path: str,
     blob = torch.frombuffer(bytearray(schema_blob(batch.schema)),
                             dtype=torch.uint8).to(device)
     col_dicts = [_col_ptrs(c) for c in batch.columns]
-    cols_dev = torch.empty(_native.gpu_devcols_bytes(), dtype=torch.uint8,
-                           device=device)
+    cols_dev = torch.empty(_native.gpu_devcols_bytes(len(col_dicts)),
+                           dtype=torch.uint8, device=device)
     psize = torch.empty(R, dtype=torch.int64, device=device)
     _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], R, psize.data_ptr(), _stream())
@@ -852,8 +852,8 @@ def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,
     blob = torch.frombuffer(bytearray(schema_blob(batch.schema)),
                             dtype=torch.uint8).to(device)
     col_dicts = [_col_ptrs(c) for c in batch.columns]
-    cols_dev = torch.empty(_native.gpu_devcols_bytes(), dtype=torch.uint8,
-                           device=device)
+    cols_dev = torch.empty(_native.gpu_devcols_bytes(len(col_dicts)),
+                           dtype=torch.uint8, device=device)
     psize = torch.empty(R, dtype=torch.int64, device=device)
     _native.gpu_size_records(col_dicts, cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], R, psize.data_ptr(), _stream())

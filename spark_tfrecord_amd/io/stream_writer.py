@@ -69,7 +69,7 @@ class ShardWriter:
         if self.schema is None:
             self.schema = schema_from_arrow(table.schema)
         batch = table_to_batch(table, self.schema)
-        if self._eng == "gpu" and len(batch.columns) <= 64:
+        if self._eng == "gpu":
             from ..engine import gpu as gpu_engine
 
             raw = gpu_engine.encode_batch_from_cpu(batch, self.record_type)

@@ -130,8 +130,7 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
             batch = table_to_batch(chunk, schema)
             fname = P.part_file_name(shard_offset + s, codec, job_id)
             fpath = os.path.join(out_dir, fname)
-            # device DevCols holds at most 64 fields; wider schemas encode on host
-            if eng == "gpu" and len(batch.columns) <= 64:
+            if eng == "gpu":
                 from ..engine import gpu as gpu_engine
                 dev_batch = gpu_engine.batch_to_device(batch)
                 if codec is None:
@@ -206,7 +205,7 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
             return d
 
         if (eng == "gpu" and record_type != "ByteArray" and num_shards == 1
-                and table.num_rows > 0 and len(data_schema.fields) <= 64):
+                and table.num_rows > 0):
             # MI355X path: encode every row ONCE on the GPU, then split into
             # per-partition file images by gathering framed records in HBM
             # (frames are concatenable — no re-serialization per partition)

@@ -429,3 +429,14 @@ class TestWaveRecords:
         with pytest.raises(RuntimeError, match="CRC"):
             g.decode_buffer_to_cpu(np.frombuffer(bytes(bad), np.uint8),
                                    schema, "Example", verify_crc=True)
+
+
+class TestWideSchemaGpu:
+    def test_100_fields_gpu_roundtrip(self, tmp_sandbox):
+        out = str(tmp_sandbox / "wide")
+        data = {f"f{i:03d}": np.arange(200, dtype=np.int64) + i
+                for i in range(100)}
+        stf.write_tfrecord(data, out, engine="gpu")
+        df = stf.read_tfrecord(out, engine="gpu").sort("f000")
+        assert len(df.columns) == 100 and df.count() == 200
+        assert df.collect()[5]["f099"] == 104

@@ -49,6 +49,10 @@ enum CodecErr : int32_t {
   ERR_KIND_MISMATCH = -3,
   ERR_BAD_WIRETYPE = -4,
   ERR_OVERFLOW = -5,
+  // feature name longer than the 16-bit length the inference kernel's
+  // nameref packing carries (the reference has no such names in practice;
+  // erroring beats silently corrupting the ref)
+  ERR_NAME_TOO_LONG = -6,
 };
 
 // ---------------------------------------------------------------------------

@@ -57,6 +57,15 @@ __device__ inline void stage_crc_tables(uint32_t (*lds)[256]) {
   __syncthreads();
 }
 
+// Error buffers are int32[2]: [0] = codec error code, [1] = 1 + a record
+// index that hit it (atomicMax keeps one deterministically; Python adds any
+// launch base offset and includes the index in the raised message).
+__device__ inline void set_err(int32_t* err, int32_t code, i64 r) {
+  err[0] = code;
+  i64 v = r + 1;
+  atomicMax(&err[1], v > 0x7FFFFFFF ? (int32_t)0x7FFFFFFF : (int32_t)v);
+}
+
 // ---------------------------------------------------------------------------
 // CRC verification: one record per lane. err_out[0] stays 0 on success, else
 // holds 1 + first-bad-record index (atomicMin keeps the earliest).
@@ -130,9 +139,9 @@ __global__ void scan_records_kernel(const u8* __restrict__ data,
     } else {
       rc = scan_record(data, off[r], len[r], fmt, schema, st);
     }
-    if (rc != ERR_OK) err[0] = rc;
+    if (rc != ERR_OK) set_err(err, rc, r);
     for (int f = 0; f < F; ++f)
-      if (st[f].err != ERR_OK) err[0] = st[f].err;
+      if (st[f].err != ERR_OK) set_err(err, st[f].err, r);
   }
 }
 
@@ -171,7 +180,7 @@ __global__ void extract_fields_kernel(const u8* __restrict__ data, i64 R, int F,
                                m.val_base ? m.val_base[r] : 0,
                                m.byte_base ? m.byte_base[r] : 0,
                                m.list_base ? m.list_base[r] : 0);
-    if (rc != ERR_OK) err[0] = rc;
+    if (rc != ERR_OK) set_err(err, rc, r);
   }
 }
 
@@ -211,7 +220,7 @@ __global__ void emit_records_kernel(const FieldColumn* __restrict__ cols,
     u32 crc = 0;
     i64 emitted = emit_record_payload_fused(o, cols, schema, fmt, r, &crc,
                                             tab);
-    if (emitted != payload) err[0] = ERR_OVERFLOW;
+    if (emitted != payload) set_err(err, ERR_OVERFLOW, r);
     write_frame_header_footer_crc(file, frame_off[r], payload, crc, tab);
   }
 }
@@ -631,6 +640,7 @@ __device__ inline int32_t infer_features_body_dev(const u8* data, const u8* p,
         if (kind_seen) code = 6 + kind_seen;  // FeatureList infers 2-D
       }
     }
+    if (key_len > 0xFFFF) return ERR_NAME_TOO_LONG;  // nameref packs len in 16 bits
     long long nameref = ((long long)(key - data) << 16) | (long long)key_len;
     u64 hash = fnv1a64(key, key_len);
     if (!lds_infer_merge(ltab, hash, nameref, code) &&
@@ -662,7 +672,7 @@ __global__ void infer_codes_kernel(const u8* __restrict__ data,
       u64 tag;
       const u8* np = read_varint(p, end, &tag);
       if (!np) {
-        err[0] = ERR_BAD_VARINT;
+        set_err(err, ERR_BAD_VARINT, r);
         break;
       }
       p = np;
@@ -673,20 +683,20 @@ __global__ void infer_codes_kernel(const u8* __restrict__ data,
         u64 blen;
         p = read_varint(p, end, &blen);
         if (!p || (u64)(end - p) < blen) {
-          err[0] = ERR_TRUNCATED;
+          set_err(err, ERR_TRUNCATED, r);
           break;
         }
         int32_t rc = infer_features_body_dev(data, p, p + blen, is_fl,
                                              is_fl ? lseq : lctx, table, nslots);
         if (rc != ERR_OK) {
-          err[0] = rc;
+          set_err(err, rc, r);
           break;
         }
         p += blen;
       } else {
         p = skip_field(p, end, wt);
         if (!p) {
-          err[0] = ERR_TRUNCATED;
+          set_err(err, ERR_TRUNCATED, r);
           break;
         }
       }
@@ -1180,6 +1190,26 @@ void file_mmap_drop(const std::string& path) {
   drop_mapping_locked(path);
 }
 
+// Flush a mapped file's pages to backing store before the caller publishes
+// it by rename: DMA writes through a hipHostRegister'd mapping don't set
+// CPU page-table dirty bits, so an explicit msync is the reliable writeback
+// barrier on disk-backed filesystems (on tmpfs it is a near-free no-op) —
+// matching write_file_atomic's fsync-before-rename durability.
+void file_mmap_sync(const std::string& path) {
+  std::lock_guard<std::mutex> lk(g_mmap_mu);
+  auto it = g_mmap_cache.find(path);
+  if (it != g_mmap_cache.end() && it->second.ptr) {
+    if (msync(it->second.ptr, it->second.n, MS_SYNC) != 0)
+      throw std::runtime_error("msync failed: " + path + ": " +
+                               strerror(errno));
+  }
+  int fd = ::open(path.c_str(), O_RDONLY);
+  if (fd >= 0) {
+    (void)fsync(fd);
+    ::close(fd);
+  }
+}
+
 
 void gpu_memcpy_d2h(uintptr_t dst, uintptr_t src, i64 n, uintptr_t stream) {
   HIP_CHECK(hipMemcpyAsync((void*)dst, (const void*)src, (size_t)n,
@@ -1223,6 +1253,7 @@ void register_gpu(py::module_& m) {
   m.def("file_mmap_pinned", &file_mmap_pinned, py::arg("path"), py::arg("n"),
         py::arg("writable"));
   m.def("file_mmap_drop", &file_mmap_drop);
+  m.def("file_mmap_sync", &file_mmap_sync);
   m.def("gpu_memcpy_d2h", &gpu_memcpy_d2h);
   m.def("gpu_memcpy_h2d", &gpu_memcpy_h2d);
   m.def("gpu_devcols_bytes", &gpu_devcols_bytes);

@@ -79,6 +79,9 @@ _WRITE_SLICES = int(os.environ.get("TFREC_WRITE_SLICES", 8))
 # exp/exp_sdma.py), so "overlapped" compute contends with the copy kernels
 # and the extra syncs cost more than the hidden scan. Off by default.
 _PRESCAN = os.environ.get("TFREC_PRESCAN", "0") == "1"
+# msync mapped-DMA writes before the publishing rename (durability parity
+# with write_file_atomic's fsync; near-free on tmpfs). TFREC_SYNC=0 disables.
+_SYNC_WRITES = os.environ.get("TFREC_SYNC", "1") != "0"
 
 
 def read_file_to_device(path: str, device="cuda") -> torch.Tensor:
@@ -177,6 +180,8 @@ def device_to_file(img: torch.Tensor, path: str):
     if pinned:
         _multi_dma(ptr, img.data_ptr(), n, _native.gpu_memcpy_d2h,
                    after_main=False)
+        if _SYNC_WRITES:
+            _native.file_mmap_sync(path)
         return
     _write_file_staged(img, path)
 
@@ -213,6 +218,8 @@ def _write_file_staged(img: torch.Tensor, path: str):
             pos = nxt
             which = nwhich
         _os.ftruncate(fd, n)
+        if _SYNC_WRITES:
+            _os.fsync(fd)
     finally:
         _os.close(fd)
 
@@ -319,7 +326,7 @@ def _prescan_prefix(data, pre_ranges, block_counts, nblocks, N, arrived,
     blob = torch.frombuffer(bytearray(schema_blob(schema)),
                             dtype=torch.uint8).to(device)
     stats1 = torch.empty((K, F, 6), dtype=torch.int64, device=device)
-    err1 = torch.zeros(1, dtype=torch.int32, device=device)
+    err1 = torch.zeros(2, dtype=torch.int32, device=device)
     off1 = (pos[:K] + 12).contiguous()
     lens1 = lens[:K].contiguous()
     _native.gpu_scan_records(data.data_ptr(), off1.data_ptr(),
@@ -436,7 +443,8 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                             dtype=torch.uint8).to(device)
     # FieldStat[R][F] as int64 [R,F,6]: pos,len,nvals,nbytes,nlists,(kind|err)
     stats = torch.empty((R, F, 6), dtype=torch.int64, device=device)
-    err = torch.zeros(1, dtype=torch.int32, device=device)
+    # err buffers are int32[2]: (codec error code, 1 + offending record index)
+    err = torch.zeros(2, dtype=torch.int32, device=device)
     # frame CRC verification is FUSED into the structure scan: the record
     # bytes are CRC'd while L2-hot from the parse. (A concurrent side-stream
     # CRC pass was tried and reverted: every host<->device copy on this
@@ -468,10 +476,14 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                                  err.data_ptr(),
                                  crc_err.data_ptr() if fuse_crc else 0,
                                  _stream())
+    crc_err_pre = None
     if fuse_crc and r0 > 0:
-        # prescanned rows skipped the fused path: verify them separately
+        # prescanned rows skipped the fused path: verify them separately.
+        # Separate error buffer — the fused kernel reports indices relative
+        # to its r0-based sub-launch, this one reports absolute indices.
+        crc_err_pre = torch.full((1,), -1, dtype=torch.int64, device=device)
         _native.gpu_crc_verify(data.data_ptr(), off.data_ptr(),
-                               lens.data_ptr(), r0, crc_err.data_ptr(),
+                               lens.data_ptr(), r0, crc_err_pre.data_ptr(),
                                _stream(), 0)
 
     # Per-field exclusive prefix sums ([F, R+1] x 3 planes) in ONE rocprim
@@ -496,12 +508,23 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     totals_h = totals.cpu()  # one sync for all allocations
     if fuse_crc:
         bad = int(crc_err.item())
-        if bad != -1:
+        # fused-scan indices are relative to the r0-based sub-launch
+        bad_abs = (bad - 1 + r0) if bad != -1 else -1
+        if crc_err_pre is not None:
+            bp = int(crc_err_pre.item())
+            if bp != -1:
+                bad_abs = bp - 1 if bad_abs == -1 else min(bad_abs, bp - 1)
+        if bad_abs != -1:
             raise RuntimeError(
-                f"corrupt TFRecord: bad CRC in record {bad - 1 if bad > 0 else bad}")
-    rc = int(err.item()) or (int(err1.item()) if err1 is not None else 0)
+                f"corrupt TFRecord: bad CRC in record {bad_abs}")
+    rc = int(err[0].item())
+    rec = int(err[1].item()) - 1 + r0  # scan indices are relative to r0
+    if rc == 0 and err1 is not None:
+        rc = int(err1[0].item())
+        rec = int(err1[1].item()) - 1  # prescan launch base is record 0
     if rc != 0:
-        raise RuntimeError(f"TFRecord decode failed (native error "
+        where = f" in record {rec}" if rec >= 0 else ""
+        raise RuntimeError(f"TFRecord decode failed{where} (native error "
                            f"{rc}; kind mismatch or malformed record)")
 
     metas = []
@@ -554,9 +577,11 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                   o["f32_vals"] if kind == KIND_FLOAT else o["bytes_data"])
         cols.append(WireColumn(kind, seq, presence, val_base[i], values,
                                elem_off, list_base[i] if seq else None, sub_off))
-    if int(err.item()) != 0:
-        raise RuntimeError(f"TFRecord decode failed in value extraction "
-                           f"(native error {int(err.item())})")
+    if int(err[0].item()) != 0:
+        rec = int(err[1].item()) - 1
+        where = f" (record {rec})" if rec >= 0 else ""
+        raise RuntimeError(f"TFRecord decode failed in value extraction"
+                           f"{where} (native error {int(err[0].item())})")
     return RecordBatch(schema, cols, R)
 
 
@@ -601,6 +626,13 @@ def _col_ptrs(col: WireColumn) -> dict:
     }
 
 
+def _check_emit_err(err: torch.Tensor):
+    if int(err[0].item()) != 0:
+        rec = int(err[1].item()) - 1
+        where = f" in record {rec}" if rec >= 0 else ""
+        raise RuntimeError(f"TFRecord encode failed: size/emit mismatch{where}")
+
+
 def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
     """Device wire-form batch -> framed file image as a device u8 tensor."""
     check_native()
@@ -629,12 +661,11 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
     frame_off = excl_sum(psize)
     total = int(frame_off[-1].item())
     file = torch.empty(total, dtype=torch.uint8, device=device)
-    err = torch.zeros(1, dtype=torch.int32, device=device)
+    err = torch.zeros(2, dtype=torch.int32, device=device)
     _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], 0, R, frame_off.data_ptr(),
                              file.data_ptr(), err.data_ptr(), _stream())
-    if int(err.item()) != 0:
-        raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+    _check_emit_err(err)
     return file
 
 
@@ -666,14 +697,13 @@ def write_batch_to_file(batch: RecordBatch, path: str,
     bounds = frame_off[ridx].cpu()  # one sync: slice byte bounds + total
     total = int(bounds[-1])
     file = torch.empty(total, dtype=torch.uint8, device=device)
-    err = torch.zeros(1, dtype=torch.int32, device=device)
+    err = torch.zeros(2, dtype=torch.int32, device=device)
     ptr, pinned = _native.file_mmap_pinned(path, total, True)
     if not pinned:
         _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                                  FMT[record_type], 0, R, frame_off.data_ptr(),
                                  file.data_ptr(), err.data_ptr(), _stream())
-        if int(err.item()) != 0:
-            raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+        _check_emit_err(err)
         _write_file_staged(file, path)
         return total
     main = torch.cuda.current_stream()
@@ -691,8 +721,9 @@ def write_batch_to_file(batch: RecordBatch, path: str,
                                    w.cuda_stream)
     for w in streams:
         w.synchronize()
-    if int(err.item()) != 0:
-        raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+    if _SYNC_WRITES:
+        _native.file_mmap_sync(path)
+    _check_emit_err(err)
     return total
 
 
@@ -860,7 +891,7 @@ def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,
     frame_off = excl_sum(psize)
     total = int(frame_off[-1].item())
     file = torch.empty(total, dtype=torch.uint8, device=device)
-    err = torch.zeros(1, dtype=torch.int32, device=device)
+    err = torch.zeros(2, dtype=torch.int32, device=device)
     _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], 0, R, frame_off.data_ptr(),
                              file.data_ptr(), err.data_ptr(), _stream())
@@ -878,8 +909,7 @@ def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,
     counts = torch.bincount(codes, minlength=num_parts)
     row_bound = torch.nn.functional.pad(torch.cumsum(counts, 0), (1, 0))
     byte_bound = dst_off[row_bound].cpu().numpy()
-    if int(err.item()) != 0:
-        raise RuntimeError("TFRecord encode failed: size/emit mismatch")
+    _check_emit_err(err)
     ranges = [(p, int(byte_bound[p]), int(byte_bound[p + 1]))
               for p in range(num_parts) if byte_bound[p + 1] > byte_bound[p]]
     return out, ranges
@@ -889,28 +919,50 @@ def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,
 # Schema inference on device: hash-table lattice kernel.
 # ---------------------------------------------------------------------------
 
-_INFER_SLOTS = 2048  # 1024 context + 1024 sequence feature names
+_INFER_SLOTS = 2048       # 1024 context + 1024 sequence feature names
+_INFER_SLOTS_MAX = 1 << 21  # growth cap (1M distinct names per half)
+_ERR_OVERFLOW = -5
+_ERR_NAME_TOO_LONG = -6
 
 
 def infer_codes_device(data: torch.Tensor, off: torch.Tensor,
                        lens: torch.Tensor, record_type: str) -> dict:
     """Device records -> {feature name: lattice code}. The kernel interns
     names into a device hash table and max-merges per-feature codes; the
-    host only reads back the (tiny) table and resolves name strings."""
+    host only reads back the (tiny) table and resolves name strings. The
+    table GROWS on overflow (datasets with thousands of distinct feature
+    names rerun the kernel with 4x the slots; the scan is cheap relative to
+    the decode it precedes, so the retry is simpler than spilling)."""
     check_native()
     device = data.device
     fmt = FMT["SequenceExample" if record_type == "SequenceExample" else "Example"]
-    table = torch.zeros((_INFER_SLOTS, 3), dtype=torch.int64, device=device)
-    err = torch.zeros(1, dtype=torch.int32, device=device)
     R = off.numel()
-    if R:
-        _native.gpu_infer_codes(data.data_ptr(), off.data_ptr(), lens.data_ptr(),
-                                R, fmt, table.data_ptr(), _INFER_SLOTS,
-                                err.data_ptr(), _stream())
-    tab = table.cpu().numpy()
-    if int(err.item()) != 0:
-        raise RuntimeError("malformed record during schema inference "
-                           f"(native error {int(err.item())})")
+    slots = _INFER_SLOTS
+    while True:
+        table = torch.zeros((slots, 3), dtype=torch.int64, device=device)
+        err = torch.zeros(2, dtype=torch.int32, device=device)
+        if R:
+            _native.gpu_infer_codes(data.data_ptr(), off.data_ptr(),
+                                    lens.data_ptr(), R, fmt, table.data_ptr(),
+                                    slots, err.data_ptr(), _stream())
+        tab = table.cpu().numpy()
+        rc = int(err[0].item())
+        if rc == _ERR_OVERFLOW:
+            if slots < _INFER_SLOTS_MAX:
+                slots *= 4
+                continue
+            raise RuntimeError(
+                "schema inference failed: more than "
+                f"{_INFER_SLOTS_MAX // 2} distinct feature names")
+        if rc == _ERR_NAME_TOO_LONG:
+            raise RuntimeError(
+                "schema inference failed: a feature name exceeds 65535 bytes")
+        if rc != 0:
+            rec = int(err[1].item()) - 1
+            where = f" (record {rec})" if rec >= 0 else ""
+            raise RuntimeError(f"malformed record during schema inference"
+                               f"{where} (native error {rc})")
+        break
     used = np.nonzero(tab[:, 0])[0]
     if len(used) == 0:
         return {}

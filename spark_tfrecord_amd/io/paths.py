@@ -25,7 +25,8 @@ from typing import Dict, List, Optional, Tuple
 __all__ = ["normalize_codec", "codec_extension", "compress_bytes",
            "decompress_file", "list_data_files", "partition_values_of",
            "apply_save_mode", "part_file_name", "write_file_atomic",
-           "write_success_marker", "SaveModeError"]
+           "write_success_marker", "SaveModeError", "hidden_tmp_path",
+           "escape_path_name", "unescape_path_name"]
 
 # codec option values accepted, mirroring Hadoop codec class names + shortcuts
 _CODEC_ALIASES = {
@@ -147,9 +148,24 @@ def part_file_name(shard: int, codec: Optional[str], job_id: str) -> str:
     return f"part-{shard:05d}-{job_id}.tfrecord{codec_extension(codec)}"
 
 
+def hidden_tmp_path(final_path: str, suffix: str = "inprogress") -> str:
+    """In-progress temp name for `final_path`, HIDDEN from readers: the
+    basename gets a leading '.' so _is_data_file never lists a crashed
+    job's leftovers as data (the analog of Spark's _temporary staging)."""
+    d, base = os.path.split(final_path)
+    return os.path.join(d, f".{base}.{suffix}")
+
+
+# Temp-file suffixes from older layouts; excluded from reads as well so a
+# partial file from a crashed pre-fix job is never scanned as data.
+_TMP_MARKERS = (".inprogress", ".__tmp.")
+
+
 def _is_data_file(name: str) -> bool:
     base = os.path.basename(name)
-    return not (base.startswith("_") or base.startswith("."))
+    if base.startswith("_") or base.startswith("."):
+        return False
+    return not any(m in base for m in _TMP_MARKERS)
 
 
 def list_data_files(path) -> List[str]:
@@ -178,8 +194,51 @@ def list_data_files(path) -> List[str]:
     return sorted(paths)
 
 
+# ---------------------------------------------------------------------------
+# Hive-style partition path escaping. Spark escapes these characters in
+# `col=value/` components via ExternalCatalogUtils.escapePathName (the layer
+# above the reference library; its layout tests TFRecordIOSuite.scala:140-151
+# rely on it). A value containing '/', '=' or '%' must not change the
+# directory structure or collide with the escape syntax itself.
+# ---------------------------------------------------------------------------
+
+_NEEDS_ESCAPE = set('"#%\'*/:=?\\{[]^\x7f') | {chr(c) for c in range(0x20)}
+
+
+def escape_path_name(value: str) -> str:
+    """Escape a partition value for use as the `value` of a `col=value/`
+    path component (Hive/Spark %XX escaping)."""
+    out = []
+    for ch in value:
+        if ch in _NEEDS_ESCAPE:
+            out.append(f"%{ord(ch):02X}")
+        else:
+            out.append(ch)
+    return "".join(out)
+
+
+def unescape_path_name(comp: str) -> str:
+    """Inverse of escape_path_name (tolerates malformed % sequences)."""
+    out = []
+    i = 0
+    n = len(comp)
+    while i < n:
+        ch = comp[i]
+        if ch == "%" and i + 2 < n + 1 and i + 3 <= n:
+            try:
+                out.append(chr(int(comp[i + 1:i + 3], 16)))
+                i += 3
+                continue
+            except ValueError:
+                pass
+        out.append(ch)
+        i += 1
+    return "".join(out)
+
+
 def partition_values_of(file_path: str, base_dir: str) -> Dict[str, str]:
-    """Extract `col=value` partition components between base_dir and the file."""
+    """Extract `col=value` partition components between base_dir and the
+    file, unescaping Hive-style %XX sequences in both name and value."""
     rel = os.path.relpath(os.path.dirname(os.path.abspath(file_path)),
                           os.path.abspath(base_dir))
     out: Dict[str, str] = {}
@@ -188,7 +247,7 @@ def partition_values_of(file_path: str, base_dir: str) -> Dict[str, str]:
     for comp in rel.split(os.sep):
         if "=" in comp:
             k, v = comp.split("=", 1)
-            out[k] = v
+            out[unescape_path_name(k)] = unescape_path_name(v)
     return out
 
 
@@ -221,9 +280,10 @@ def apply_save_mode(path: str, mode: str) -> bool:
 
 def write_file_atomic(data: bytes, final_path: str):
     """Temp file + rename: a torn write never becomes visible (the engine's
-    stand-in for Spark's task-commit rename protocol)."""
+    stand-in for Spark's task-commit rename protocol). The temp name is
+    dot-prefixed so a crashed job's leftover is never listed as data."""
     os.makedirs(os.path.dirname(final_path), exist_ok=True)
-    tmp = final_path + f".__tmp.{uuid.uuid4().hex[:8]}"
+    tmp = hidden_tmp_path(final_path, f"tmp.{uuid.uuid4().hex[:8]}")
     with open(tmp, "wb") as f:
         f.write(data)
         f.flush()

@@ -24,7 +24,8 @@ from ..infer import (
     merge_code_maps,
     schema_from_codes,
 )
-from ..schema import LongType, StringType, StructField, StructType
+from ..schema import (LongType, StringType, StructField, StructType,
+                      validate_schema_for_record_type)
 from ..utils import IOMetrics, StageTimer
 from .. import _native
 from . import paths as P
@@ -154,6 +155,9 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
         schema = StructType([f for f in schema.fields if f.name in set(columns)])
         part_cols = [c for c in part_cols if c in set(columns)]
     data_schema = StructType([f for f in schema.fields if f.name not in part_cols])
+    # an Example cannot carry 2-D ragged fields: reject like the reference's
+    # deserializer construction would (TFRecordDeserializer.scala:148-175)
+    validate_schema_for_record_type(data_schema, record_type)
 
     # Files needing host bytes (compressed, or CPU engine) are loaded and
     # inflated by a thread pool — gzip/zlib release the GIL, so multi-file

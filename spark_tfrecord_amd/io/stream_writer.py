@@ -22,7 +22,7 @@ from typing import Optional
 
 from .. import engine as engine_mod
 from ..arrow_interop import schema_from_arrow, table_to_batch
-from ..schema import StructType
+from ..schema import StructType, validate_schema_for_record_type
 from . import paths as P
 from .writer import normalize_input
 
@@ -41,7 +41,7 @@ class ShardWriter:
         self.codec = P.normalize_codec(codec)
         self._eng = engine_mod.resolve_engine(engine)
         self.rows_written = 0
-        self._tmp = path + ".inprogress"
+        self._tmp = P.hidden_tmp_path(path)
         os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
         self._f = open(self._tmp, "wb")
         self._gz = (zlib.compressobj(6, zlib.DEFLATED, 16 + 15)
@@ -68,6 +68,7 @@ class ShardWriter:
         table = normalize_input(data, self.schema)
         if self.schema is None:
             self.schema = schema_from_arrow(table.schema)
+        validate_schema_for_record_type(self.schema, self.record_type)
         batch = table_to_batch(table, self.schema)
         if self._eng == "gpu":
             from ..engine import gpu as gpu_engine
@@ -91,6 +92,8 @@ class ShardWriter:
             return
         if self._gz is not None:
             self._f.write(self._gz.flush())
+        self._f.flush()
+        os.fsync(self._f.fileno())
         self._f.close()
         os.replace(self._tmp, self.path)
         self._closed = True

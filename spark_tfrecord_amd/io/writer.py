@@ -20,7 +20,7 @@ from .. import engine as engine_mod
 from ..arrow_interop import schema_from_arrow, schema_to_arrow, table_to_batch
 from ..columnar import RecordBatch
 from ..engine import cpu as cpu_engine
-from ..schema import BinaryType, StructType
+from ..schema import BinaryType, StructType, validate_schema_for_record_type
 from ..utils import IOMetrics, StageTimer
 from . import paths as P
 
@@ -59,11 +59,15 @@ def normalize_input(data, schema: Optional[StructType]) -> pa.Table:
 
 
 def _partition_dir_value(v) -> str:
+    """Directory-safe `value` for a `col=value/` component: Hive default
+    partition for nulls, Hive/Spark %XX escaping for special characters
+    (a value containing '/', '=' or '%' must not alter the layout)."""
     if v is None:
         return "__HIVE_DEFAULT_PARTITION__"
     if isinstance(v, float) and v == int(v):
         return str(int(v))
-    return str(v)
+    s = str(v)
+    return P.escape_path_name(s) if s else "__HIVE_DEFAULT_PARTITION__"
 
 
 def _factorize_partitions(table: pa.Table, partition_by: Sequence[str]):
@@ -135,7 +139,7 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
                 dev_batch = gpu_engine.batch_to_device(batch)
                 if codec is None:
                     # encode sliced + DMA straight into the (temp) file mapping
-                    tmp = fpath + ".inprogress"
+                    tmp = P.hidden_tmp_path(fpath)
                     nbytes = gpu_engine.write_batch_to_file(dev_batch, tmp,
                                                             record_type)
                     os.replace(tmp, fpath)
@@ -173,6 +177,7 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
     table = normalize_input(data, schema)
     if schema is None:
         schema = schema_from_arrow(table.schema)
+    validate_schema_for_record_type(schema, record_type)
     eng = engine_mod.resolve_engine(engine)
 
     if record_type == "ByteArray":
@@ -222,7 +227,7 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
                 if codec is None:
                     # atomic like write_file_atomic: land bytes in a temp
                     # file, then rename over the final name
-                    tmp = fpath + ".inprogress"
+                    tmp = P.hidden_tmp_path(fpath)
                     gpu_engine.device_to_file(img[lo:hi], tmp)
                     os.replace(tmp, fpath)
                 else:

@@ -32,7 +32,7 @@ from .. import _native
 from ..engine import resolve_engine
 from ..infer import byte_array_schema, infer_codes_from_buffer, schema_from_codes
 from ..io import paths as P
-from ..schema import StructType
+from ..schema import StructType, validate_schema_for_record_type
 
 __all__ = ["init_distributed", "shard_files", "infer_schema_distributed",
            "write_tfrecord_distributed", "read_tfrecord_distributed"]
@@ -220,6 +220,7 @@ def write_tfrecord_distributed(data, path: str, record_type: str = "Example",
     table = normalize_input(data, schema)
     if schema is None:
         schema = schema_from_arrow(table.schema)
+    validate_schema_for_record_type(schema, record_type)
     eng = resolve_engine(engine)
 
     # save-mode coordination: rank 0 prepares, all wait, everyone honors skip
@@ -317,7 +318,7 @@ def write_tfrecord_distributed(data, path: str, record_type: str = "Example",
 
                 raw_t = torch.cat(blobs) if len(blobs) > 1 else blobs[0]
                 if codec is None:
-                    tmp = fpath + ".inprogress"
+                    tmp = P.hidden_tmp_path(fpath)
                     gpu_engine.device_to_file(raw_t.contiguous(), tmp)
                     os.replace(tmp, fpath)
                 else:

@@ -34,6 +34,7 @@ __all__ = [
     "KIND_INT64",
     "wire_kind_of",
     "is_sequence_field",
+    "validate_schema_for_record_type",
     "merge_types",
     "type_precedence",
 ]
@@ -235,6 +236,23 @@ def is_sequence_field(dt: DataType) -> bool:
     (2-D ragged), i.e. ArrayType(ArrayType(_)).
     Reference: TFRecordSerializer.scala:45-47."""
     return isinstance(dt, ArrayType) and isinstance(dt.elementType, ArrayType)
+
+
+def validate_schema_for_record_type(schema: "StructType", record_type: str):
+    """Reject schemas the chosen record type cannot carry, like the
+    reference's converter construction does (TFRecordSerializer.scala:147-180
+    raises for types unsupported by the record type): a 2-D ragged column
+    (SequenceExample FeatureList) has no representation in an Example —
+    silently dropping it would be data loss.
+    """
+    if record_type != "Example":
+        return
+    for f in schema.fields:
+        if is_sequence_field(f.dataType):
+            raise TypeError(
+                f"Cannot convert field '{f.name}' of type "
+                f"{f.dataType.simple_string()} with recordType 'Example': "
+                f"nested arrays require recordType 'SequenceExample'")
 
 
 # ---------------------------------------------------------------------------

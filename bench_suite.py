@@ -104,7 +104,12 @@ def bench_partitionby(args):
         "score": pa.array(rng.random(rows).astype(np.float32)),
         "feats": pa.array(list(rng.random((rows, 8)).astype(np.float32))),
     })
-    d = _workdir("partby", 0)  # shared dir: ranks write distinct part files
+    # shared dir: ranks write distinct part files; rank 0 alone prepares it
+    if world == 1 or rank == 0:
+        d = _workdir("partby", 0)
+    else:
+        base = "/dev/shm" if os.path.isdir("/dev/shm") else tempfile.gettempdir()
+        d = os.path.join(base, "tfrec_suite_partby_r0")
     out = os.path.join(d, "t")
     eng = "gpu" if use_cuda else "cpu"
 
@@ -114,6 +119,8 @@ def bench_partitionby(args):
         if world > 1:
             import torch.distributed as td
             td.barrier()
+
+    sync()  # setup rmtree (rank 0) must complete before anyone writes
 
     def one_rep():
         if world > 1:
@@ -151,22 +158,6 @@ def bench_infer(args):
     use_cuda = torch.cuda.is_available()
     if world > 1:
         D.init_distributed()
-    rows = args.rows // max(world, 1)
-    rng = np.random.default_rng(7)
-    # ragged 2-D: FeatureList of FloatList
-    rag = [[list(rng.random(rng.integers(1, 6)).astype(float))
-            for _ in range(int(rng.integers(1, 5)))] for _ in range(rows)]
-    data = {"sid": np.arange(rows, dtype=np.int64), "rag": rag}
-    schema = stf.StructType([
-        stf.StructField("sid", stf.LongType(), True),
-        stf.StructField("rag", stf.ArrayType(stf.ArrayType(stf.FloatType())), True),
-    ])
-    d = _workdir("infer", rank if world > 1 else 0)
-    out = os.path.join(d, "t")
-    stf.write_tfrecord(data, out, record_type="SequenceExample", schema=schema,
-                       engine="gpu" if use_cuda else "cpu", mode="overwrite")
-    files = [os.path.join(out, f) for f in sorted(os.listdir(out))
-             if not f.startswith("_")]
 
     def sync():
         if use_cuda:
@@ -174,6 +165,34 @@ def bench_infer(args):
         if world > 1:
             import torch.distributed as td
             td.barrier()
+
+    # ONE shared dataset (the distributed contract: every rank passes the
+    # SAME file list; rank r scans records r::world of the chosen file and
+    # the lattice codes are max-all-reduced)
+    rows = args.rows
+    d = _workdir("infer", 0) if (world == 1 or rank == 0) else None
+    if d is None:
+        base = "/dev/shm" if os.path.isdir("/dev/shm") else tempfile.gettempdir()
+        d = os.path.join(base, "tfrec_suite_infer_r0")
+    out = os.path.join(d, "t")
+    if rank == 0:
+        rng = np.random.default_rng(7)
+        # ragged 2-D: FeatureList of FloatList
+        rag = [[list(rng.random(rng.integers(1, 6)).astype(float))
+                for _ in range(int(rng.integers(1, 5)))] for _ in range(rows)]
+        data = {"sid": np.arange(rows, dtype=np.int64), "rag": rag}
+        schema = stf.StructType([
+            stf.StructField("sid", stf.LongType(), True),
+            stf.StructField("rag",
+                            stf.ArrayType(stf.ArrayType(stf.FloatType())), True),
+        ])
+        stf.write_tfrecord(data, out, record_type="SequenceExample",
+                           schema=schema,
+                           engine="gpu" if use_cuda else "cpu",
+                           mode="overwrite")
+    sync()
+    files = [os.path.join(out, f) for f in sorted(os.listdir(out))
+             if not f.startswith("_")]
 
     def one_rep():
         if world > 1:
@@ -189,10 +208,10 @@ def bench_infer(args):
     sync()
     el = time.perf_counter() - t0
     assert "rag" in [f.name for f in s.fields]
-    total = rows * world * args.reps
+    total = rows * args.reps  # shared file: `rows` records scanned per rep
     _emit(rank, "records/sec schema inference (SequenceExample lattice)",
           total / el, "records/s",
-          {"rows_total": rows * world, "parallelism": f"dp{world}",
+          {"rows_total": rows, "parallelism": f"dp{world}",
            "engine": "gpu" if use_cuda else "cpu", "reps": args.reps}, el,
           world if use_cuda else 0)
 
@@ -203,14 +222,26 @@ def bench_gzip_bytearray(args):
 
     import spark_tfrecord_amd as stf
 
+    from spark_tfrecord_amd.parallel import dist as D
+
     rank, world = _env_world()
     use_cuda = torch.cuda.is_available()
+    if world > 1:
+        D.init_distributed()
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        if world > 1:
+            import torch.distributed as td
+            td.barrier()
+
     rows = args.rows // max(world, 1)
-    rng = np.random.default_rng(9)
+    rng = np.random.default_rng(9 + rank)
     payloads = [rng.bytes(200) for _ in range(rows)]
     import pyarrow as pa
     table = pa.table({"byteArray": pa.array(payloads, type=pa.large_binary())})
-    d = _workdir("gzba", rank if world > 1 else 0)
+    d = _workdir("gzba", rank)
     out = os.path.join(d, "t")
     # many shards, like one Spark task per file in the reference: gzip is
     # sequential PER file, so shards are the parallelism axis on read
@@ -220,18 +251,18 @@ def bench_gzip_bytearray(args):
                  for f in os.listdir(out) if not f.startswith("_"))
     eng = "gpu" if use_cuda else "cpu"
     stf.read_tfrecord(out, record_type="ByteArray", engine=eng)  # warmup
-    if use_cuda:
-        torch.cuda.synchronize()
+    sync()
     t0 = time.perf_counter()
     for _ in range(args.reps):
         df = stf.read_tfrecord(out, record_type="ByteArray", engine=eng)
         assert df.count() == rows
-    if use_cuda:
-        torch.cuda.synchronize()
+    sync()
     el = time.perf_counter() - t0
-    _emit(rank, "rows/sec gzip ByteArray read", rows * args.reps / el, "rows/s",
-          {"rows": rows, "gz_bytes": nbytes, "engine": eng,
-           "reps": args.reps}, el, 1 if use_cuda else 0)
+    _emit(rank, "rows/sec gzip ByteArray read",
+          rows * world * args.reps / el, "rows/s",
+          {"rows_total": rows * world, "gz_bytes_per_rank": nbytes,
+           "engine": eng, "parallelism": f"dp{world}", "reps": args.reps}, el,
+          world if use_cuda else 0)
 
 
 BENCHES = {"plumbing": bench_plumbing, "partitionby": bench_partitionby,

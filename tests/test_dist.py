@@ -161,6 +161,162 @@ def test_distributed_save_mode_ignore(tmp_path):
     _run(_w_save_mode_ignore, tmp_path)
 
 
+def _w_blob_exchange(rank, tmp, extra):
+    """Every branch of _all_to_all_blobs on gloo: bytes in, tensors in,
+    zero-length blobs, wildly uneven sizes. extra = per-blob scale factor."""
+    import torch
+    import torch.distributed as dist
+
+    from spark_tfrecord_amd.parallel.dist import _all_to_all_blobs
+
+    world = dist.get_world_size()
+    scale = int(extra or 1)
+
+    def payload(src, dst):
+        if (src + dst) % 3 == 0:
+            return b""  # empty-blob branch
+        n = ((src * 7 + dst * 13) % 11) * 1024 * scale + src + dst
+        return bytes([((src * 251) ^ (dst * 17) ^ (i & 0xFF)) & 0xFF
+                      for i in range(min(n, 64))]) * max(1, n // 64)
+
+    send = [payload(rank, d) for d in range(world)]
+    if rank % 2 == 1:  # tensor-input branch
+        send = [torch.frombuffer(bytearray(b), dtype=torch.uint8)
+                if len(b) else torch.zeros(0, dtype=torch.uint8)
+                for b in send]
+    recv = _all_to_all_blobs(send, torch.device("cpu"))
+    assert len(recv) == world
+    for src in range(world):
+        got = recv[src]
+        if hasattr(got, "numpy"):
+            got = bytes(got.cpu().numpy().tobytes())
+        want = payload(src, rank)
+        want_n = len(want) if isinstance(want, (bytes, bytearray)) else want.numel()
+        assert len(got) == want_n, (rank, src, len(got), want_n)
+        assert got == (want if isinstance(want, bytes) else bytes(want)), \
+            (rank, src)
+
+
+def _w_partitioned_uneven(rank, tmp, extra):
+    """Skew: rank 0 holds ALL rows of partition 'hot'; ranks >= world//2
+    hold no rows at all (zero-row ranks must still participate in every
+    collective); 3 partitions < world so most ranks own no partition."""
+    import torch.distributed as dist
+
+    from spark_tfrecord_amd.parallel import write_tfrecord_distributed
+
+    import pyarrow as pa
+
+    world = dist.get_world_size()
+    out = os.path.join(tmp, "uneven")
+    if rank == 0:
+        data = {"p": ["hot"] * 50 + ["a", "b"],
+                "x": np.arange(52, dtype=np.int64)}
+    elif rank < world // 2:
+        data = {"p": ["a", "b"], "x": np.array([100 + rank, 200 + rank], np.int64)}
+    else:
+        data = pa.table({"p": pa.array([], type=pa.large_utf8()),
+                         "x": pa.array([], type=pa.int64())})
+    write_tfrecord_distributed(data, out, partition_by=["p"], mode="overwrite")
+    dist.barrier()
+    if rank == 0:
+        import spark_tfrecord_amd as stf
+
+        dirs = sorted(d for d in os.listdir(out) if d.startswith("p="))
+        assert dirs == ["p=a", "p=b", "p=hot"], dirs
+        df = stf.read_tfrecord(out)
+        n_expected = 52 + 2 * (world // 2 - 1)
+        assert df.count() == n_expected, (df.count(), n_expected)
+        rows = df.collect()
+        hot = sorted(r["x"] for r in rows if r["p"] == "hot")
+        assert hot == list(range(50)), hot
+
+
+def _w_infer_sparse(rank, tmp, extra):
+    """Schema inference with fewer records than ranks: most ranks scan an
+    EMPTY slice (codes = {}) and must still agree on the merged schema."""
+    import torch.distributed as dist
+
+    import spark_tfrecord_amd as stf
+    from spark_tfrecord_amd.parallel import infer_schema_distributed
+
+    path = os.path.join(tmp, "sparse")
+    if rank == 0:
+        # three records with DIFFERENT inferred types for 'v' in ONE file
+        # (frames concatenate): long scalar, float scalar, long array
+        os.makedirs(path, exist_ok=True)
+        blobs = []
+        for i, (d, dt) in enumerate([
+                ({"v": [1]}, stf.LongType()),
+                ({"v": [2.5]}, stf.FloatType()),
+                ({"v": [[1, 2, 3]]}, stf.ArrayType(stf.LongType()))]):
+            td = os.path.join(tmp, f"sparse_src{i}")
+            stf.write_tfrecord(d, td, mode="overwrite", schema=stf.StructType(
+                [stf.StructField("v", dt, True)]))
+            part = next(f for f in os.listdir(td) if f.startswith("part-"))
+            with open(os.path.join(td, part), "rb") as f:
+                blobs.append(f.read())
+        with open(os.path.join(path, "part-00000-m.tfrecord"), "wb") as f:
+            f.write(b"".join(blobs))
+    dist.barrier()
+    files = [os.path.join(path, "part-00000-m.tfrecord")]
+    schema = infer_schema_distributed(files, "Example")
+    # precedence-max merge (TensorFlowInferSchema.scala:194-228):
+    # Long < Float < Arr[Long] -> Arr[Long]
+    assert schema["v"].dataType == stf.ArrayType(stf.LongType()), schema
+
+
+def _w_infer_empty_dataset(rank, tmp, extra):
+    """All files empty: every rank must raise the same ValueError."""
+    import pytest
+    import torch.distributed as dist
+
+    from spark_tfrecord_amd.parallel import infer_schema_distributed
+
+    path = os.path.join(tmp, "empty")
+    if rank == 0:
+        os.makedirs(path, exist_ok=True)
+        open(os.path.join(path, "part-00000-x.tfrecord"), "wb").close()
+    dist.barrier()
+    files = [os.path.join(path, "part-00000-x.tfrecord")]
+    with pytest.raises(ValueError, match="no non-empty"):
+        infer_schema_distributed(files, "Example")
+
+
+def _w_infer_bytearray(rank, tmp, extra):
+    """ByteArray short-circuit: fixed schema, no file IO, no collectives."""
+    import spark_tfrecord_amd as stf
+    from spark_tfrecord_amd.parallel import infer_schema_distributed
+
+    schema = infer_schema_distributed([], "ByteArray")
+    assert [f.name for f in schema.fields] == ["byteArray"]
+    assert isinstance(schema.fields[0].dataType, stf.BinaryType)
+
+
+def _w_read_fewer_files_than_ranks(rank, tmp, extra):
+    """Round-robin sharding with 2 files over a larger world: ranks beyond
+    the file count get a valid EMPTY DataFrame with the right schema."""
+    import torch.distributed as dist
+
+    import spark_tfrecord_amd as stf
+    from spark_tfrecord_amd.parallel import read_tfrecord_distributed
+
+    world = dist.get_world_size()
+    out = os.path.join(tmp, "few")
+    if rank == 0:
+        stf.write_tfrecord({"x": np.arange(6, dtype=np.int64)}, out,
+                           mode="overwrite", num_shards=2)
+    dist.barrier()
+    df = read_tfrecord_distributed(out)
+    local = sorted(r["x"] for r in df.collect())
+    gathered = [None] * world
+    dist.all_gather_object(gathered, local)
+    assert sorted(x for g in gathered for x in g) == list(range(6))
+    if rank >= 2:
+        assert local == []
+        assert [f.name for f in df.schema.fields] == ["x"]
+
+
 def _w_partitioned_any_world(rank, tmp, extra):
     import torch.distributed as dist
 
@@ -190,3 +346,43 @@ def _w_partitioned_any_world(rank, tmp, extra):
 
 def test_distributed_partitioned_write_world4(tmp_path):
     _run(_w_partitioned_any_world, tmp_path, world=4)
+
+
+# -- world=8 hardening (the dp8 shapes the driver's SCALE run exercises) -----
+
+@pytest.mark.timeout(420)
+def test_blob_exchange_world8(tmp_path):
+    _run(_w_blob_exchange, tmp_path, extra=1, world=8)
+
+
+@pytest.mark.timeout(600)
+@pytest.mark.skipif(os.environ.get("TFREC_BIG_DIST") != "1",
+                    reason="set TFREC_BIG_DIST=1 for the GB-scale exchange")
+def test_blob_exchange_world8_large(tmp_path):
+    # ~11 MB x 8 peers x 8 ranks ~ 0.7 GB total in flight over gloo
+    _run(_w_blob_exchange, tmp_path, extra=1024, world=8)
+
+
+@pytest.mark.timeout(420)
+def test_distributed_partitioned_uneven_world8(tmp_path):
+    _run(_w_partitioned_uneven, tmp_path, world=8)
+
+
+@pytest.mark.timeout(420)
+def test_distributed_infer_sparse_world8(tmp_path):
+    _run(_w_infer_sparse, tmp_path, world=8)
+
+
+@pytest.mark.timeout(240)
+def test_distributed_infer_empty_dataset(tmp_path):
+    _run(_w_infer_empty_dataset, tmp_path)
+
+
+@pytest.mark.timeout(240)
+def test_distributed_infer_bytearray(tmp_path):
+    _run(_w_infer_bytearray, tmp_path)
+
+
+@pytest.mark.timeout(420)
+def test_distributed_read_fewer_files_than_ranks(tmp_path):
+    _run(_w_read_fewer_files_than_ranks, tmp_path, world=5)

@@ -440,3 +440,82 @@ class TestWideSchemaGpu:
         df = stf.read_tfrecord(out, engine="gpu").sort("f000")
         assert len(df.columns) == 100 and df.count() == 200
         assert df.collect()[5]["f099"] == 104
+
+
+class TestStreamOrderingStress:
+    def test_mmap_cache_dma_churn(self, tmp_path):
+        """Size churn over two cached paths: every write retruncates and
+        re-registers the inode mapping while reads DMA from it on the side
+        streams — the stress shape for the drop-before-ftruncate ordering
+        in file_mmap_pinned (kernels.hip mapping cache)."""
+        g = _gpu_engine()
+        rng = np.random.default_rng(42)
+        paths = [str(tmp_path / f"st{i}.tfrecord") for i in range(2)]
+        for it in range(25):
+            n = int(rng.integers(1, 4000))
+            b = make_batch(n, seed=it)
+            p = paths[it % 2]
+            g.write_batch_to_file(g.batch_to_device(b), p, "Example")
+            out = g.batch_to_host(g.read_file_to_batch_pipelined(
+                p, b.schema, "Example", verify_crc=True))
+            assert_batches_equal(b, out)
+
+    def test_interleaved_raw_and_batch_io(self, tmp_path):
+        """device_to_file / read_file_to_device interleaved with the framed
+        pipeline on the same two DMA streams must not reorder."""
+        import torch
+        g = _gpu_engine()
+        raw_path = str(tmp_path / "raw.bin")
+        rec_path = str(tmp_path / "rec.tfrecord")
+        batch = make_batch(1000, seed=3)
+        for it in range(10):
+            raw = torch.arange(
+                it + 1, dtype=torch.float32, device="cuda").view(torch.uint8)
+            g.device_to_file(raw.contiguous(), raw_path)
+            g.write_batch_to_file(g.batch_to_device(batch), rec_path, "Example")
+            back = g.read_file_to_device(raw_path)
+            assert torch.equal(back.cpu(), raw.cpu())
+            out = g.batch_to_host(g.read_file_to_batch_pipelined(
+                rec_path, batch.schema, "Example"))
+            assert_batches_equal(batch, out)
+
+
+class TestInferenceLimits:
+    def test_5000_distinct_names_grows_table(self):
+        """>1024 names per table half forces the ERR_OVERFLOW retry with a
+        grown table (2048-slot hard cap lifted)."""
+        import torch
+        g = _gpu_engine()
+        F, R = 5000, 2
+        schema = stf.StructType([
+            stf.StructField(f"feat_{i:05d}", stf.LongType(), True)
+            for i in range(F)])
+        cols = [column_from_values(np.arange(R, dtype=np.int64),
+                                   stf.LongType(), True, f"feat_{i:05d}")
+                for i in range(F)]
+        img = cpu_engine.encode_batch(RecordBatch(schema, cols, R), "Example")
+        data = torch.frombuffer(bytearray(img), dtype=torch.uint8).cuda()
+        off, lens = g.scan_frames_device(data)
+        codes = g.infer_codes_device(data, off, lens, "Example")
+        assert len(codes) == F
+        assert set(codes.values()) == {1}  # all long scalars
+        assert sorted(codes)[0] == "feat_00000"
+
+
+class TestErrorIndices:
+    def test_kind_mismatch_reports_record_index(self):
+        """Record 3 carries a float where the schema says int64: the decode
+        error must name a record index."""
+        import torch
+        schema_l = stf.StructType([stf.StructField("x", stf.LongType(), True)])
+        schema_f = stf.StructType([stf.StructField("x", stf.FloatType(), True)])
+        good = cpu_engine.encode_batch(RecordBatch(schema_l, [
+            column_from_values(np.arange(3, dtype=np.int64), stf.LongType(),
+                               True, "x")], 3), "Example")
+        bad = cpu_engine.encode_batch(RecordBatch(schema_f, [
+            column_from_values(np.array([1.5], np.float32), stf.FloatType(),
+                               True, "x")], 1), "Example")
+        g = _gpu_engine()
+        with pytest.raises(RuntimeError, match=r"record 3"):
+            g.decode_buffer_to_cpu(np.frombuffer(good + bad, np.uint8),
+                                   schema_l, "Example", verify_crc=False)

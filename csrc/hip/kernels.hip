@@ -329,6 +329,461 @@ __global__ void gather_payloads_wave_kernel(const u8* __restrict__ data,
     copy_wave(out + dst_off[r], data + off[r], len[r]);
 }
 
+// ---------------------------------------------------------------------------
+// Wave-cooperative PROTO decode/encode for large records (roadmap r1 #5).
+// One-lane-per-record starves the chip on big records because every payload
+// byte funnels through one lane's serial CRC/copy loops. For records above
+// kWaveRecordBytes a whole wave owns one record:
+//  - structure walk (tags, lens, keys): ALL lanes execute it in lockstep —
+//    the loads are wave-uniform (same address), so they broadcast from one
+//    cache-line fetch, and keeping control flow uniform lets the walk call
+//    wave-cooperative helpers at the value runs;
+//  - value runs, which carry ~all the bytes of a big record, are processed
+//    by all 64 lanes: packed floats / string bytes via lane-strided u64
+//    copies, packed int64 via a parallel varint decode (per-lane terminator
+//    counts + a shfl exclusive scan give each lane its output slots);
+//  - the frame CRC is crc32c_wave (64 chunk CRCs folded with the GF(2)
+//    shift operator) instead of the one-lane fused cursor.
+// ---------------------------------------------------------------------------
+
+// Inclusive shfl scan; returns this lane's inclusive prefix of v.
+__device__ inline i64 wave_incl_scan(i64 v) {
+  int lane = threadIdx.x & 63;
+  for (int d = 1; d < 64; d <<= 1) {
+    i64 y = __shfl_up(v, d);
+    if (lane >= d) v += y;
+  }
+  return v;
+}
+
+// Parallel decode of one packed-int64 run [p, p+n) into out[base...].
+// Varint starts are positions q with q==0 or MSB-clear at q-1; each lane
+// owns the values STARTING in its contiguous chunk (a varint may spill into
+// the next lane's chunk — reads cross, ownership doesn't). Returns the
+// total value count (uniform across the wave).
+__device__ inline i64 wave_extract_i64_packed(const u8* __restrict__ p, i64 n,
+                                              i64* __restrict__ out, i64 base) {
+  int lane = threadIdx.x & 63;
+  i64 chunk = (n + 63) / 64;
+  i64 lo = (i64)lane * chunk;
+  i64 hi = lo + chunk < n ? lo + chunk : n;
+  i64 cnt = 0;
+  for (i64 i = lo; i < hi; ++i) cnt += !(p[i] & 0x80);
+  i64 incl = wave_incl_scan(cnt);
+  i64 excl = incl - cnt;
+  i64 total = __shfl(incl, 63);
+  // first varint start in my chunk (skip a varint spanning in from the left)
+  i64 q = lo;
+  i64 pre = 0;
+  if (lo > 0 && lo < hi) {
+    if (p[lo - 1] & 0x80) {
+      while (q < hi && (p[q] & 0x80)) ++q;
+      if (q < hi) {
+        ++q;  // consumed the spanning varint's terminator
+        pre = 1;
+      }
+    }
+  }
+  i64 idx = base + excl + pre;
+  while (q < hi) {
+    u64 v = 0;
+    int shift = 0;
+    i64 j = q;
+    while (j < n && shift < 70) {
+      u8 b = p[j++];
+      v |= (u64)(b & 0x7F) << shift;
+      shift += 7;
+      if (!(b & 0x80)) break;
+    }
+    out[idx++] = (i64)v;
+    q = j;
+  }
+  return total;
+}
+
+// Wave variant of extract_list_body: all lanes walk in lockstep, value runs
+// go cooperative. vi/bi stay register-uniform (every lane computes the same
+// updates).
+__device__ inline int32_t extract_list_body_wave(const u8* __restrict__ p,
+                                                 const u8* end, int32_t kind,
+                                                 const DecodeDst& dst, i64* vi,
+                                                 i64* bi) {
+  int lane = threadIdx.x & 63;
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    if (fieldno != 1) {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+      continue;
+    }
+    if (kind == KIND_BYTES) {
+      u64 len;
+      p = read_varint(p, end, &len);
+      if (!p || (u64)(end - p) < len) return ERR_TRUNCATED;
+      copy_wave(dst.bytes_data + *bi, p, (i64)len);
+      if (lane == 0) dst.elem_len[*vi] = (i64)len;
+      *bi += (i64)len;
+      *vi += 1;
+      p += len;
+    } else if (kind == KIND_FLOAT) {
+      if (wt == 2) {
+        u64 len;
+        p = read_varint(p, end, &len);
+        if (!p || (u64)(end - p) < len) return ERR_TRUNCATED;
+        u64 nv = len / 4;
+        copy_wave((u8*)(dst.f32_vals + *vi), p, (i64)(nv * 4));
+        *vi += (i64)nv;
+        p += len;
+      } else {
+        if (end - p < 4) return ERR_TRUNCATED;
+        if (lane == 0) {
+          float v;
+          __builtin_memcpy(&v, p, 4);
+          dst.f32_vals[*vi] = v;
+        }
+        (*vi)++;
+        p += 4;
+      }
+    } else {  // INT64
+      if (wt == 2) {
+        u64 len;
+        p = read_varint(p, end, &len);
+        if (!p || (u64)(end - p) < len) return ERR_TRUNCATED;
+        *vi += wave_extract_i64_packed(p, (i64)len, dst.i64_vals, *vi);
+        p += len;
+      } else {
+        u64 v;
+        p = read_varint(p, end, &v);
+        if (!p) return ERR_BAD_VARINT;
+        if (lane == 0) dst.i64_vals[*vi] = (i64)v;
+        (*vi)++;
+      }
+    }
+  }
+  return ERR_OK;
+}
+
+__device__ inline int32_t extract_feature_body_wave(const u8* __restrict__ p,
+                                                    const u8* end, int32_t kind,
+                                                    const DecodeDst& dst,
+                                                    i64* vi, i64* bi) {
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    if (fieldno >= 1 && fieldno <= 3 && wt == 2) {
+      u64 len;
+      p = read_varint(p, end, &len);
+      if (!p || (u64)(end - p) < len) return ERR_TRUNCATED;
+      int32_t rc = extract_list_body_wave(p, p + len, kind, dst, vi, bi);
+      if (rc != ERR_OK) return rc;
+      p += len;
+    } else {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+    }
+  }
+  return ERR_OK;
+}
+
+__device__ inline int32_t extract_field_wave(const u8* __restrict__ data,
+                                             i64 pos, i64 len, int32_t kind,
+                                             int32_t is_seq, const DecodeDst& dst,
+                                             i64 val_base, i64 byte_base,
+                                             i64 list_base) {
+  if (pos < 0) return ERR_OK;
+  int lane = threadIdx.x & 63;
+  const u8* p = data + pos;
+  const u8* end = p + len;
+  i64 vi = val_base;
+  i64 bi = byte_base;
+  if (!is_seq) return extract_feature_body_wave(p, end, kind, dst, &vi, &bi);
+  i64 li = list_base;
+  while (p < end) {
+    u64 tag;
+    p = read_varint(p, end, &tag);
+    if (!p) return ERR_BAD_VARINT;
+    u32 fieldno = (u32)(tag >> 3), wt = (u32)(tag & 7);
+    if (fieldno == 1 && wt == 2) {
+      u64 flen;
+      p = read_varint(p, end, &flen);
+      if (!p || (u64)(end - p) < flen) return ERR_TRUNCATED;
+      i64 v_before = vi;
+      int32_t rc = extract_feature_body_wave(p, p + flen, kind, dst, &vi, &bi);
+      if (rc != ERR_OK) return rc;
+      if (lane == 0) dst.sub_count[li] = vi - v_before;
+      ++li;
+      p += flen;
+    } else {
+      p = skip_field(p, end, wt);
+      if (!p) return ERR_TRUNCATED;
+    }
+  }
+  return ERR_OK;
+}
+
+// One record per WAVE structure scan: lane 0 runs the (cheap, structure-only)
+// two-pass scan; the payload CRC — the byte-proportional part — runs
+// wave-cooperatively. Used when records are large and few.
+__global__ void scan_records_wave_kernel(const u8* __restrict__ data,
+                                         const i64* __restrict__ off,
+                                         const i64* __restrict__ len, i64 R,
+                                         int32_t fmt,
+                                         const u8* __restrict__ schema_blob,
+                                         int F, FieldStat* __restrict__ stats,
+                                         int32_t* __restrict__ err,
+                                         unsigned long long* crc_err) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  SchemaView schema = schema_view(schema_blob);
+  i64 wave = (blockIdx.x * (i64)blockDim.x + threadIdx.x) / 64;
+  i64 nwaves = ((i64)gridDim.x * blockDim.x) / 64;
+  int lane = threadIdx.x & 63;
+  for (i64 r = wave; r < R; r += nwaves) {
+    if (lane == 0) {
+      FieldStat* st = stats + r * F;
+      for (int f = 0; f < F; ++f) field_stat_clear(&st[f]);
+      int32_t rc = scan_record(data, off[r], len[r], fmt, schema, st);
+      if (rc != ERR_OK) set_err(err, rc, r);
+      for (int f = 0; f < F; ++f)
+        if (st[f].err != ERR_OK) set_err(err, st[f].err, r);
+    }
+    if (crc_err) {
+      const u8* h = data + off[r] - 12;
+      u32 payload = crc32c_wave(h + 12, len[r], tab);
+      if (lane == 0) {
+        u32 len_crc, data_crc;
+        __builtin_memcpy(&len_crc, h + 8, 4);
+        __builtin_memcpy(&data_crc, h + 12 + len[r], 4);
+        bool ok = mask_crc(crc32c_sw(h, 8, 0, tab)) == len_crc &&
+                  mask_crc(payload) == data_crc;
+        if (!ok) atomicMin(crc_err, (unsigned long long)(r + 1));
+      }
+    }
+  }
+}
+
+// One (record, field) pair per WAVE value extraction.
+__global__ void extract_fields_wave_kernel(const u8* __restrict__ data, i64 R,
+                                           int F,
+                                           const FieldStat* __restrict__ stats,
+                                           const DevFieldDst* __restrict__ metas,
+                                           int32_t* __restrict__ err) {
+  i64 total = R * F;
+  i64 wave = (blockIdx.x * (i64)blockDim.x + threadIdx.x) / 64;
+  i64 nwaves = ((i64)gridDim.x * blockDim.x) / 64;
+  for (i64 idx = wave; idx < total; idx += nwaves) {
+    i64 r = idx / F;
+    int f = (int)(idx - r * F);
+    const DevFieldDst& m = metas[f];
+    const FieldStat& st = stats[r * F + f];
+    DecodeDst dst{m.i64_vals, m.f32_vals, m.bytes_data, m.elem_len, m.sub_count};
+    int32_t rc = extract_field_wave(data, st.pos, st.len, m.kind, m.is_seq, dst,
+                                    m.val_base ? m.val_base[r] : 0,
+                                    m.byte_base ? m.byte_base[r] : 0,
+                                    m.list_base ? m.list_base[r] : 0);
+    if (rc != ERR_OK && (threadIdx.x & 63) == 0) set_err(err, rc, r);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Wave-cooperative EMIT for large records. Structure bytes are stored by
+// lane 0 (all lanes track the cursor in registers); value runs are emitted
+// by the whole wave; the frame CRC re-reads the emitted payload wave-wide
+// (L2-hot) instead of the one-lane fused WriteCur.
+// ---------------------------------------------------------------------------
+
+// Emits a packed-int64 run's varints in parallel: lane l takes values
+// [v0 + l*chunk ...), sums its varint sizes, a shfl scan places its output
+// cursor, then it emits serially. Returns total packed bytes (uniform).
+__device__ inline i64 wave_emit_i64_packed(u8* __restrict__ o,
+                                           const i64* __restrict__ vals,
+                                           i64 v0, i64 v1) {
+  int lane = threadIdx.x & 63;
+  i64 n = v1 - v0;
+  i64 chunk = (n + 63) / 64;
+  i64 lo = v0 + (i64)lane * chunk;
+  i64 hi = lo + chunk < v1 ? lo + chunk : v1;
+  i64 sz = 0;
+  for (i64 v = lo; v < hi; ++v) sz += varint_size((u64)vals[v]);
+  i64 incl = wave_incl_scan(sz);
+  u8* q = o + (incl - sz);
+  for (i64 v = lo; v < hi; ++v) q = write_varint(q, (u64)vals[v]);
+  return __shfl(incl, 63);
+}
+
+struct WaveCur {
+  u8* base;
+  i64 pos;  // uniform across the wave
+};
+
+__device__ inline void wvc_put(WaveCur& w, u8 b) {
+  if ((threadIdx.x & 63) == 0) w.base[w.pos] = b;
+  w.pos += 1;
+}
+
+__device__ inline void wvc_varint(WaveCur& w, u64 v) {
+  if ((threadIdx.x & 63) == 0) write_varint(w.base + w.pos, v);
+  w.pos += varint_size(v);
+}
+
+__device__ inline void wvc_bytes(WaveCur& w, const u8* s, i64 n) {
+  copy_wave(w.base + w.pos, s, n);
+  w.pos += n;
+}
+
+__device__ inline void wvc_list_body(WaveCur& w, const FieldColumn& c, i64 v0,
+                                     i64 v1) {
+  if (c.kind == KIND_FLOAT) {
+    i64 n = v1 - v0;
+    if (n) {
+      wvc_put(w, 0x0A);
+      wvc_varint(w, (u64)(4 * n));
+      wvc_bytes(w, (const u8*)(c.f32_vals + v0), 4 * n);
+    }
+  } else if (c.kind == KIND_INT64) {
+    i64 packed = 0;
+    for (i64 v = v0; v < v1; ++v) packed += varint_size((u64)c.i64_vals[v]);
+    if (packed) {
+      wvc_put(w, 0x0A);
+      wvc_varint(w, (u64)packed);
+      w.pos += wave_emit_i64_packed(w.base + w.pos, c.i64_vals, v0, v1);
+    }
+  } else {
+    for (i64 v = v0; v < v1; ++v) {
+      i64 b0 = c.elem_off[v];
+      i64 blen = c.elem_off[v + 1] - b0;
+      wvc_put(w, 0x0A);
+      wvc_varint(w, (u64)blen);
+      wvc_bytes(w, c.bytes_data + b0, blen);
+    }
+  }
+}
+
+__device__ inline void wvc_feature_body(WaveCur& w, const FieldColumn& c,
+                                        i64 v0, i64 v1) {
+  i64 body = list_body_size(c, v0, v1);
+  wvc_put(w, (u8)((c.kind << 3) | 2));
+  wvc_varint(w, (u64)body);
+  wvc_list_body(w, c, v0, v1);
+}
+
+__device__ inline void wvc_features_entry(WaveCur& w, const FieldColumn& c,
+                                          const SchemaView& s, int f, i64 v0,
+                                          i64 v1) {
+  i64 fb = feature_body_size(c, v0, v1);
+  i64 klen = s.name_len(f);
+  wvc_put(w, 0x0A);
+  wvc_varint(w, (u64)klen);
+  wvc_bytes(w, s.name(f), klen);
+  wvc_put(w, 0x12);
+  wvc_varint(w, (u64)fb);
+  wvc_feature_body(w, c, v0, v1);
+}
+
+__device__ inline void wvc_feature_lists_entry(WaveCur& w, const FieldColumn& c,
+                                               const SchemaView& s, int f,
+                                               i64 r) {
+  i64 flb = feature_list_body_size(c, r);
+  i64 klen = s.name_len(f);
+  wvc_put(w, 0x0A);
+  wvc_varint(w, (u64)klen);
+  wvc_bytes(w, s.name(f), klen);
+  wvc_put(w, 0x12);
+  wvc_varint(w, (u64)flb);
+  for (i64 j = c.list_off[r]; j < c.list_off[r + 1]; ++j) {
+    i64 fb = feature_body_size(c, c.sub_off[j], c.sub_off[j + 1]);
+    wvc_put(w, 0x0A);
+    wvc_varint(w, (u64)fb);
+    wvc_feature_body(w, c, c.sub_off[j], c.sub_off[j + 1]);
+  }
+}
+
+__device__ inline i64 emit_record_payload_wave(u8* o, const FieldColumn* cols,
+                                               const SchemaView& s, int32_t fmt,
+                                               i64 r) {
+  i64 ctx_body = 0, fl_body = 0;
+  for (int f = 0; f < s.nfields; ++f) {
+    const FieldColumn& c = cols[f];
+    if (!c.presence[r]) continue;
+    if (!c.is_seq) {
+      i64 e = features_entry_size(c, s, f, c.row_off[r], c.row_off[r + 1]);
+      ctx_body += 1 + varint_size((u64)e) + e;
+    } else {
+      i64 e = feature_lists_entry_size(c, s, f, r);
+      fl_body += 1 + varint_size((u64)e) + e;
+    }
+  }
+  WaveCur w{o, 0};
+  auto emit_ctx = [&]() {
+    for (int f = 0; f < s.nfields; ++f) {
+      const FieldColumn& c = cols[f];
+      if (!c.presence[r] || c.is_seq) continue;
+      i64 e = features_entry_size(c, s, f, c.row_off[r], c.row_off[r + 1]);
+      wvc_put(w, 0x0A);
+      wvc_varint(w, (u64)e);
+      wvc_features_entry(w, c, s, f, c.row_off[r], c.row_off[r + 1]);
+    }
+  };
+  auto emit_fl = [&]() {
+    for (int f = 0; f < s.nfields; ++f) {
+      const FieldColumn& c = cols[f];
+      if (!c.presence[r] || !c.is_seq) continue;
+      i64 e = feature_lists_entry_size(c, s, f, r);
+      wvc_put(w, 0x0A);
+      wvc_varint(w, (u64)e);
+      wvc_feature_lists_entry(w, c, s, f, r);
+    }
+  };
+  if (fmt == FMT_EXAMPLE) {
+    wvc_put(w, 0x0A);
+    wvc_varint(w, (u64)ctx_body);
+    emit_ctx();
+  } else {
+    if (ctx_body) {
+      wvc_put(w, 0x0A);
+      wvc_varint(w, (u64)ctx_body);
+      emit_ctx();
+    }
+    wvc_put(w, 0x12);
+    wvc_varint(w, (u64)fl_body);
+    emit_fl();
+  }
+  return w.pos;
+}
+
+// One record per WAVE emit: payload cooperatively, then wave CRC of the
+// freshly-stored (L2-hot) payload, then lane 0 writes the frame header.
+__global__ void emit_records_wave_kernel(const FieldColumn* __restrict__ cols,
+                                         const u8* __restrict__ schema_blob,
+                                         int32_t fmt, i64 r0, i64 R,
+                                         const i64* __restrict__ frame_off,
+                                         u8* __restrict__ file,
+                                         int32_t* __restrict__ err) {
+  __shared__ uint32_t tab[8][256];
+  stage_crc_tables(tab);
+  SchemaView schema = schema_view(schema_blob);
+  i64 wave = (blockIdx.x * (i64)blockDim.x + threadIdx.x) / 64;
+  i64 nwaves = ((i64)gridDim.x * blockDim.x) / 64;
+  int lane = threadIdx.x & 63;
+  for (i64 r = r0 + wave; r < R; r += nwaves) {
+    i64 payload = (frame_off[r + 1] - frame_off[r]) - kFrameOverhead;
+    u8* o = file + frame_off[r] + 12;
+    i64 emitted = emit_record_payload_wave(o, cols, schema, fmt, r);
+    if (emitted != payload && lane == 0) set_err(err, ERR_OVERFLOW, r);
+    // make every lane's payload stores visible before other lanes re-read
+    // them for the CRC (lockstep wave => the fence alone is the barrier)
+    __threadfence_block();
+    u32 crc = crc32c_wave(o, payload, tab);
+    if (lane == 0)
+      write_frame_header_footer_crc(file, frame_off[r], payload, crc, tab);
+  }
+}
+
 // ByteArray framing: payload r occupies [elem_off[r], elem_off[r+1]) of src.
 // The copy streams through a WriteCur so the frame CRC comes from the same
 // register windows the stores use (no payload re-read).
@@ -770,18 +1225,27 @@ void gpu_crc_verify(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
 
 void gpu_scan_records(uintptr_t data, uintptr_t off, uintptr_t len, i64 R,
                       int32_t fmt, uintptr_t schema_blob, int F, uintptr_t stats,
-                      uintptr_t err, uintptr_t crc_err, uintptr_t stream) {
-  hipLaunchKernelGGL(scan_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const u8*)data, (const i64*)off,
-                     (const i64*)len, R, fmt, (const u8*)schema_blob, F,
-                     (FieldStat*)stats, (int32_t*)err,
-                     (unsigned long long*)crc_err);
+                      uintptr_t err, uintptr_t crc_err, uintptr_t stream,
+                      i64 avg_bytes) {
+  if (avg_bytes > kWaveRecordBytes) {  // few big records: one per wave
+    hipLaunchKernelGGL(scan_records_wave_kernel, dim3(grid_for(R * 64)),
+                       dim3(kBlock), 0, (hipStream_t)stream, (const u8*)data,
+                       (const i64*)off, (const i64*)len, R, fmt,
+                       (const u8*)schema_blob, F, (FieldStat*)stats,
+                       (int32_t*)err, (unsigned long long*)crc_err);
+  } else {
+    hipLaunchKernelGGL(scan_records_kernel, dim3(grid_for(R)), dim3(kBlock), 0,
+                       (hipStream_t)stream, (const u8*)data, (const i64*)off,
+                       (const i64*)len, R, fmt, (const u8*)schema_blob, F,
+                       (FieldStat*)stats, (int32_t*)err,
+                       (unsigned long long*)crc_err);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
 void gpu_extract_fields(uintptr_t data, i64 R, int F, uintptr_t stats,
                         py::list metas, uintptr_t meta_dev, uintptr_t err,
-                        uintptr_t stream) {
+                        uintptr_t stream, i64 avg_bytes) {
   std::vector<DevFieldDst> host_metas(F);
   for (int f = 0; f < F; ++f) {
     py::dict d = metas[f].cast<py::dict>();
@@ -800,10 +1264,17 @@ void gpu_extract_fields(uintptr_t data, i64 R, int F, uintptr_t stats,
   HIP_CHECK(hipMemcpyAsync((void*)meta_dev, host_metas.data(),
                            sizeof(DevFieldDst) * F, hipMemcpyHostToDevice,
                            (hipStream_t)stream));
-  hipLaunchKernelGGL(extract_fields_kernel, dim3(grid_for(R * F)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const u8*)data, R, F,
-                     (const FieldStat*)stats, (const DevFieldDst*)meta_dev,
-                     (int32_t*)err);
+  if (avg_bytes > kWaveRecordBytes) {
+    hipLaunchKernelGGL(extract_fields_wave_kernel, dim3(grid_for(R * F * 64)),
+                       dim3(kBlock), 0, (hipStream_t)stream, (const u8*)data,
+                       R, F, (const FieldStat*)stats,
+                       (const DevFieldDst*)meta_dev, (int32_t*)err);
+  } else {
+    hipLaunchKernelGGL(extract_fields_kernel, dim3(grid_for(R * F)),
+                       dim3(kBlock), 0, (hipStream_t)stream, (const u8*)data,
+                       R, F, (const FieldStat*)stats,
+                       (const DevFieldDst*)meta_dev, (int32_t*)err);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -845,11 +1316,20 @@ void gpu_size_records(py::list cols, uintptr_t cols_dev, uintptr_t schema_blob,
 
 void gpu_emit_records(uintptr_t cols_dev, uintptr_t schema_blob, int32_t fmt,
                       i64 r0, i64 R, uintptr_t frame_off, uintptr_t file,
-                      uintptr_t err, uintptr_t stream) {
-  hipLaunchKernelGGL(emit_records_kernel, dim3(grid_for(R - r0)), dim3(kBlock), 0,
-                     (hipStream_t)stream, (const FieldColumn*)cols_dev,
-                     (const u8*)schema_blob, fmt, r0, R, (const i64*)frame_off,
-                     (u8*)file, (int32_t*)err);
+                      uintptr_t err, uintptr_t stream, i64 avg_bytes) {
+  if (avg_bytes > kWaveRecordBytes) {
+    hipLaunchKernelGGL(emit_records_wave_kernel, dim3(grid_for((R - r0) * 64)),
+                       dim3(kBlock), 0, (hipStream_t)stream,
+                       (const FieldColumn*)cols_dev, (const u8*)schema_blob,
+                       fmt, r0, R, (const i64*)frame_off, (u8*)file,
+                       (int32_t*)err);
+  } else {
+    hipLaunchKernelGGL(emit_records_kernel, dim3(grid_for(R - r0)),
+                       dim3(kBlock), 0, (hipStream_t)stream,
+                       (const FieldColumn*)cols_dev, (const u8*)schema_blob,
+                       fmt, r0, R, (const i64*)frame_off, (u8*)file,
+                       (int32_t*)err);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -1232,10 +1712,19 @@ void register_gpu(py::module_& m) {
   m.def("gpu_crc_verify", &gpu_crc_verify, py::arg("data"), py::arg("off"),
         py::arg("len"), py::arg("R"), py::arg("err_out"), py::arg("stream"),
         py::arg("avg_bytes") = 0);
-  m.def("gpu_scan_records", &gpu_scan_records);
-  m.def("gpu_extract_fields", &gpu_extract_fields);
+  m.def("gpu_scan_records", &gpu_scan_records, py::arg("data"), py::arg("off"),
+        py::arg("len"), py::arg("R"), py::arg("fmt"), py::arg("schema_blob"),
+        py::arg("F"), py::arg("stats"), py::arg("err"), py::arg("crc_err"),
+        py::arg("stream"), py::arg("avg_bytes") = 0);
+  m.def("gpu_extract_fields", &gpu_extract_fields, py::arg("data"),
+        py::arg("R"), py::arg("F"), py::arg("stats"), py::arg("metas"),
+        py::arg("meta_dev"), py::arg("err"), py::arg("stream"),
+        py::arg("avg_bytes") = 0);
   m.def("gpu_size_records", &gpu_size_records);
-  m.def("gpu_emit_records", &gpu_emit_records);
+  m.def("gpu_emit_records", &gpu_emit_records, py::arg("cols_dev"),
+        py::arg("schema_blob"), py::arg("fmt"), py::arg("r0"), py::arg("R"),
+        py::arg("frame_off"), py::arg("file"), py::arg("err"),
+        py::arg("stream"), py::arg("avg_bytes") = 0);
   m.def("gpu_frame_bytes", &gpu_frame_bytes, py::arg("src"),
         py::arg("elem_off"), py::arg("frame_off"), py::arg("R"),
         py::arg("file"), py::arg("stream"), py::arg("avg_bytes") = 0);

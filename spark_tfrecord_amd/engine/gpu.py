@@ -467,6 +467,8 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
             r0 = K
         else:
             err1 = None
+    # average record size steers the one-per-lane vs one-per-wave kernels
+    avg_bytes = max(0, (data.numel() // R) - 16) if R else 0
     if R > r0:
         stride = F * 6 * 8
         _native.gpu_scan_records(data.data_ptr(), off.data_ptr() + r0 * 8,
@@ -475,7 +477,7 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
                                  stats.data_ptr() + r0 * stride,
                                  err.data_ptr(),
                                  crc_err.data_ptr() if fuse_crc else 0,
-                                 _stream())
+                                 _stream(), avg_bytes)
     crc_err_pre = None
     if fuse_crc and r0 > 0:
         # prescanned rows skipped the fused path: verify them separately.
@@ -560,7 +562,8 @@ def decode_device(data: torch.Tensor, off: torch.Tensor, lens: torch.Tensor,
     meta_dev = torch.empty(F * _native.gpu_devmeta_bytes(), dtype=torch.uint8,
                            device=device)
     _native.gpu_extract_fields(data.data_ptr(), R, F, stats.data_ptr(), metas,
-                               meta_dev.data_ptr(), err.data_ptr(), _stream())
+                               meta_dev.data_ptr(), err.data_ptr(), _stream(),
+                               avg_bytes)
 
     cols: List[WireColumn] = []
     for i, (f, o) in enumerate(zip(fields, outs)):
@@ -664,7 +667,8 @@ def encode_device(batch: RecordBatch, record_type: str) -> torch.Tensor:
     err = torch.zeros(2, dtype=torch.int32, device=device)
     _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], 0, R, frame_off.data_ptr(),
-                             file.data_ptr(), err.data_ptr(), _stream())
+                             file.data_ptr(), err.data_ptr(), _stream(),
+                             total // max(R, 1))
     _check_emit_err(err)
     return file
 
@@ -702,7 +706,8 @@ def write_batch_to_file(batch: RecordBatch, path: str,
     if not pinned:
         _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                                  FMT[record_type], 0, R, frame_off.data_ptr(),
-                                 file.data_ptr(), err.data_ptr(), _stream())
+                                 file.data_ptr(), err.data_ptr(), _stream(),
+                                 total // R)
         _check_emit_err(err)
         _write_file_staged(file, path)
         return total
@@ -712,7 +717,7 @@ def write_batch_to_file(batch: RecordBatch, path: str,
         _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                                  FMT[record_type], ridx[s], ridx[s + 1],
                                  frame_off.data_ptr(), file.data_ptr(),
-                                 err.data_ptr(), _stream())
+                                 err.data_ptr(), _stream(), total // R)
         b0, b1 = int(bounds[s]), int(bounds[s + 1])
         if b1 > b0:
             w = streams[s % len(streams)]
@@ -894,7 +899,8 @@ def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,
     err = torch.zeros(2, dtype=torch.int32, device=device)
     _native.gpu_emit_records(cols_dev.data_ptr(), blob.data_ptr(),
                              FMT[record_type], 0, R, frame_off.data_ptr(),
-                             file.data_ptr(), err.data_ptr(), _stream())
+                             file.data_ptr(), err.data_ptr(), _stream(),
+                             total // max(R, 1))
     codes = torch.as_tensor(np.ascontiguousarray(part_codes, np.int64),
                             device=device)
     order = torch.argsort(codes, stable=True)

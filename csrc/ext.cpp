@@ -777,6 +777,9 @@ PYBIND11_MODULE(_native, m) {
   m.def("crc32c_combine",
         [](u32 c1, u32 c2, u64 len2) { return crc32c_combine(c1, c2, len2); },
         "crc(A||B) from crc(A), crc(B), len(B) (GF(2) shift operator)");
+  m.def("crc32c_combine_fast",
+        [](u32 c1, u32 c2, u64 len2) { return crc32c_combine_fast(c1, c2, len2); },
+        "crc(A||B) via register-only GF(2^32) field multiply (wave-CRC form)");
   m.def("scan_frames", &scan_frames, py::arg("data"), py::arg("verify_crc") = true,
         "Scan TFRecord frames -> (payload offsets, payload lengths)");
   m.def("scan_frame_headers", &scan_frame_headers, py::arg("data"),

@@ -247,16 +247,18 @@ __device__ inline u32 crc32c_wave(const u8* p, i64 n,
   u32 acc = __shfl(my, 0);
   if (chunk < n) {
     // every lane folds redundantly in lockstep (uniform control flow, no
-    // divergence around the cross-lane reads); acc ends identical wave-wide
-    CrcMat op;
-    crc32c_shift_op(op, (u64)chunk);
+    // divergence around the cross-lane reads); acc ends identical
+    // wave-wide. The fold is the register-only GF(2^32) field form —
+    // the CrcMat version kept 128 B matrices in per-lane SCRATCH and made
+    // this fold 30x more expensive than the chunk CRCs themselves.
+    u32 op = crc32c_shift_elem((u64)chunk);
     for (int i = 1; i < 64; ++i) {
       i64 si = (i64)i * chunk;
       if (si >= n) break;
       u32 ci = __shfl(my, i);
       i64 li = (si + chunk < n ? chunk : n - si);
-      acc = (li == chunk) ? (crcmat_times(op, acc) ^ ci)
-                          : crc32c_combine(acc, ci, (u64)li);
+      acc = (li == chunk) ? (crc_gfmul(acc, op) ^ ci)
+                          : crc32c_combine_fast(acc, ci, (u64)li);
     }
   }
   return acc;

@@ -68,31 +68,97 @@ def codec_from_path(path: str) -> Optional[str]:
 # Gzip output is written as a SINGLE standard gzip member whose deflate
 # stream carries Z_FULL_FLUSH sync points every _GZ_SEGMENT of input (each
 # flush resets the dictionary and byte-aligns with the 00 00 FF FF empty
-# stored block, pigz-style). Any gzip reader — TensorFlow included — decodes
-# the file normally; OUR reader finds the sync markers and inflates the
-# segments in parallel, verifying the member's CRC32 trailer (false-positive
-# markers fall back to sequential inflate). Matches the reference's
-# isSplitable=false model: gzip files still read as whole files, just on
-# more than one core.
-_GZ_SEGMENT = 4 << 20
+# stored block, pigz-style), and whose header carries an FEXTRA subfield
+# ('T','S') listing every segment's (compressed, uncompressed) length.
+# Any gzip reader — TensorFlow included — decodes the file normally (extra
+# fields are skipped per RFC 1952); OUR readers get exact segment extents:
+# the host inflates segments on a thread pool, the GPU engine inflates all
+# segments in one kernel launch (one segment per lane, csrc/hip/inflate.hip)
+# with output offsets known up front. Table-less gzip (foreign files, or
+# > _GZ_MAX_SEGS segments) falls back to the marker scan, then to
+# sequential inflate. Matches the reference's isSplitable=false model:
+# gzip files still read as whole files, just on more than one core/CU.
+_GZ_SEGMENT = int(os.environ.get("TFREC_GZ_SEGMENT", 256 << 10))
 _GZ_MARK = b"\x00\x00\xff\xff"
+_GZ_MAX_SEGS = 8189  # FEXTRA payload cap: 65535 bytes / 8 per segment
+
+
+def _gz_segment_size(total: int) -> int:
+    """Segment size: the knob, grown so the table fits FEXTRA's 64 KiB."""
+    seg = _GZ_SEGMENT
+    while total > seg * _GZ_MAX_SEGS:
+        seg *= 2
+    return seg
 
 
 def compress_bytes(data: bytes, codec: Optional[str]) -> bytes:
     if codec is None:
         return data
     if codec == "gzip":
-        c = zlib.compressobj(6, zlib.DEFLATED, 16 + 15)  # gzip wrapper
-        out = []
-        for pos in range(0, len(data), _GZ_SEGMENT):
-            out.append(c.compress(data[pos:pos + _GZ_SEGMENT]))
-            if pos + _GZ_SEGMENT < len(data):
-                out.append(c.flush(zlib.Z_FULL_FLUSH))
-        out.append(c.flush())
-        return b"".join(out)
+        seg = _gz_segment_size(len(data))
+        c = zlib.compressobj(6, zlib.DEFLATED, -15)  # raw deflate body
+        chunks: List[bytes] = []
+        seg_lens: List[Tuple[int, int]] = []  # (comp_len, uncomp_len)
+        n = len(data)
+        pos = 0
+        while True:
+            hi = min(pos + seg, n)
+            body = c.compress(data[pos:hi])
+            body += (c.flush() if hi == n else c.flush(zlib.Z_FULL_FLUSH))
+            chunks.append(body)
+            seg_lens.append((len(body), hi - pos))
+            pos = hi
+            if pos >= n:
+                break
+        import struct as _struct
+
+        payload = _struct.pack("<BBH", 1, 0, len(seg_lens)) + b"".join(
+            _struct.pack("<II", c_, u_) for c_, u_ in seg_lens)
+        extra = b"TS" + _struct.pack("<H", len(payload)) + payload
+        hdr = (b"\x1f\x8b\x08\x04" + b"\x00\x00\x00\x00" + b"\x00\xff" +
+               _struct.pack("<H", len(extra)) + extra)
+        trailer = _struct.pack("<II", zlib.crc32(data) & 0xFFFFFFFF,
+                               n % (1 << 32))
+        return hdr + b"".join(chunks) + trailer
     if codec == "deflate":
         return zlib.compress(data, 6)
     raise ValueError(codec)
+
+
+def parse_gz_segments(raw: bytes):
+    """Parse our FEXTRA 'TS' segment table from a gzip blob.
+
+    Returns (body_off, [(comp_len, uncomp_len), ...], crc32, isize) or None
+    when the blob is not a single-member gzip with our table."""
+    import struct as _struct
+
+    if len(raw) < 20 or raw[:3] != b"\x1f\x8b\x08" or not (raw[3] & 0x04):
+        return None
+    flg = raw[3]
+    if flg & ~0x04:  # any flag other than FEXTRA shifts fields we don't walk
+        return None
+    xlen = int.from_bytes(raw[10:12], "little")
+    extra = raw[12:12 + xlen]
+    body_off = 12 + xlen
+    pos = 0
+    while pos + 4 <= len(extra):
+        si, ln = extra[pos:pos + 2], int.from_bytes(extra[pos + 2:pos + 4],
+                                                    "little")
+        sub = extra[pos + 4:pos + 4 + ln]
+        pos += 4 + ln
+        if si != b"TS" or len(sub) < 4:
+            continue
+        ver, _, nseg = _struct.unpack("<BBH", sub[:4])
+        if ver != 1 or len(sub) < 4 + 8 * nseg:
+            return None
+        segs = [_struct.unpack("<II", sub[4 + 8 * i:12 + 8 * i])
+                for i in range(nseg)]
+        crc = int.from_bytes(raw[-8:-4], "little")
+        isize = int.from_bytes(raw[-4:], "little")
+        if sum(c_ for c_, _ in segs) != len(raw) - body_off - 8:
+            return None  # truncated/concatenated: not a clean single member
+        return body_off, segs, crc, isize
+    return None
 
 
 def _gunzip_parallel(raw: bytes) -> Optional[bytes]:

@@ -148,3 +148,22 @@ class TestCrcCombine:
                 _native.crc32c(np.frombuffer(a, np.uint8)),
                 _native.crc32c(np.frombuffer(b, np.uint8)), len(b))
             assert got == want
+
+
+class TestCrcCombineFast:
+    def test_field_form_matches_matrix_and_direct(self):
+        """crc32c_combine_fast (register-only GF(2^32) multiply, the wave
+        kernels' fold) must equal the matrix form and the direct CRC of the
+        concatenation for arbitrary splits."""
+        import numpy as np
+        from spark_tfrecord_amd import _native
+
+        rng = np.random.default_rng(11)
+        for n in (0, 1, 2, 7, 8, 9, 63, 64, 65, 255, 4096, 100_001):
+            blob = rng.integers(0, 256, max(n, 1)).astype(np.uint8)[:n].tobytes()
+            for cut in {0, n // 3, n // 2, n - 1 if n else 0, n}:
+                a, b = blob[:cut], blob[cut:]
+                want = _native.crc32c(blob)
+                ca, cb = _native.crc32c(a), _native.crc32c(b)
+                assert _native.crc32c_combine(ca, cb, len(b)) == want
+                assert _native.crc32c_combine_fast(ca, cb, len(b)) == want

@@ -15,7 +15,8 @@ from pathlib import Path
 
 ROOT = Path(__file__).resolve().parent
 OUT = ROOT / "spark_tfrecord_amd" / "_native.so"
-SOURCES = [ROOT / "csrc" / "ext.cpp", ROOT / "csrc" / "hip" / "kernels.hip"]
+SOURCES = [ROOT / "csrc" / "ext.cpp", ROOT / "csrc" / "hip" / "kernels.hip",
+           ROOT / "csrc" / "hip" / "inflate.hip"]
 HEADERS = list((ROOT / "csrc").rglob("*.h"))
 
 

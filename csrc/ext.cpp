@@ -758,7 +758,8 @@ u32 masked_crc32c_py(py::buffer data) {
 
 }  // namespace
 
-void register_gpu(py::module_& m);  // defined in csrc/hip/kernels.hip
+void register_gpu(py::module_& m);      // defined in csrc/hip/kernels.hip
+void register_inflate(py::module_& m);  // defined in csrc/hip/inflate.hip
 
 PYBIND11_MODULE(_native, m) {
   m.doc() = "MI355X-native TFRecord codec (host + gfx950 kernels)";
@@ -800,6 +801,7 @@ PYBIND11_MODULE(_native, m) {
   m.attr("FMT_BYTE_ARRAY") = static_cast<int>(FMT_BYTE_ARRAY);
 #ifdef TFREC_WITH_HIP
   register_gpu(m);
+  register_inflate(m);
   m.attr("HAS_GPU_KERNELS") = true;
 #else
   m.attr("HAS_GPU_KERNELS") = false;

@@ -828,19 +828,115 @@ def read_file_to_batch_pipelined(path: str, schema: StructType, record_type: str
                          prescan=prescan)
 
 
-def read_files_to_batch(paths, schema: StructType, record_type: str,
-                        verify_crc: bool = True, device="cuda"):
-    """Decode MANY uncompressed files as ONE GPU pipeline: their images are
-    concatenated in HBM (TFRecord frames are concatenable, so the frame
-    chain spans file boundaries exactly), scanned and decoded once, and the
-    per-file row counts recovered with a searchsorted over frame offsets.
-    Returns (device RecordBatch, np.ndarray row_counts per file). Collapses
-    the per-file launch/sync overhead that dominates partitioned datasets
-    with many small files."""
+def gz_device_meta(path: str):
+    """Segment table of OUR gzip files (None for foreign/table-less gzip):
+    gates the device-inflate read path."""
+    from ..io import paths as P
+
+    if not getattr(_native, "HAS_GPU_KERNELS", False):
+        return None
+    return P.parse_gz_segments_file(path)
+
+
+def _device_inflate_group(data: torch.Tensor, gz_items, device) -> bool:
+    """Inflate several gzip files' full-flush segments in ONE kernel launch
+    (one segment per lane, csrc/hip/inflate.hip), each file's output landing
+    at its slice of `data`. gz_items = [(path, parse_gz_segments_file meta,
+    out_base)]. Compressed bodies DMA straight from the page cache (pinned
+    mmaps). Returns False when the kernel reported any malformed segment —
+    the caller redoes those files on host zlib. The gzip CRC32 trailer is
+    NOT checked here: the TFRecord layer CRC32C-verifies every record of the
+    inflated bytes (a corrupt stream also breaks the frame chain)."""
     import os as _os
 
+    comp_sizes = [_os.path.getsize(p) for p, _, _ in gz_items]
+    comp_total = sum(comp_sizes)
+    comp = torch.empty(max(comp_total, 1), dtype=torch.uint8,
+                       device=device)[:comp_total]
+    main = torch.cuda.current_stream()
+    streams = _dma_streams()
+    in_off = []
+    in_len = []
+    out_off = []
+    out_len = []
+    used = []
+    coff = 0
+    for k, (p, meta, out_base) in enumerate(gz_items):
+        body_off, segs, _crc, _isize = meta
+        ptr, pinned = _native.file_mmap_pinned(p, comp_sizes[k], False)
+        if pinned:
+            st = streams[k % len(streams)]
+            st.wait_stream(main)
+            _native.gpu_memcpy_h2d(comp.data_ptr() + coff, ptr, comp_sizes[k],
+                                   st.cuda_stream)
+            used.append(st)
+        else:
+            _read_file_staged(p, comp[coff:coff + comp_sizes[k]])
+        so = coff + body_off
+        uo = out_base
+        for c, u in segs:
+            in_off.append(so)
+            in_len.append(c)
+            out_off.append(uo)
+            out_len.append(u)
+            so += c
+            uo += u
+        coff += comp_sizes[k]
+    for st in set(used):
+        main.wait_stream(st)
+    meta_np = np.array([in_off, in_len, out_off, out_len], np.int64)
+    meta_dev = torch.as_tensor(meta_np).to(device)
+    err = torch.full((1,), -1, dtype=torch.int64, device=device)
+    _native.gpu_inflate_segments(
+        comp.data_ptr(), meta_dev[0].data_ptr(), meta_dev[1].data_ptr(),
+        meta_dev[2].data_ptr(), meta_dev[3].data_ptr(), len(in_off),
+        data.data_ptr(), err.data_ptr(), _stream())
+    return int(err.item()) == -1
+
+
+def read_gzip_file_to_device(path: str, device="cuda") -> Optional[torch.Tensor]:
+    """Our-format gzip file -> decompressed bytes in HBM via the device
+    inflater. None when the file carries no segment table or the kernel
+    rejected a segment (callers fall back to host zlib)."""
+    meta = gz_device_meta(path)
+    if meta is None:
+        return None
+    total_u = sum(u for _, u in meta[1])
+    data = torch.empty(max(total_u, 1), dtype=torch.uint8,
+                       device=device)[:total_u]
+    if not _device_inflate_group(data, [(path, meta, 0)], device):
+        return None
+    return data
+
+
+def read_files_to_batch(paths, schema: StructType, record_type: str,
+                        verify_crc: bool = True, device="cuda"):
+    """Decode MANY files as ONE GPU pipeline: their (decompressed) images
+    are concatenated in HBM (TFRecord frames are concatenable, so the frame
+    chain spans file boundaries exactly), scanned and decoded once, and the
+    per-file row counts recovered with a searchsorted over frame offsets.
+    Uncompressed files DMA straight from the page cache; gzip files bearing
+    our segment table are inflated by the device inflater (all files'
+    segments in one launch). Returns (device RecordBatch, np.ndarray
+    row_counts per file)."""
+    import os as _os
+
+    from ..io import paths as P
+
     check_native()
-    sizes = [_os.path.getsize(p) for p in paths]
+    metas = []
+    sizes = []
+    for p in paths:
+        if P.codec_from_path(p) == "gzip":
+            meta = P.parse_gz_segments_file(p)
+            if meta is None:
+                raise ValueError(f"gzip file without segment table: {p} "
+                                 "(host path required)")
+            metas.append(meta)
+            sizes.append(sum(u for _, u in meta[1]))
+        else:
+            metas.append(None)
+            sizes.append(_os.path.getsize(p))
     n = sum(sizes)
     if n == 0:
         z = torch.zeros(0, dtype=torch.int64, device=device)
@@ -852,8 +948,12 @@ def read_files_to_batch(paths, schema: StructType, record_type: str,
     np.cumsum(sizes, out=bounds[1:])
     main = torch.cuda.current_stream()
     streams = _dma_streams()
+    gz_items = []
     for i, p in enumerate(paths):
         if not sizes[i]:
+            continue
+        if metas[i] is not None:
+            gz_items.append((p, metas[i], int(bounds[i])))
             continue
         ptr, pinned = _native.file_mmap_pinned(p, sizes[i], False)
         if pinned:
@@ -863,6 +963,15 @@ def read_files_to_batch(paths, schema: StructType, record_type: str,
             main.wait_stream(st)
         else:
             _read_file_staged(p, data[int(bounds[i]):int(bounds[i + 1])])
+    if gz_items and not _device_inflate_group(data, gz_items, device):
+        # rare fallback: a segment the kernel rejected — host zlib fills
+        # the affected files' slices
+        for p, meta, base in gz_items:
+            blob = np.frombuffer(P.decompress_file(p), np.uint8)
+            stage = pinned_buffer("gzfb", blob.size)
+            stage.numpy()[:blob.size] = blob
+            data[base:base + blob.size].copy_(stage[:blob.size],
+                                              non_blocking=True)
     off, lens = scan_frames_device(data)
     batch = decode_device(data, off, lens, schema, record_type, verify_crc)
     frame_start = off - 12

@@ -162,23 +162,34 @@ def parse_gz_segments(raw: bytes):
 
 
 def _gunzip_parallel(raw: bytes) -> Optional[bytes]:
-    """Parallel segmented inflate of a single-member gzip blob written with
-    full-flush sync points. Returns None when the fast path does not apply
-    (multi-member, extra header fields, marker false positive, CRC mismatch)."""
-    if len(raw) < 20 or raw[:3] != b"\x1f\x8b\x08" or raw[3] != 0:
-        return None  # flags would shift the header; let zlib handle it
-    body = raw[10:-8]
-    crc_want = int.from_bytes(raw[-8:-4], "little")
-    isize = int.from_bytes(raw[-4:], "little")
-    cuts = []
-    p = body.find(_GZ_MARK)
-    while p != -1:
-        cuts.append(p + 4)
-        p = body.find(_GZ_MARK, p + 4)
-    if not cuts:
-        return None
-    bounds = [0] + cuts + [len(body)]
-    segs = [body[bounds[i]:bounds[i + 1]] for i in range(len(bounds) - 1)]
+    """Parallel segmented inflate of a single-member gzip blob. With our
+    FEXTRA table the segment extents are exact; a table-less blob written
+    with full-flush sync points falls back to the marker scan. Returns None
+    when neither fast path applies (multi-member, other header fields,
+    marker false positive, CRC mismatch)."""
+    meta = parse_gz_segments(raw)
+    if meta is not None:
+        body_off, seg_lens, crc_want, isize = meta
+        segs = []
+        pos = body_off
+        for c, _u in seg_lens:
+            segs.append(raw[pos:pos + c])
+            pos += c
+    else:
+        if len(raw) < 20 or raw[:3] != b"\x1f\x8b\x08" or raw[3] != 0:
+            return None  # flags would shift the header; let zlib handle it
+        body = raw[10:-8]
+        crc_want = int.from_bytes(raw[-8:-4], "little")
+        isize = int.from_bytes(raw[-4:], "little")
+        cuts = []
+        p = body.find(_GZ_MARK)
+        while p != -1:
+            cuts.append(p + 4)
+            p = body.find(_GZ_MARK, p + 4)
+        if not cuts:
+            return None
+        bounds = [0] + cuts + [len(body)]
+        segs = [body[bounds[i]:bounds[i + 1]] for i in range(len(bounds) - 1)]
 
     def inflate(seg):
         d = zlib.decompressobj(-15)
@@ -194,6 +205,51 @@ def _gunzip_parallel(raw: bytes) -> Optional[bytes]:
     if len(out) % (1 << 32) != isize or (zlib.crc32(out) & 0xFFFFFFFF) != crc_want:
         return None
     return out
+
+
+def parse_gz_segments_file(path: str):
+    """parse_gz_segments reading only the header and trailer of a file:
+    returns (body_off, [(comp_len, uncomp_len), ...], crc32, isize) or None.
+    The GPU reader uses this to size device buffers without touching the
+    compressed body on the host (it DMAs straight from the page cache)."""
+    import struct as _struct
+
+    try:
+        size = os.path.getsize(path)
+        if size < 20:
+            return None
+        with open(path, "rb") as f:
+            head = f.read(12)
+            if head[:3] != b"\x1f\x8b\x08" or not (head[3] & 0x04):
+                return None
+            if head[3] & ~0x04:
+                return None
+            xlen = int.from_bytes(head[10:12], "little")
+            extra = f.read(xlen)
+            f.seek(size - 8)
+            trailer = f.read(8)
+    except OSError:
+        return None
+    body_off = 12 + xlen
+    pos = 0
+    while pos + 4 <= len(extra):
+        si = extra[pos:pos + 2]
+        ln = int.from_bytes(extra[pos + 2:pos + 4], "little")
+        sub = extra[pos + 4:pos + 4 + ln]
+        pos += 4 + ln
+        if si != b"TS" or len(sub) < 4:
+            continue
+        ver, _, nseg = _struct.unpack("<BBH", sub[:4])
+        if ver != 1 or len(sub) < 4 + 8 * nseg:
+            return None
+        segs = [_struct.unpack("<II", sub[4 + 8 * i:12 + 8 * i])
+                for i in range(nseg)]
+        if sum(c_ for c_, _ in segs) != size - body_off - 8:
+            return None
+        crc = int.from_bytes(trailer[:4], "little")
+        isize = int.from_bytes(trailer[4:], "little")
+        return body_off, segs, crc, isize
+    return None
 
 
 def decompress_file(path: str) -> bytes:

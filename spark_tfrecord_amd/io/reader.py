@@ -45,14 +45,31 @@ def infer_schema_of_paths(files: List[str], record_type: str,
     if record_type == "ByteArray":
         return byte_array_schema()
     for f in files:
-        if engine == "gpu" and P.codec_from_path(f) is None:
+        if engine == "gpu" and P.codec_from_path(f) in (None, "gzip"):
             import os as _os
 
             if _os.path.getsize(f) == 0:
                 continue
             from ..engine import gpu as gpu_engine
 
-            data = gpu_engine.read_file_to_device(f)
+            if P.codec_from_path(f) == "gzip":
+                data = gpu_engine.read_gzip_file_to_device(f)
+                if data is None:  # foreign gzip: host inflate + lattice
+                    data_np = _load_file(f)
+                    if data_np.size == 0:
+                        continue
+                    off_h, lens_h = _native.scan_frames(data_np, False)
+                    if len(off_h) == 0:
+                        continue
+                    codes = infer_codes_from_buffer(data_np, off_h, lens_h,
+                                                    record_type)
+                    if codes:
+                        return schema_from_codes(codes)
+                    return StructType([])
+                if data.numel() == 0:
+                    continue
+            else:
+                data = gpu_engine.read_file_to_device(f)
             off, lens = gpu_engine.scan_frames_device(data)
             if off.numel() == 0:
                 continue
@@ -167,7 +184,17 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     from concurrent.futures import ThreadPoolExecutor
 
     def _needs_host_bytes(fpath: str) -> bool:
-        return eng != "gpu" or P.codec_from_path(fpath) is not None
+        if eng != "gpu":
+            return True
+        codec = P.codec_from_path(fpath)
+        if codec is None:
+            return False
+        if codec == "gzip":
+            # our gzip files carry a segment table: the device inflater
+            # decompresses them in HBM (foreign gzip stays on host zlib)
+            from ..engine import gpu as gpu_engine
+            return gpu_engine.gz_device_meta(fpath) is None
+        return True
 
     workers = min(32, (os.cpu_count() or 8))
     pool = ThreadPoolExecutor(max_workers=workers)

@@ -966,12 +966,19 @@ def read_files_to_batch(paths, schema: StructType, record_type: str,
     if gz_items and not _device_inflate_group(data, gz_items, device):
         # rare fallback: a segment the kernel rejected — host zlib fills
         # the affected files' slices
+        import logging
+
+        logging.getLogger(__name__).warning(
+            "device inflate rejected a segment; host zlib fallback for "
+            "%d gzip file(s)", len(gz_items))
         for p, meta, base in gz_items:
             blob = np.frombuffer(P.decompress_file(p), np.uint8)
             stage = pinned_buffer("gzfb", blob.size)
             stage.numpy()[:blob.size] = blob
             data[base:base + blob.size].copy_(stage[:blob.size],
                                               non_blocking=True)
+            # the staging buffer is reused next iteration: wait out the copy
+            torch.cuda.current_stream().synchronize()
     off, lens = scan_frames_device(data)
     batch = decode_device(data, off, lens, schema, record_type, verify_crc)
     frame_start = off - 12

@@ -86,7 +86,20 @@ struct BitRd {
   int n;  // bits buffered; < 0 => underflow (sticky error)
 };
 
+// Word-wise refill: ONE unaligned u64 load tops the buffer up to >=57 bits
+// (byte-at-a-time refills cost a dependent global load per input byte — at
+// the inflater's intrinsically low occupancy every one of those stalls the
+// wave for the full L2 latency; measured 15x slower).
 __device__ inline void br_refill(BitRd& b) {
+  if (b.end - b.p >= 8) {
+    u64 w;
+    __builtin_memcpy(&w, b.p, 8);
+    int take = (64 - b.n) >> 3;
+    b.buf |= w << b.n;
+    b.p += take;
+    b.n += take << 3;
+    return;
+  }
   while (b.n <= 56 && b.p < b.end) {
     b.buf |= (u64)(*b.p++) << b.n;
     b.n += 8;
@@ -160,14 +173,24 @@ __device__ inline bool build_huff16(const uint16_t* lens, int nsym,
   return true;
 }
 
+// Peek-based canonical decode: bit-reverse the next 15 buffered bits once,
+// then every candidate length is a shift+compare — no per-bit buffer ops.
 __device__ inline int huff_decode(BitRd& br, const uint16_t* base,
                                   const uint16_t* rank, const uint16_t* sym) {
-  u32 code = 0;
+  if (br.n < 15) br_refill(br);  // short tail: zero bits pad the peek
+  u32 rev = __brev((u32)br.buf) >> 17;
   for (int l = 1; l <= 15; ++l) {
-    code = (code << 1) | br_bits(br, 1);
+    u32 code = rev >> (15 - l);
     u32 idx = code - base[l];
-    if (idx < (u32)(rank[l + 1] - rank[l])) return sym[rank[l] + idx];
-    if (br.n < 0) return -1;
+    if (idx < (u32)(rank[l + 1] - rank[l])) {
+      if (br.n < l) {
+        br.n = -1 << 20;  // consumed past the stream end
+        return -1;
+      }
+      br.buf >>= l;
+      br.n -= l;
+      return sym[rank[l] + idx];
+    }
   }
   return -1;
 }
@@ -194,7 +217,14 @@ __device__ inline int inflate_one(const u8* in, i64 ilen, u8* dst, i64 expect,
       if (br.n < 0 || ((len ^ nlen) & 0xFFFFu) != 0xFFFFu) return 2;
       const u8* src = br.p - (br.n >> 3);  // rewind buffered bytes
       if (src + len > br.end || opos + (i64)len > expect) return 3;
-      for (u32 i = 0; i < len; ++i) dst[opos + i] = src[i];
+      u8* dp = dst + opos;
+      u32 i = 0;
+      for (; i + 8 <= len; i += 8) {
+        u64 w;
+        __builtin_memcpy(&w, src + i, 8);
+        __builtin_memcpy(dp + i, &w, 8);
+      }
+      for (; i < len; ++i) dp[i] = src[i];
       opos += len;
       br.p = src + len;
       br.buf = 0;

@@ -174,12 +174,12 @@ def wire_to_arrow(col: WireColumn, dt: DataType, nullable: bool, name: str,
         raise RuntimeError(f"Feature '{name}' dimensionality does not match schema")
     presence = np.asarray(col.presence)
     mask = presence.astype(bool)
-    if not nullable and not mask.all():
+    all_present = bool(mask.all())
+    if not nullable and not all_present:
         r = int(np.argmin(mask))
         raise ValueError(f"Feature '{name}' is required but missing (row {r})")
     row_off = np.asarray(col.row_off).astype(np.int64, copy=False)
-    row_len = row_off[1:] - row_off[:-1]
-    valid_buf = _validity(presence)
+    valid_buf = None if all_present else np.packbits(mask, bitorder="little")
 
     if seq:
         # 2-D ragged: rows -> lists (sub-lists) -> values
@@ -210,18 +210,20 @@ def wire_to_arrow(col: WireColumn, dt: DataType, nullable: bool, name: str,
                                      children=[inner_vals])
 
     # Scalar: head element per present row (reference head semantics).
-    if bool((mask & (row_len == 0)).any()):
-        r = int(np.argmax(mask & (row_len == 0)))
-        raise ValueError(
-            f"Feature '{name}' is present but empty; cannot read scalar (row {r})")
+    row_len = row_off[1:] - row_off[:-1]
     # identity fast path — every row present with exactly one value (the
     # overwhelmingly common scalar-column case): values ARE the heads, no
     # boolean gather (r01 finding: these 1M-element gathers were ~half the
     # engine->API read gap)
-    if valid_buf is None and int(row_off[-1]) == num_rows:
+    if all_present and int(row_off[-1]) == num_rows and \
+            not bool((row_len != 1).any()):
         if col.kind == KIND_BYTES:
-            return _bytes_arrow(col, dt, row_off[:-1])
+            return _bytes_arrow(col, dt, None)
         return pa.array(_numeric_np(col, dt))
+    if bool((mask & (row_len == 0)).any()):
+        r = int(np.argmax(mask & (row_len == 0)))
+        raise ValueError(
+            f"Feature '{name}' is present but empty; cannot read scalar (row {r})")
     heads = row_off[:-1][mask]
     if col.kind == KIND_BYTES:
         present_arr = _bytes_arrow(col, dt, heads)

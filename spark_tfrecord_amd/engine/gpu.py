@@ -265,6 +265,19 @@ def device_to_bytes(img: torch.Tensor) -> bytes:
     return buf.numpy()[:n].tobytes()
 
 
+def device_to_pinned_view(img: torch.Tensor, tag: str = "d2h_img") -> np.ndarray:
+    """One link-speed D2H of a device image into the reusable pinned buffer;
+    returns a numpy VIEW of it (valid until the tag's next use). Partitioned
+    writes slice this view per partition file — registering a fresh mmap per
+    part file costs ~0.16 ms/MB in hipHostRegister alone, ~10x the bytes'
+    DMA time for many small fresh files (r01 config-3 cliff)."""
+    n = img.numel()
+    buf = pinned_buffer(tag, n)
+    buf[:n].copy_(img, non_blocking=True)
+    torch.cuda.synchronize()
+    return buf.numpy()[:n]
+
+
 # ---------------------------------------------------------------------------
 # GPU frame scan: parallel frame-boundary discovery + chain validation.
 # ---------------------------------------------------------------------------

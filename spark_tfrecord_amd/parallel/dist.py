@@ -306,28 +306,51 @@ def write_tfrecord_distributed(data, path: str, record_type: str = "Example",
             for combo, ln in hdr:
                 mine.setdefault(combo, []).append(blob[pos:pos + ln])
                 pos += ln
+        # device blobs: concatenate into ONE image, one pinned D2H, then
+        # write the per-partition files from view slices on a thread pool
+        # (per-file mmap registration on fresh inodes costs more than the
+        # bytes' DMA — see io/writer.py partition path)
+        dev_jobs: List[Tuple[tuple, int, int]] = []
+        host_jobs: List[Tuple[tuple, bytes]] = []
+        dev_parts: List[torch.Tensor] = []
+        pos = 0
         for combo, blobs in sorted(mine.items(), key=lambda kv: str(kv[0])):
+            if blobs and isinstance(blobs[0], torch.Tensor):
+                raw_t = (torch.cat(blobs) if len(blobs) > 1
+                         else blobs[0]).contiguous()
+                dev_parts.append(raw_t)
+                dev_jobs.append((combo, pos, pos + raw_t.numel()))
+                pos += raw_t.numel()
+            else:
+                host_jobs.append((combo, b"".join(blobs)))
+        view = None
+        if dev_parts:
+            from ..engine import gpu as gpu_engine
+
+            img = torch.cat(dev_parts) if len(dev_parts) > 1 else dev_parts[0]
+            view = gpu_engine.device_to_pinned_view(img, tag="dist_d2h")
+
+        def _write_combo(combo, payload):
             sub_dir = os.path.join(
                 path, *(f"{c}={_partition_dir_value(v)}"
                         for c, v in zip(partition_by, combo)))
             os.makedirs(sub_dir, exist_ok=True)
-            fname = P.part_file_name(rank, codec, job_id)
-            fpath = os.path.join(sub_dir, fname)
-            if blobs and isinstance(blobs[0], torch.Tensor):
-                from ..engine import gpu as gpu_engine
+            fpath = os.path.join(sub_dir, P.part_file_name(rank, codec, job_id))
+            P.write_file_atomic(
+                payload if codec is None else P.compress_bytes(
+                    payload if isinstance(payload, bytes)
+                    else payload.tobytes(), codec), fpath)
 
-                raw_t = torch.cat(blobs) if len(blobs) > 1 else blobs[0]
-                if codec is None:
-                    tmp = P.hidden_tmp_path(fpath)
-                    gpu_engine.device_to_file(raw_t.contiguous(), tmp)
-                    os.replace(tmp, fpath)
-                else:
-                    P.write_file_atomic(P.compress_bytes(
-                        gpu_engine.device_to_bytes(raw_t.contiguous()), codec),
-                        fpath)
-            else:
-                raw = b"".join(blobs)  # TFRecord frames concatenate losslessly
-                P.write_file_atomic(P.compress_bytes(raw, codec), fpath)
+        from concurrent.futures import ThreadPoolExecutor
+
+        njobs = len(dev_jobs) + len(host_jobs)
+        if njobs:
+            with ThreadPoolExecutor(max_workers=min(16, njobs)) as ex:
+                futs = [ex.submit(_write_combo, c, view[lo:hi])
+                        for c, lo, hi in dev_jobs]
+                futs += [ex.submit(_write_combo, c, b) for c, b in host_jobs]
+                for f in futs:
+                    f.result()
 
     dist.barrier()
     if rank == 0:

@@ -59,13 +59,16 @@ __device__ __constant__ uint8_t kClOrder[19] = {16, 17, 18, 0, 8,  7, 9,
                                                 6,  10, 5,  11, 4, 12, 3,
                                                 13, 2,  14, 1,  15};
 
-// Per-lane LDS scratch: 936 B x 64 lanes = 58.5 KiB static shared.
+// Per-lane LDS scratch: ~1 KB x 64 lanes < 64 KiB static shared. The decode
+// loop's per-length lookup is ONE u32 load — (count << 16) | base packed —
+// with the rank read only on the hit (the loop is the latency chain at the
+// inflater's low occupancy; separate base/count/rank reads tripled it).
 struct LaneScratch {
+  u32 bc_lit[16];          // (cnt << 16) | first canonical code, per length
+  u32 bc_dist[16];
   uint16_t sym[320];       // lit/len symbols [0,288) + dist symbols [288,320)
                            // (tail doubles as CL-table space during header)
-  uint16_t base_lit[16];   // first canonical code of each bit length
-  uint16_t rank_lit[17];   // first symbol-table rank; cnt[l]=rank[l+1]-rank[l]
-  uint16_t base_dist[16];
+  uint16_t rank_lit[17];   // first symbol-table rank per length
   uint16_t rank_dist[17];
   u8 lens4[160];           // 320 nibble-packed code lengths
 };
@@ -121,10 +124,9 @@ __device__ inline u32 br_bits(BitRd& b, int k) {
 }
 
 // Canonical-Huffman build from nibble-packed lengths [len_off, len_off+nsym).
-// Emits (base, rank, sym); rejects over-subscribed codes.
+// Emits (bc = (cnt<<16)|base, rank, sym); rejects over-subscribed codes.
 __device__ inline bool build_huff4(const u8* lens4, int len_off, int nsym,
-                                   uint16_t* base, uint16_t* rank,
-                                   uint16_t* sym) {
+                                   u32* bc, uint16_t* rank, uint16_t* sym) {
   for (int l = 0; l < 17; ++l) rank[l] = 0;
   for (int s = 0; s < nsym; ++s) {
     u32 L = get_len4(lens4, len_off + s);
@@ -133,7 +135,7 @@ __device__ inline bool build_huff4(const u8* lens4, int len_off, int nsym,
   u32 code = 0, k = 0;
   for (int l = 1; l <= 15; ++l) {
     u32 cnt = rank[l];
-    base[l] = (uint16_t)code;
+    bc[l] = (cnt << 16) | code;
     rank[l] = (uint16_t)k;
     k += cnt;
     code = (code + cnt) << 1;
@@ -151,15 +153,14 @@ __device__ inline bool build_huff4(const u8* lens4, int len_off, int nsym,
 
 // Same build from a u16 length array (the 19-symbol code-length alphabet).
 __device__ inline bool build_huff16(const uint16_t* lens, int nsym,
-                                    uint16_t* base, uint16_t* rank,
-                                    uint16_t* sym) {
+                                    u32* bc, uint16_t* rank, uint16_t* sym) {
   for (int l = 0; l < 17; ++l) rank[l] = 0;
   for (int s = 0; s < nsym; ++s)
     if (lens[s]) ++rank[lens[s]];
   u32 code = 0, k = 0;
   for (int l = 1; l <= 15; ++l) {
     u32 cnt = rank[l];
-    base[l] = (uint16_t)code;
+    bc[l] = (cnt << 16) | code;
     rank[l] = (uint16_t)k;
     k += cnt;
     code = (code + cnt) << 1;
@@ -175,14 +176,14 @@ __device__ inline bool build_huff16(const uint16_t* lens, int nsym,
 
 // Peek-based canonical decode: bit-reverse the next 15 buffered bits once,
 // then every candidate length is a shift+compare — no per-bit buffer ops.
-__device__ inline int huff_decode(BitRd& br, const uint16_t* base,
+__device__ inline int huff_decode(BitRd& br, const u32* bc,
                                   const uint16_t* rank, const uint16_t* sym) {
   if (br.n < 15) br_refill(br);  // short tail: zero bits pad the peek
   u32 rev = __brev((u32)br.buf) >> 17;
   for (int l = 1; l <= 15; ++l) {
-    u32 code = rev >> (15 - l);
-    u32 idx = code - base[l];
-    if (idx < (u32)(rank[l + 1] - rank[l])) {
+    u32 w = bc[l];
+    u32 idx = (rev >> (15 - l)) - (w & 0xFFFFu);
+    if (idx < (w >> 16)) {
       if (br.n < l) {
         br.n = -1 << 20;  // consumed past the stream end
         return -1;
@@ -254,14 +255,14 @@ __device__ inline int inflate_one(const u8* in, i64 ilen, u8* dst, i64 expect,
       uint16_t* cl_sym = &L.sym[280];   // <= 19 entries
       for (int i = 0; i < 19; ++i) cl_lens[i] = 0;
       for (int i = 0; i < hclen; ++i) cl_lens[kClOrder[i]] = br_bits(br, 3);
-      if (br.n < 0 || !build_huff16(cl_lens, 19, L.base_dist, L.rank_dist,
+      if (br.n < 0 || !build_huff16(cl_lens, 19, L.bc_dist, L.rank_dist,
                                     cl_sym))
         return 6;
       int total = hlit + hdist;
       int n = 0;
       u32 prev = 0;
       while (n < total) {
-        int s = huff_decode(br, L.base_dist, L.rank_dist, cl_sym);
+        int s = huff_decode(br, L.bc_dist, L.rank_dist, cl_sym);
         if (s < 0) return 7;
         if (s < 16) {
           set_len4(L.lens4, n++, (u32)s);
@@ -279,13 +280,13 @@ __device__ inline int inflate_one(const u8* in, i64 ilen, u8* dst, i64 expect,
         if (br.n < 0) return 8;
       }
     }
-    if (!build_huff4(L.lens4, 0, hlit, L.base_lit, L.rank_lit, L.sym))
+    if (!build_huff4(L.lens4, 0, hlit, L.bc_lit, L.rank_lit, L.sym))
       return 9;
-    if (!build_huff4(L.lens4, dist_off, hdist, L.base_dist, L.rank_dist,
+    if (!build_huff4(L.lens4, dist_off, hdist, L.bc_dist, L.rank_dist,
                      &L.sym[288]))
       return 10;
     for (;;) {
-      int s = huff_decode(br, L.base_lit, L.rank_lit, L.sym);
+      int s = huff_decode(br, L.bc_lit, L.rank_lit, L.sym);
       if (s < 0) return 11;
       if (s < 256) {
         if (opos >= expect) return 12;
@@ -296,7 +297,7 @@ __device__ inline int inflate_one(const u8* in, i64 ilen, u8* dst, i64 expect,
         s -= 257;
         if (s >= 29) return 13;
         i64 mlen = kLenBase[s] + (i64)br_bits(br, kLenExtra[s]);
-        int d = huff_decode(br, L.base_dist, L.rank_dist, &L.sym[288]);
+        int d = huff_decode(br, L.bc_dist, L.rank_dist, &L.sym[288]);
         if (d < 0 || d >= 30) return 14;
         i64 dist = kDistBase[d] + (i64)br_bits(br, kDistExtra[d]);
         if (br.n < 0) return 15;

@@ -993,11 +993,31 @@ def read_files_to_batch(paths, schema: StructType, record_type: str,
             # the staging buffer is reused next iteration: wait out the copy
             torch.cuda.current_stream().synchronize()
     off, lens = scan_frames_device(data)
-    batch = decode_device(data, off, lens, schema, record_type, verify_crc)
     frame_start = off - 12
     counts = torch.searchsorted(
         frame_start, torch.as_tensor(bounds, device=device)).cpu().numpy()
-    return batch, np.diff(counts)
+    row_counts = np.diff(counts)
+    if schema is None:
+        # FUSED schema inference: the reference scans the first non-empty
+        # file in a separate job (DefaultSource.scala:31-39); here the file
+        # image is already in HBM, so the lattice kernel runs over that
+        # file's frames and the decode reuses the same image — a
+        # schema-less read costs ONE pass instead of two.
+        from ..infer import byte_array_schema, schema_from_codes
+
+        if record_type == "ByteArray":
+            schema = byte_array_schema()
+        else:
+            i = next((k for k in range(len(paths)) if row_counts[k] > 0), None)
+            if i is None:
+                raise ValueError(
+                    "Could not infer schema: no non-empty TFRecord files found")
+            r0, r1 = int(counts[i]), int(counts[i + 1])
+            codes = infer_codes_device(data, off[r0:r1].contiguous(),
+                                       lens[r0:r1].contiguous(), record_type)
+            schema = schema_from_codes(codes)
+    batch = decode_device(data, off, lens, schema, record_type, verify_crc)
+    return batch, row_counts
 
 
 def encode_partitions_device(batch: RecordBatch, part_codes: np.ndarray,

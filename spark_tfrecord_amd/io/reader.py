@@ -160,21 +160,6 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     eng = engine_mod.resolve_engine(engine)
 
     metrics = IOMetrics("read")
-    if schema is None:
-        with StageTimer(metrics, "infer_schema"):
-            schema = (byte_array_schema() if record_type == "ByteArray"
-                      else infer_schema_of_paths(files, record_type, eng))
-    if columns is not None:
-        missing = [c for c in columns if c not in [f.name for f in schema.fields]
-                   and c not in part_cols]
-        if missing:
-            raise KeyError(f"columns not in schema: {missing}")
-        schema = StructType([f for f in schema.fields if f.name in set(columns)])
-        part_cols = [c for c in part_cols if c in set(columns)]
-    data_schema = StructType([f for f in schema.fields if f.name not in part_cols])
-    # an Example cannot carry 2-D ragged fields: reject like the reference's
-    # deserializer construction would (TFRecordDeserializer.scala:148-175)
-    validate_schema_for_record_type(data_schema, record_type)
 
     # Files needing host bytes (compressed, or CPU engine) are loaded and
     # inflated by a thread pool — gzip/zlib release the GIL, so multi-file
@@ -195,6 +180,37 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
             from ..engine import gpu as gpu_engine
             return gpu_engine.gz_device_meta(fpath) is None
         return True
+
+    # Schema-less GPU reads DEFER inference into the first group pipeline:
+    # the file image is scanned once in HBM, the lattice kernel runs over
+    # the first non-empty file's frames, and the decode reuses the image —
+    # one pass instead of the reference's extra inference job.
+    deferred_infer = (schema is None and eng == "gpu" and bool(files)
+                      and not _needs_host_bytes(files[0]))
+    if schema is None and not deferred_infer:
+        with StageTimer(metrics, "infer_schema"):
+            schema = (byte_array_schema() if record_type == "ByteArray"
+                      else infer_schema_of_paths(files, record_type, eng))
+
+    def _resolve(schema):
+        nonlocal part_cols
+        if columns is not None:
+            missing = [c for c in columns
+                       if c not in [f.name for f in schema.fields]
+                       and c not in part_cols]
+            if missing:
+                raise KeyError(f"columns not in schema: {missing}")
+            schema = StructType(
+                [f for f in schema.fields if f.name in set(columns)])
+            part_cols = [c for c in part_cols if c in set(columns)]
+        data_schema = StructType(
+            [f for f in schema.fields if f.name not in part_cols])
+        # an Example cannot carry 2-D ragged fields: reject like the
+        # reference's deserializer would (TFRecordDeserializer.scala:148-175)
+        validate_schema_for_record_type(data_schema, record_type)
+        return data_schema
+
+    data_schema = None if deferred_infer else _resolve(schema)
 
     workers = min(32, (os.cpu_count() or 8))
     pool = ThreadPoolExecutor(max_workers=workers)
@@ -244,11 +260,30 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
                     gbytes += sz
                     metrics.add(files=1, nbytes=sz)
                     i += 1
-                batch, row_counts = gpu_engine.read_files_to_batch(
-                    group, data_schema, record_type, verify_crc=verify_crc)
+                if data_schema is None:
+                    try:
+                        batch, row_counts = gpu_engine.read_files_to_batch(
+                            group, None, record_type, verify_crc=verify_crc)
+                        data_schema = _resolve(batch.schema)
+                    except ValueError:
+                        # this group was entirely empty: resolve the schema
+                        # from the full file list, then decode normally
+                        data_schema = _resolve(
+                            byte_array_schema() if record_type == "ByteArray"
+                            else infer_schema_of_paths(files, record_type, eng))
+                        batch, row_counts = gpu_engine.read_files_to_batch(
+                            group, data_schema, record_type,
+                            verify_crc=verify_crc)
+                else:
+                    batch, row_counts = gpu_engine.read_files_to_batch(
+                        group, data_schema, record_type, verify_crc=verify_crc)
                 if batch.num_rows == 0 and not part_cols:
                     continue
                 t = batch_to_table(gpu_engine.batch_to_host(batch))
+                if [f.name for f in data_schema.fields] != t.column_names:
+                    # deferred inference + column projection: decode carried
+                    # the full inferred schema; project here
+                    t = t.select([f.name for f in data_schema.fields])
                 _append_with_parts(t, group, row_counts)
                 continue
             for j in range(i, min(i + window, len(files))):

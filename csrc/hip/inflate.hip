@@ -175,25 +175,47 @@ __device__ inline bool build_huff16(const uint16_t* lens, int nsym,
 }
 
 // Peek-based canonical decode: bit-reverse the next 15 buffered bits once,
-// then every candidate length is a shift+compare — no per-bit buffer ops.
+// then every candidate length is a shift+compare. The (cnt|base) words for
+// lengths 1..8 are INDEPENDENT of the bitstream, so they are batch-loaded
+// up front — one LDS wait covers the whole common case instead of a
+// dependent load per candidate length (the inflater runs at 1-2 waves per
+// CU, so every serialized memory wait is raw wall time; PMC showed
+// VALU/busy = 0.04 with the load-per-iteration form).
 __device__ inline int huff_decode(BitRd& br, const u32* bc,
                                   const uint16_t* rank, const uint16_t* sym) {
   if (br.n < 15) br_refill(br);  // short tail: zero bits pad the peek
   u32 rev = __brev((u32)br.buf) >> 17;
-  for (int l = 1; l <= 15; ++l) {
-    u32 w = bc[l];
-    u32 idx = (rev >> (15 - l)) - (w & 0xFFFFu);
-    if (idx < (w >> 16)) {
-      if (br.n < l) {
-        br.n = -1 << 20;  // consumed past the stream end
-        return -1;
+  u32 b1 = bc[1], b2 = bc[2], b3 = bc[3], b4 = bc[4];
+  u32 b5 = bc[5], b6 = bc[6], b7 = bc[7], b8 = bc[8];
+  u32 idx;
+  int l;
+  if ((idx = (rev >> 14) - (b1 & 0xFFFFu)) < (b1 >> 16)) l = 1;
+  else if ((idx = (rev >> 13) - (b2 & 0xFFFFu)) < (b2 >> 16)) l = 2;
+  else if ((idx = (rev >> 12) - (b3 & 0xFFFFu)) < (b3 >> 16)) l = 3;
+  else if ((idx = (rev >> 11) - (b4 & 0xFFFFu)) < (b4 >> 16)) l = 4;
+  else if ((idx = (rev >> 10) - (b5 & 0xFFFFu)) < (b5 >> 16)) l = 5;
+  else if ((idx = (rev >> 9) - (b6 & 0xFFFFu)) < (b6 >> 16)) l = 6;
+  else if ((idx = (rev >> 8) - (b7 & 0xFFFFu)) < (b7 >> 16)) l = 7;
+  else if ((idx = (rev >> 7) - (b8 & 0xFFFFu)) < (b8 >> 16)) l = 8;
+  else {
+    l = 16;
+    for (int k = 9; k <= 15; ++k) {
+      u32 w = bc[k];
+      idx = (rev >> (15 - k)) - (w & 0xFFFFu);
+      if (idx < (w >> 16)) {
+        l = k;
+        break;
       }
-      br.buf >>= l;
-      br.n -= l;
-      return sym[rank[l] + idx];
     }
+    if (l > 15) return -1;
   }
-  return -1;
+  if (br.n < l) {
+    br.n = -1 << 20;  // consumed past the stream end
+    return -1;
+  }
+  br.buf >>= l;
+  br.n -= l;
+  return sym[rank[l] + idx];
 }
 
 // Inflate one raw-deflate segment into dst[0, expect). Returns 0 on

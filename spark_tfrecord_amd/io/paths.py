@@ -78,11 +78,12 @@ def codec_from_path(path: str) -> Optional[str]:
 # > _GZ_MAX_SEGS segments) falls back to the marker scan, then to
 # sequential inflate. Matches the reference's isSplitable=false model:
 # gzip files still read as whole files, just on more than one core/CU.
-# 64 KiB segments: device inflation is one segment per LANE and Huffman
+# 32 KiB segments: device inflation is one segment per LANE and Huffman
 # decode is bit-serial, so wall time ~= segment size — smaller segments buy
-# parallelism directly (a 256 MB file yields ~4000 lanes). Compression-ratio
-# cost of the extra dictionary resets measured ~2-3% at level 6.
-_GZ_SEGMENT = int(os.environ.get("TFREC_GZ_SEGMENT", 64 << 10))
+# parallelism AND a lower single-file latency floor directly. Compression-
+# ratio cost of the extra dictionary resets is a few % at level 6
+# (TFREC_GZ_SEGMENT overrides for ratio-sensitive datasets).
+_GZ_SEGMENT = int(os.environ.get("TFREC_GZ_SEGMENT", 32 << 10))
 _GZ_MARK = b"\x00\x00\xff\xff"
 _GZ_MAX_SEGS = 8189  # FEXTRA payload cap: 65535 bytes / 8 per segment
 

@@ -220,8 +220,12 @@ __device__ inline int huff_decode(BitRd& br, const u32* bc,
 
 // Inflate one raw-deflate segment into dst[0, expect). Returns 0 on
 // success, a small nonzero cause code otherwise (any nonzero => the Python
-// side redoes the FILE on the host zlib path).
-__device__ inline int inflate_one(const u8* in, i64 ilen, u8* dst, i64 expect,
+// side redoes the FILE on the host zlib path). `in`/`dst` are __restrict__:
+// without it the compiler must order every bit-refill load after all
+// pending output stores (they could alias), serializing the decode on L2
+// store latency.
+__device__ inline int inflate_one(const u8* __restrict__ in, i64 ilen,
+                                  u8* __restrict__ dst, i64 expect,
                                   LaneScratch& L) {
   BitRd br{in, in + ilen, 0, 0};
   i64 opos = 0;
@@ -307,15 +311,36 @@ __device__ inline int inflate_one(const u8* in, i64 ilen, u8* dst, i64 expect,
     if (!build_huff4(L.lens4, dist_off, hdist, L.bc_dist, L.rank_dist,
                      &L.sym[288]))
       return 10;
+    // literal accumulation window: byte-per-literal global stores made the
+    // literal-heavy path store-bound; 8 literals flush as one u64 store
+    // (flushed before matches, which may read the freshly-written bytes)
+    u64 lw = 0;
+    int ln = 0;
+    auto flush_lits = [&]() {
+      if (!ln) return;
+      if (opos + 8 <= expect) {
+        // bytes past ln are garbage but lie before future output: they are
+        // overwritten by construction (opos+8 <= expect)
+        __builtin_memcpy(dst + opos, &lw, 8);
+      } else {
+        for (int i = 0; i < ln; ++i) dst[opos + i] = (u8)(lw >> (8 * i));
+      }
+      opos += ln;
+      lw = 0;
+      ln = 0;
+    };
     for (;;) {
       int s = huff_decode(br, L.bc_lit, L.rank_lit, L.sym);
       if (s < 0) return 11;
       if (s < 256) {
-        if (opos >= expect) return 12;
-        dst[opos++] = (u8)s;
+        if (opos + ln >= expect) return 12;
+        lw |= (u64)(u8)s << (8 * ln);
+        if (++ln == 8) flush_lits();
       } else if (s == 256) {
+        flush_lits();
         break;
       } else {
+        flush_lits();
         s -= 257;
         if (s >= 29) return 13;
         i64 mlen = kLenBase[s] + (i64)br_bits(br, kLenExtra[s]);

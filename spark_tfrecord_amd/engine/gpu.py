@@ -1170,11 +1170,14 @@ def batch_to_device(batch: RecordBatch, device="cuda") -> RecordBatch:
 def batch_to_host(batch: RecordBatch) -> RecordBatch:
     """Device batch -> host numpy batch. All D2H copies go into PINNED
     host tensors (torch's caching host allocator reuses them) issued
-    async back-to-back with ONE sync — pageable `.cpu()` per tensor runs at
-    ~8 GB/s with a hidden staging copy; this runs at link speed. The numpy
-    arrays are zero-copy views of the pinned tensors (kept alive by the
-    view's base)."""
-    pending = []
+    async round-robin over the TWO side streams with one sync at the end —
+    pageable `.cpu()` per tensor runs at ~8 GB/s with a hidden staging
+    copy, one pinned stream at ~35, two at link speed. The numpy arrays
+    are zero-copy views of the pinned tensors (kept alive by the view's
+    base)."""
+    streams = _dma_streams()
+    main = torch.cuda.current_stream()
+    state = {"used": False, "k": 0}
 
     def h(t):
         if not isinstance(t, torch.Tensor):
@@ -1182,8 +1185,15 @@ def batch_to_host(batch: RecordBatch) -> RecordBatch:
         if not t.is_cuda:
             return t.numpy()
         host = torch.empty_like(t, device="cpu", pin_memory=True)
-        host.copy_(t, non_blocking=True)
-        pending.append(True)
+        st = streams[state["k"] % len(streams)]
+        state["k"] += 1
+        if not state["used"]:
+            for s in streams:
+                s.wait_stream(main)
+            state["used"] = True
+        with torch.cuda.stream(st):
+            host.copy_(t, non_blocking=True)
+        t.record_stream(st)
         return host.numpy()
 
     cols = [WireColumn(c.kind, c.is_seq, h(c.presence), h(c.row_off), h(c.values),
@@ -1191,8 +1201,9 @@ def batch_to_host(batch: RecordBatch) -> RecordBatch:
                        h(c.list_off) if c.list_off is not None else None,
                        h(c.sub_off) if c.sub_off is not None else None)
             for c in batch.columns]
-    if pending:
-        torch.cuda.current_stream().synchronize()
+    if state["used"]:
+        for s in streams:
+            s.synchronize()
     return RecordBatch(batch.schema, cols, batch.num_rows)
 
 

@@ -700,3 +700,55 @@ class TestDeviceInflate:
             f.write(bytes(raw))
         with pytest.raises(Exception):
             stf.read_tfrecord(out, engine="gpu").collect()
+
+
+class TestDeferredInference:
+    """Schema-less GPU reads fuse inference into the group pipeline; the
+    semantics must stay exactly the reference's (first non-empty file
+    decides; columns project; partition dirs still discovered)."""
+
+    def test_schemaless_read_matches_explicit(self, tmp_sandbox):
+        out = str(tmp_sandbox / "d")
+        data = {"a": np.arange(1000, dtype=np.int64),
+                "b": [[float(i), i / 3] for i in range(1000)],
+                "s": [f"v{i}" for i in range(1000)]}
+        stf.write_tfrecord(data, out, num_shards=3)
+        got = stf.read_tfrecord(out, engine="gpu").sort("a").collect()
+        assert len(got) == 1000
+        assert got[5]["s"] == "v5"
+        assert got[7]["b"] == [pytest.approx(7.0), pytest.approx(7 / 3)]
+
+    def test_schemaless_with_columns_projection(self, tmp_sandbox):
+        out = str(tmp_sandbox / "p")
+        stf.write_tfrecord({"a": np.arange(50, dtype=np.int64),
+                            "b": np.arange(50, dtype=np.int64) * 2}, out)
+        df = stf.read_tfrecord(out, engine="gpu", columns=["b"])
+        assert df.columns == ["b"]
+        assert sorted(r["b"] for r in df.collect()) == [2 * i for i in range(50)]
+        with pytest.raises(KeyError):
+            stf.read_tfrecord(out, engine="gpu", columns=["nope"])
+
+    def test_schemaless_partitioned(self, tmp_sandbox):
+        out = str(tmp_sandbox / "part")
+        stf.write_tfrecord({"p": ["x", "y"] * 10,
+                            "v": np.arange(20, dtype=np.int64)}, out,
+                           partition_by=["p"])
+        rows = stf.read_tfrecord(out, engine="gpu").sort("v").collect()
+        assert len(rows) == 20
+        assert rows[0]["p"] == "x" and rows[1]["p"] == "y"
+
+    def test_first_file_empty_fallback(self, tmp_sandbox):
+        out = str(tmp_sandbox / "empty_first")
+        stf.write_tfrecord({"x": np.arange(5, dtype=np.int64)}, out)
+        # an empty part file that sorts FIRST: the fused path must fall
+        # through to the non-empty file for the schema
+        open(os.path.join(out, "part-00000-aaa.tfrecord"), "wb").close()
+        rows = stf.read_tfrecord(out, engine="gpu").collect()
+        assert sorted(r["x"] for r in rows) == list(range(5))
+
+    def test_all_files_empty_raises(self, tmp_sandbox):
+        out = str(tmp_sandbox / "all_empty")
+        os.makedirs(out)
+        open(os.path.join(out, "part-00000-aaa.tfrecord"), "wb").close()
+        with pytest.raises(ValueError, match="no non-empty"):
+            stf.read_tfrecord(out, engine="gpu")

@@ -952,6 +952,9 @@ def read_files_to_batch(paths, schema: StructType, record_type: str,
             sizes.append(_os.path.getsize(p))
     n = sum(sizes)
     if n == 0:
+        if schema is None:
+            raise ValueError(
+                "Could not infer schema: no non-empty TFRecord files found")
         z = torch.zeros(0, dtype=torch.int64, device=device)
         return (decode_device(torch.zeros(0, dtype=torch.uint8, device=device),
                               z, z.clone(), schema, record_type, verify_crc),

@@ -752,3 +752,31 @@ class TestDeferredInference:
         open(os.path.join(out, "part-00000-aaa.tfrecord"), "wb").close()
         with pytest.raises(ValueError, match="no non-empty"):
             stf.read_tfrecord(out, engine="gpu")
+
+
+class TestMixedDatasetRead:
+    def test_uncompressed_plus_gzip_plus_foreign(self, tmp_sandbox):
+        """One dataset mixing uncompressed parts (device DMA), our gzip
+        (device inflate) and a foreign table-less gzip (host fallback):
+        the reader must merge all three transparently."""
+        import gzip as _gzip
+
+        from spark_tfrecord_amd.io import paths as P
+
+        out = str(tmp_sandbox / "mix")
+        stf.write_tfrecord({"x": np.arange(0, 100, dtype=np.int64)}, out,
+                           num_shards=2)
+        stf.write_tfrecord({"x": np.arange(100, 200, dtype=np.int64)}, out,
+                           codec="gzip", mode="append", shard_offset=2,
+                           write_success=False)
+        # foreign gzip: raw frames compressed by the gzip module (no table)
+        from spark_tfrecord_amd.engine import cpu as cpu_engine
+        from spark_tfrecord_amd.columnar import RecordBatch, column_from_values
+        schema = stf.StructType([stf.StructField("x", stf.LongType(), True)])
+        b = RecordBatch(schema, [column_from_values(
+            np.arange(200, 300, dtype=np.int64), stf.LongType(), True, "x")], 100)
+        raw = cpu_engine.encode_batch(b, "Example")
+        with open(os.path.join(out, "part-00009-foreign.tfrecord.gz"), "wb") as f:
+            f.write(_gzip.compress(raw, 6))
+        df = stf.read_tfrecord(out, engine="gpu")
+        assert sorted(r["x"] for r in df.collect()) == list(range(300))

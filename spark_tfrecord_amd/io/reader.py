@@ -213,9 +213,9 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
     data_schema = None if deferred_infer else _resolve(schema)
 
     workers = min(32, (os.cpu_count() or 8))
-    pool = ThreadPoolExecutor(max_workers=workers)
     window = 2 * workers
     futures: dict = {}
+    _pool = P.shared_pool  # persistent process-wide pool (no spin-up)
 
     def _host_task(fpath: str):
         data = _load_file(fpath)
@@ -227,7 +227,7 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
 
     def _blob(i: int):
         if i not in futures:
-            futures[i] = pool.submit(_host_task, files[i])
+            futures[i] = _pool().submit(_host_task, files[i])
         return futures.pop(i).result()
 
     # GPU path: consecutive uncompressed files are decoded as ONE pipeline
@@ -288,7 +288,7 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
                 continue
             for j in range(i, min(i + window, len(files))):
                 if _needs_host_bytes(files[j]) and j not in futures:
-                    futures[j] = pool.submit(_host_task, files[j])
+                    futures[j] = _pool().submit(_host_task, files[j])
             metrics.add(files=1, nbytes=os.path.getsize(fpath))
             got = _blob(i)
             i += 1
@@ -305,7 +305,8 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
             t = batch_to_table(batch)
             _append_with_parts(t, [fpath], [t.num_rows])
     finally:
-        pool.shutdown(wait=False)
+        for f in futures.values():  # abandon prefetches from a failed read
+            f.cancel()
     if not tables:
         full = StructType(list(data_schema.fields) +
                           [StructField(c, StringType(), True) for c in part_cols])

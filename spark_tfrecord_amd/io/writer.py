@@ -124,41 +124,57 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
         if metrics is not None:
             metrics.add(rows=rows, nbytes=len(payload), files=1)
 
-    with ThreadPoolExecutor(max_workers=min(8, max(num_shards, 1))) as pool:
-        futs = []
-        tag_futs = {}
-        for s in range(num_shards):
-            lo, hi = int(bounds[s]), int(bounds[s + 1])
-            if num_shards > 1 and hi == lo:
-                continue
-            chunk = table.slice(lo, hi - lo)
-            batch = table_to_batch(chunk, schema)
-            fname = P.part_file_name(shard_offset + s, codec, job_id)
-            fpath = os.path.join(out_dir, fname)
-            if eng == "gpu":
-                from ..engine import gpu as gpu_engine
+    if num_shards == 1:
+        # no pool for a single shard: thread spin-up dominates small writes
+        batch = table_to_batch(table, schema)
+        fpath = os.path.join(out_dir, P.part_file_name(shard_offset, codec,
+                                                       job_id))
+        if eng == "gpu":
+            from ..engine import gpu as gpu_engine
 
-                # encode on device, ONE pinned D2H, write/compress on a
-                # worker thread while the next shard encodes. (A mapped-DMA
-                # write to the fresh temp inode pays ~0.16 ms/MB of
-                # hipHostRegister each time — slower than the page-cache
-                # write itself; the mapped path is for stable-inode rewrites
-                # like engine-level write_batch_to_file.)
-                dev_batch = gpu_engine.batch_to_device(batch)
-                img = gpu_engine.encode_device(dev_batch, record_type)
-                tag = f"encw{s % 2}"
-                prev = tag_futs.get(tag)
-                if prev is not None:
-                    prev.result()  # the tag's pinned buffer is being reused
-                raw = gpu_engine.device_to_pinned_view(img, tag=tag)
-            else:
-                raw = cpu_engine.encode_batch(batch, record_type)
-            f = pool.submit(_compress_write, raw, fpath, hi - lo)
-            if eng == "gpu":
-                tag_futs[tag] = f
-            futs.append(f)
-        for f in futs:
-            f.result()
+            img = gpu_engine.encode_device(
+                gpu_engine.batch_to_device(batch), record_type)
+            raw = gpu_engine.device_to_pinned_view(img, tag="encw0")
+        else:
+            raw = cpu_engine.encode_batch(batch, record_type)
+        _compress_write(raw, fpath, R)
+        return
+
+    pool = P.shared_pool()
+    futs = []
+    tag_futs = {}
+    for s in range(num_shards):
+        lo, hi = int(bounds[s]), int(bounds[s + 1])
+        if num_shards > 1 and hi == lo:
+            continue
+        chunk = table.slice(lo, hi - lo)
+        batch = table_to_batch(chunk, schema)
+        fname = P.part_file_name(shard_offset + s, codec, job_id)
+        fpath = os.path.join(out_dir, fname)
+        if eng == "gpu":
+            from ..engine import gpu as gpu_engine
+
+            # encode on device, ONE pinned D2H, write/compress on a
+            # worker thread while the next shard encodes. (A mapped-DMA
+            # write to the fresh temp inode pays ~0.16 ms/MB of
+            # hipHostRegister each time — slower than the page-cache
+            # write itself; the mapped path is for stable-inode rewrites
+            # like engine-level write_batch_to_file.)
+            dev_batch = gpu_engine.batch_to_device(batch)
+            img = gpu_engine.encode_device(dev_batch, record_type)
+            tag = f"encw{s % 2}"
+            prev = tag_futs.get(tag)
+            if prev is not None:
+                prev.result()  # the tag's pinned buffer is being reused
+            raw = gpu_engine.device_to_pinned_view(img, tag=tag)
+        else:
+            raw = cpu_engine.encode_batch(batch, record_type)
+        f = pool.submit(_compress_write, raw, fpath, hi - lo)
+        if eng == "gpu":
+            tag_futs[tag] = f
+        futs.append(f)
+    for f in futs:
+        f.result()
 
 
 def write_tfrecord(data, path: str, record_type: str = "Example",
@@ -232,7 +248,6 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
             # mmap registration would cost ~0.16 ms/MB each on the fresh
             # temp inodes — the r01 config-3 cliff)
             view = gpu_engine.device_to_pinned_view(img)
-            from concurrent.futures import ThreadPoolExecutor
 
             def _write_part(p, lo, hi):
                 sub_dir = part_dir(combos[p])
@@ -243,10 +258,9 @@ def write_tfrecord(data, path: str, record_type: str = "Example",
                 P.write_file_atomic(payload, fpath)
                 return len(payload)
 
-            with ThreadPoolExecutor(max_workers=min(16, len(ranges) or 1)) as ex:
-                for nb in ex.map(lambda a: _write_part(*a), ranges):
-                    if metrics is not None:
-                        metrics.add(nbytes=nb, files=1)
+            for nb in P.shared_pool().map(lambda a: _write_part(*a), ranges):
+                if metrics is not None:
+                    metrics.add(nbytes=nb, files=1)
             if metrics is not None:
                 metrics.add(rows=table.num_rows)
         else:

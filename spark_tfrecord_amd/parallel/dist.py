@@ -341,16 +341,14 @@ def write_tfrecord_distributed(data, path: str, record_type: str = "Example",
                     payload if isinstance(payload, bytes)
                     else payload.tobytes(), codec), fpath)
 
-        from concurrent.futures import ThreadPoolExecutor
-
         njobs = len(dev_jobs) + len(host_jobs)
         if njobs:
-            with ThreadPoolExecutor(max_workers=min(16, njobs)) as ex:
-                futs = [ex.submit(_write_combo, c, view[lo:hi])
-                        for c, lo, hi in dev_jobs]
-                futs += [ex.submit(_write_combo, c, b) for c, b in host_jobs]
-                for f in futs:
-                    f.result()
+            ex = P.shared_pool()
+            futs = [ex.submit(_write_combo, c, view[lo:hi])
+                    for c, lo, hi in dev_jobs]
+            futs += [ex.submit(_write_combo, c, b) for c, b in host_jobs]
+            for f in futs:
+                f.result()
 
     dist.barrier()
     if rank == 0:

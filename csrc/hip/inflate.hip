@@ -85,28 +85,46 @@ __device__ inline u32 get_len4(const u8* a, int i) {
 struct BitRd {
   const u8* p;
   const u8* end;
-  u64 buf;
-  int n;  // bits buffered; < 0 => underflow (sticky error)
+  u64 buf;    // active bits
+  int n;      // bit count; < 0 => underflow (sticky error)
+  u64 pre;    // PREFETCHED next input bytes (low-order first)
+  int pre_n;  // valid bytes in pre
 };
 
-// Word-wise refill: ONE unaligned u64 load tops the buffer up to >=57 bits
-// (byte-at-a-time refills cost a dependent global load per input byte — at
-// the inflater's intrinsically low occupancy every one of those stalls the
-// wave for the full L2 latency; measured 15x slower).
-__device__ inline void br_refill(BitRd& b) {
+// The next input word is loaded the moment the previous one is consumed —
+// >=4 symbols of decode separate the load from its first use, hiding the
+// global-load latency that otherwise sits on the serial decode chain
+// (byte-at-a-time refills measured 15x slower; unprefetched word refills
+// still stalled every ~4 symbols at the inflater's 1-wave/CU occupancy).
+__device__ inline void br_load_pre(BitRd& b) {
   if (b.end - b.p >= 8) {
-    u64 w;
-    __builtin_memcpy(&w, b.p, 8);
-    int take = (64 - b.n) >> 3;
-    b.buf |= w << b.n;
-    b.p += take;
-    b.n += take << 3;
-    return;
+    __builtin_memcpy(&b.pre, b.p, 8);
+    b.p += 8;
+    b.pre_n = 8;
+  } else {
+    b.pre = 0;
+    b.pre_n = (int)(b.end - b.p);
+    for (int i = 0; i < b.pre_n; ++i) b.pre |= (u64)b.p[i] << (8 * i);
+    b.p = b.end;
   }
-  while (b.n <= 56 && b.p < b.end) {
-    b.buf |= (u64)(*b.p++) << b.n;
-    b.n += 8;
-  }
+}
+
+__device__ inline void br_init(BitRd& b, const u8* in, i64 ilen) {
+  b.p = in;
+  b.end = in + ilen;
+  b.buf = 0;
+  b.n = 0;
+  br_load_pre(b);
+}
+
+__device__ inline void br_refill(BitRd& b) {
+  int take = (64 - b.n) >> 3;
+  if (take > b.pre_n) take = b.pre_n;
+  b.buf |= b.pre << b.n;  // bits past 64 truncate; only `take` bytes counted
+  b.n += take << 3;
+  b.pre = (take >= 8) ? 0 : (b.pre >> (take << 3));
+  b.pre_n -= take;
+  if (b.pre_n == 0 && b.p < b.end) br_load_pre(b);
 }
 
 __device__ inline u32 br_bits(BitRd& b, int k) {
@@ -135,7 +153,10 @@ __device__ inline bool build_huff4(const u8* lens4, int len_off, int nsym,
   u32 code = 0, k = 0;
   for (int l = 1; l <= 15; ++l) {
     u32 cnt = rank[l];
-    bc[l] = (cnt << 16) | code;
+    // watershed form: hit at length l iff rev15 < lim (both 15-bit
+    // justified); symbol = sym[(rev15 >> (15-l)) + offset] with offset =
+    // rank - base in wraparound u16 arithmetic
+    bc[l] = (((code + cnt) << (15 - l)) << 16) | (uint16_t)(k - code);
     rank[l] = (uint16_t)k;
     k += cnt;
     code = (code + cnt) << 1;
@@ -160,7 +181,7 @@ __device__ inline bool build_huff16(const uint16_t* lens, int nsym,
   u32 code = 0, k = 0;
   for (int l = 1; l <= 15; ++l) {
     u32 cnt = rank[l];
-    bc[l] = (cnt << 16) | code;
+    bc[l] = (((code + cnt) << (15 - l)) << 16) | (uint16_t)(k - code);
     rank[l] = (uint16_t)k;
     k += cnt;
     code = (code + cnt) << 1;
@@ -183,26 +204,24 @@ __device__ inline bool build_huff16(const uint16_t* lens, int nsym,
 // VALU/busy = 0.04 with the load-per-iteration form).
 __device__ inline int huff_decode(BitRd& br, const u32* bc,
                                   const uint16_t* rank, const uint16_t* sym) {
+  (void)rank;  // folded into bc as (rank - base); kept for the builders
   if (br.n < 15) br_refill(br);  // short tail: zero bits pad the peek
   u32 rev = __brev((u32)br.buf) >> 17;
   u32 b1 = bc[1], b2 = bc[2], b3 = bc[3], b4 = bc[4];
   u32 b5 = bc[5], b6 = bc[6], b7 = bc[7], b8 = bc[8];
-  u32 idx;
   int l;
-  if ((idx = (rev >> 14) - (b1 & 0xFFFFu)) < (b1 >> 16)) l = 1;
-  else if ((idx = (rev >> 13) - (b2 & 0xFFFFu)) < (b2 >> 16)) l = 2;
-  else if ((idx = (rev >> 12) - (b3 & 0xFFFFu)) < (b3 >> 16)) l = 3;
-  else if ((idx = (rev >> 11) - (b4 & 0xFFFFu)) < (b4 >> 16)) l = 4;
-  else if ((idx = (rev >> 10) - (b5 & 0xFFFFu)) < (b5 >> 16)) l = 5;
-  else if ((idx = (rev >> 9) - (b6 & 0xFFFFu)) < (b6 >> 16)) l = 6;
-  else if ((idx = (rev >> 8) - (b7 & 0xFFFFu)) < (b7 >> 16)) l = 7;
-  else if ((idx = (rev >> 7) - (b8 & 0xFFFFu)) < (b8 >> 16)) l = 8;
+  if (rev < (b1 >> 16)) l = 1;
+  else if (rev < (b2 >> 16)) l = 2;
+  else if (rev < (b3 >> 16)) l = 3;
+  else if (rev < (b4 >> 16)) l = 4;
+  else if (rev < (b5 >> 16)) l = 5;
+  else if (rev < (b6 >> 16)) l = 6;
+  else if (rev < (b7 >> 16)) l = 7;
+  else if (rev < (b8 >> 16)) l = 8;
   else {
     l = 16;
     for (int k = 9; k <= 15; ++k) {
-      u32 w = bc[k];
-      idx = (rev >> (15 - k)) - (w & 0xFFFFu);
-      if (idx < (w >> 16)) {
+      if (rev < (bc[k] >> 16)) {
         l = k;
         break;
       }
@@ -213,9 +232,12 @@ __device__ inline int huff_decode(BitRd& br, const u32* bc,
     br.n = -1 << 20;  // consumed past the stream end
     return -1;
   }
+  u32 w = (l <= 4 ? (l <= 2 ? (l == 1 ? b1 : b2) : (l == 3 ? b3 : b4))
+                  : (l <= 8 ? (l <= 6 ? (l == 5 ? b5 : b6) : (l == 7 ? b7 : b8))
+                            : bc[l]));
   br.buf >>= l;
   br.n -= l;
-  return sym[rank[l] + idx];
+  return sym[(uint16_t)((rev >> (15 - l)) + (u32)(uint16_t)w)];
 }
 
 // Inflate one raw-deflate segment into dst[0, expect). Returns 0 on
@@ -227,12 +249,15 @@ __device__ inline int huff_decode(BitRd& br, const u32* bc,
 __device__ inline int inflate_one(const u8* __restrict__ in, i64 ilen,
                                   u8* __restrict__ dst, i64 expect,
                                   LaneScratch& L) {
-  BitRd br{in, in + ilen, 0, 0};
+  BitRd br;
+  br_init(br, in, ilen);
   i64 opos = 0;
   for (;;) {
     // a non-final segment ends after its full-flush empty stored block:
     // all output produced and fewer bits left than any block needs
-    if (opos >= expect && (i64)(br.end - br.p) * 8 + br.n < 10) break;
+    if (opos >= expect &&
+        ((i64)(br.end - br.p) + br.pre_n) * 8 + br.n < 10)
+      break;
     u32 final = br_bits(br, 1);
     u32 btype = br_bits(br, 2);
     if (br.n < 0) return 1;
@@ -242,7 +267,8 @@ __device__ inline int inflate_one(const u8* __restrict__ in, i64 ilen,
       u32 len = br_bits(br, 16);
       u32 nlen = br_bits(br, 16);
       if (br.n < 0 || ((len ^ nlen) & 0xFFFFu) != 0xFFFFu) return 2;
-      const u8* src = br.p - (br.n >> 3);  // rewind buffered bytes
+      // rewind both the bit buffer's and the prefetch register's bytes
+      const u8* src = br.p - br.pre_n - (br.n >> 3);
       if (src + len > br.end || opos + (i64)len > expect) return 3;
       u8* dp = dst + opos;
       u32 i = 0;
@@ -256,6 +282,9 @@ __device__ inline int inflate_one(const u8* __restrict__ in, i64 ilen,
       br.p = src + len;
       br.buf = 0;
       br.n = 0;
+      br.pre = 0;
+      br.pre_n = 0;
+      br_load_pre(br);
       if (final) break;
       continue;
     }

@@ -28,6 +28,7 @@
 #include <vector>
 
 #include "codec_core.h"
+#include "inflate_core.h"
 
 namespace py = pybind11;
 using namespace tfrec;
@@ -744,6 +745,20 @@ py::dict scan_stats_debug(py::buffer data, py::array_t<i64> rec_off,
   return out;
 }
 
+// Host run of the DEFLATE segment inflater core shared with the gfx950
+// kernel (csrc/inflate_core.h): backs CPU tests of the device algorithm.
+py::bytes host_inflate_segment(py::buffer comp, i64 expect) {
+  auto info = comp.request();
+  std::string out(static_cast<size_t>(expect), '\0');
+  tfrec::inflate::LaneScratch L;
+  int rc = tfrec::inflate::inflate_one(
+      static_cast<const u8*>(info.ptr), static_cast<i64>(info.size),
+      reinterpret_cast<u8*>(&out[0]), expect, L);
+  if (rc)
+    throw std::runtime_error("inflate_one failed: cause " + std::to_string(rc));
+  return py::bytes(out);
+}
+
 u32 crc32c_py(py::buffer data) {
   py::buffer_info info;
   BufView buf = as_bytes(data, info);
@@ -778,6 +793,9 @@ PYBIND11_MODULE(_native, m) {
   m.def("crc32c_combine",
         [](u32 c1, u32 c2, u64 len2) { return crc32c_combine(c1, c2, len2); },
         "crc(A||B) from crc(A), crc(B), len(B) (GF(2) shift operator)");
+  m.def("host_inflate_segment", &host_inflate_segment, py::arg("comp"),
+        py::arg("expect"),
+        "TEST-ONLY: run the device inflater's core on host for one segment");
   m.def("crc32c_combine_fast",
         [](u32 c1, u32 c2, u64 len2) { return crc32c_combine_fast(c1, c2, len2); },
         "crc(A||B) via register-only GF(2^32) field multiply (wave-CRC form)");

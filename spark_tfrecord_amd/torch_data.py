@@ -134,6 +134,19 @@ class TFRecordIterableDataset(torch.utils.data.IterableDataset):
                 fpath, sub_schema, self.record_type, self.verify_crc)
             torch.cuda.current_stream().synchronize()  # hand off across threads
             return batch
+        if eng == "gpu" and P.codec_from_path(fpath) == "gzip":
+            # our gzip shards inflate ON the training GPU (segment table)
+            from .engine import gpu as gpu_engine
+            dev = gpu_engine.read_gzip_file_to_device(fpath)
+            if dev is not None:
+                if dev.numel() == 0:
+                    return None
+                off, lens = gpu_engine.scan_frames_device(dev)
+                batch = gpu_engine.decode_device(
+                    dev, off, lens, sub_schema, self.record_type,
+                    self.verify_crc)
+                torch.cuda.current_stream().synchronize()
+                return batch
         from .engine import cpu as cpu_engine
         data = np.frombuffer(P.decompress_file(fpath), np.uint8)
         if data.size == 0:

@@ -11,6 +11,14 @@
 
 #include "crc32c.h"  // TFR_HOSTDEV
 
+// Wave-visibility fence for the cooperative copies (no-op on host; the
+// builtin needs no runtime header, unlike __threadfence_block).
+#if defined(__HIP_DEVICE_COMPILE__)
+#define TFR_WAVE_FENCE() __builtin_amdgcn_fence(__ATOMIC_ACQ_REL, "workgroup")
+#else
+#define TFR_WAVE_FENCE() ((void)0)
+#endif
+
 namespace tfrec {
 namespace inflate {
 
@@ -226,15 +234,43 @@ TFR_HOSTDEV inline int huff_decode(BitRd& br, const u32* bc,
   return sym[(uint16_t)((rev >> (15 - l)) + (u32)(uint16_t)w)];
 }
 
+// Bulk copy: serial on host (lane < 0); lane-strided u64s when a whole
+// wave runs ONE segment uniformly (the GPU decomposition — per-lane
+// segments paid a ~5x SIMT divergence tax on 64 independent bitstreams).
+// Only used where src/dst cannot overlap within the stride window.
+TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane) {
+  if (lane < 0) {
+    i64 i = 0;
+    for (; i + 8 <= n; i += 8) {
+      u64 w;
+      __builtin_memcpy(&w, src + i, 8);
+      __builtin_memcpy(dst + i, &w, 8);
+    }
+    for (; i < n; ++i) dst[i] = src[i];
+    return;
+  }
+  for (i64 i = (i64)lane * 8; i + 8 <= n; i += 64 * 8) {
+    u64 w;
+    __builtin_memcpy(&w, src + i, 8);
+    __builtin_memcpy(dst + i, &w, 8);
+  }
+  i64 tail = n & ~((i64)7);
+  for (i64 b = tail + lane; b < n; b += 64) dst[b] = src[b];
+  TFR_WAVE_FENCE();  // other lanes may read these bytes (later matches)
+}
+
 // Inflate one raw-deflate segment into dst[0, expect). Returns 0 on
 // success, a small nonzero cause code otherwise (any nonzero => the Python
 // side redoes the FILE on the host zlib path). `in`/`dst` are __restrict__:
 // without it the compiler must order every bit-refill load after all
 // pending output stores (they could alias), serializing the decode on L2
-// store latency.
+// store latency. `lane` < 0 = host/serial; otherwise the caller runs the
+// WHOLE WAVE through this function in lockstep on one segment — every
+// value is wave-uniform (the compiler keeps it in scalar registers), and
+// `lane` only splits the big copies.
 TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
                                   u8* __restrict__ dst, i64 expect,
-                                  LaneScratch& L) {
+                                  LaneScratch& L, int lane = -1) {
   BitRd br;
   br_init(br, in, ilen);
   i64 opos = 0;
@@ -256,14 +292,7 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
       // rewind both the bit buffer's and the prefetch register's bytes
       const u8* src = br.p - br.pre_n - (br.n >> 3);
       if (src + len > br.end || opos + (i64)len > expect) return 3;
-      u8* dp = dst + opos;
-      u32 i = 0;
-      for (; i + 8 <= len; i += 8) {
-        u64 w;
-        __builtin_memcpy(&w, src + i, 8);
-        __builtin_memcpy(dp + i, &w, 8);
-      }
-      for (; i < len; ++i) dp[i] = src[i];
+      bulk_copy(dst + opos, src, (i64)len, lane);
       opos += len;
       br.p = src + len;
       br.buf = 0;
@@ -366,7 +395,9 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
         if (dist > opos || opos + mlen > expect) return 16;
         const u8* sp = dst + (opos - dist);
         u8* dp = dst + opos;
-        if (dist >= 8) {
+        if (lane >= 0 && dist >= mlen) {
+          bulk_copy(dp, sp, mlen, lane);  // non-overlapping: wave-strided
+        } else if (dist >= 8) {
           i64 i = 0;
           for (; i + 8 <= mlen; i += 8) {
             u64 w;

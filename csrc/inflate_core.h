@@ -45,10 +45,10 @@ constexpr uint8_t kClOrder[19] = {16, 17, 18, 0, 8,  7, 9,
                                                 6,  10, 5,  11, 4, 12, 3,
                                                 13, 2,  14, 1,  15};
 
-// Per-lane LDS scratch: ~1 KB x 64 lanes < 64 KiB static shared. The decode
-// loop's per-length lookup is ONE u32 load — (count << 16) | base packed —
-// with the rank read only on the hit (the loop is the latency chain at the
-// inflater's low occupancy; separate base/count/rank reads tripled it).
+// Decode scratch (~1 KB): ONE per wave on the GPU (all 64 lanes run the
+// same segment in lockstep, so accesses are wave-uniform broadcasts), or
+// plain stack state on the host. bc_* pack the watershed limit and the
+// (rank - base) offset into one u32 per code length.
 struct LaneScratch {
   u32 bc_lit[16];          // (cnt << 16) | first canonical code, per length
   u32 bc_dist[16];

@@ -1,8 +1,9 @@
 // DEFLATE (RFC 1951) segment inflater core, shared host/device.
 //
-// The gfx950 kernel (csrc/hip/inflate.hip) wires this to one-segment-per-
-// lane with per-lane LDS scratch; the host build backs the CPU tests (the
-// algorithm is bit-exact on both) and a reference path for debugging.
+// The gfx950 kernel (csrc/hip/inflate.hip) runs this one-segment-per-WAVE
+// (64 lanes in lockstep on one bitstream — wave-uniform values, lane-split
+// bulk copies, LDS scratch per wave); the host build backs the CPU tests
+// (the algorithm is bit-exact on both) and a reference path for debugging.
 // Design notes: docs/KERNELS.md "inflate_segments_kernel".
 #pragma once
 

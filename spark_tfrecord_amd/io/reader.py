@@ -117,13 +117,24 @@ def count_tfrecord(path: str, engine: str = "auto") -> int:
     eng = engine_mod.resolve_engine(engine)
     total = 0
     for f in files:
-        if eng == "gpu" and P.codec_from_path(f) is None:
+        if eng == "gpu" and P.codec_from_path(f) in (None, "gzip"):
             if os.path.getsize(f) == 0:
                 continue
             from ..engine import gpu as gpu_engine
 
-            off, _ = gpu_engine.scan_frames_device(
-                gpu_engine.read_file_to_device(f))
+            if P.codec_from_path(f) == "gzip":
+                data = gpu_engine.read_gzip_file_to_device(f)
+                if data is None:  # foreign gzip: host count below
+                    data_np = _load_file(f)
+                    if data_np.size:
+                        off_h, _ = _native.scan_frames(data_np, False)
+                        total += len(off_h)
+                    continue
+                if data.numel() == 0:
+                    continue
+            else:
+                data = gpu_engine.read_file_to_device(f)
+            off, _ = gpu_engine.scan_frames_device(data)
             total += int(off.numel())
         else:
             data = _load_file(f)

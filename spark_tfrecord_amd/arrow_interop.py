@@ -318,6 +318,14 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
         off = np.frombuffer(a.buffers()[1], np.int64, len(a) + 1 + a.offset)[a.offset:]
         data_buf = a.buffers()[2]
         data = np.frombuffer(data_buf, np.uint8, len(data_buf)) if data_buf else np.zeros(0, np.uint8)
+        if null_count == 0:
+            # all-present: elements are a contiguous byte subrange — no gather
+            # (a sliced chunk has off[0] != 0; rebase instead of copying)
+            base = int(off[0]) if R else 0
+            return WireColumn(kind, False, presence,
+                              np.arange(R + 1, dtype=np.int64),
+                              data[base:int(off[-1])] if R else data[:0],
+                              off.astype(np.int64) - base)
         lens = (off[1:] - off[:-1]).astype(np.int64)
         lens_present = lens[mask]
         starts = off[:-1][mask].astype(np.int64)
@@ -334,11 +342,14 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
         off = _np_offsets(arr)
         lens = off[1:] - off[:-1]
         lens = np.where(mask, lens, 0)
-        vals_arr = arr.values  # child array
+        vals_arr = arr.values  # FULL child array (ignores the slice window)
         child = vals_arr.to_numpy(zero_copy_only=False)
-        if null_count == 0 and arr.offset == 0 and off[0] == 0:
-            vals = numeric_cast(child[: off[-1]])
-            return WireColumn(kind, False, presence, off.copy(), vals)
+        if null_count == 0:
+            # all-present: the slice's values are the contiguous child range
+            # [off[0], off[-1]) — view it and rebase the offsets, no gather
+            base = int(off[0]) if R else 0
+            vals = numeric_cast(child[base:int(off[-1])] if R else child[:0])
+            return WireColumn(kind, False, presence, off - base, vals)
         starts = off[:-1][mask]
         idx = _ragged_gather_idx(starts.astype(np.int64), lens[mask])
         vals = numeric_cast(child[idx])
@@ -349,41 +360,45 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
     # Fast path: 1-D string/binary list (BytesList arrays like token lists)
     if not seq and isinstance(dt, ArrayType) and kind == KIND_BYTES and \
             (pa.types.is_list(arr.type) or pa.types.is_large_list(arr.type)):
-        inner = arr.values
-        if (null_count == 0 and arr.offset == 0 and inner.null_count == 0
+        inner = arr.values  # FULL child array (ignores the slice window)
+        if (null_count == 0 and inner.null_count == 0
                 and (pa.types.is_string(inner.type)
                      or pa.types.is_large_string(inner.type)
                      or pa.types.is_binary(inner.type)
                      or pa.types.is_large_binary(inner.type))):
             off1 = _np_offsets(arr).astype(np.int64)
             a = inner.cast(pa.large_binary())
-            if a.offset == 0 and off1[0] == 0:
-                E = int(off1[-1])
+            if a.offset == 0 and R:
+                # slice window = elements [E0,E1) = bytes [eoff[E0],eoff[E1])
+                E0, E1 = int(off1[0]), int(off1[-1])
                 eoff = np.frombuffer(a.buffers()[1], np.int64, len(a) + 1)
                 db = a.buffers()[2]
                 data = (np.frombuffer(db, np.uint8, len(db)) if db
                         else np.zeros(0, np.uint8))
-                return WireColumn(kind, False, presence, off1,
-                                  data[:int(eoff[E])], eoff[:E + 1])
+                b0 = int(eoff[E0])
+                return WireColumn(kind, False, presence, off1 - E0,
+                                  data[b0:int(eoff[E1])],
+                                  eoff[E0:E1 + 1] - b0)
 
     # Fast path: 2-D numeric nested list (SequenceExample FeatureList of
     # Int64List/FloatList). Offsets compose: values-per-row cumulative is
     # off2[off1] — no python loop over 1M ragged rows.
     if seq and kind != KIND_BYTES and \
             (pa.types.is_list(arr.type) or pa.types.is_large_list(arr.type)):
-        inner = arr.values
-        if (null_count == 0 and arr.offset == 0 and inner.offset == 0
-                and inner.null_count == 0
+        inner = arr.values  # FULL child list array (ignores the slice window)
+        if (null_count == 0 and inner.offset == 0 and inner.null_count == 0
                 and (pa.types.is_list(inner.type)
-                     or pa.types.is_large_list(inner.type))):
+                     or pa.types.is_large_list(inner.type)) and R):
             off1 = _np_offsets(arr).astype(np.int64)    # [R+1] sublists/row
             off2 = _np_offsets(inner).astype(np.int64)  # [L+1] values/sublist
-            if off1[0] == 0 and off2[0] == 0:
+            if off2[0] == 0:
+                # slice window = sublists [L0,L1) = values [off2[L0],off2[L1])
                 child = inner.values.to_numpy(zero_copy_only=False)
-                L = int(off1[-1])
-                vals = numeric_cast(child[: int(off2[L])])
-                return WireColumn(kind, True, presence, off2[off1].copy(), vals,
-                                  None, off1.copy(), off2[: L + 1].copy())
+                L0, L1 = int(off1[0]), int(off1[-1])
+                v0 = int(off2[L0])
+                vals = numeric_cast(child[v0:int(off2[L1])])
+                return WireColumn(kind, True, presence, off2[off1] - v0, vals,
+                                  None, off1 - L0, off2[L0:L1 + 1] - v0)
 
     # General fallback: python objects
     return column_from_values(arr.to_pylist(), dt, nullable, name)

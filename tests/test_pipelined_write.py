@@ -82,3 +82,76 @@ class TestPipelinedSingleShard:
         out = str(tmp_path / "gz")
         stf.write_tfrecord(_table(1_000), out, codec="gzip", engine="cpu")
         assert _read_ids(out) == list(range(1_000))
+
+
+class TestSlicedConversion:
+    """arrow_to_wire on sliced (offset != 0) chunks must take the rebased
+    contiguous fast paths, not the gather/to_pylist fallbacks, and must
+    produce values identical to converting the whole table."""
+
+    def _roundtrip_eq(self, tab, schema=None):
+        from spark_tfrecord_amd.arrow_interop import (schema_from_arrow,
+                                                      table_to_batch)
+        from spark_tfrecord_amd.engine import cpu as cpu_engine
+        sch = schema or schema_from_arrow(tab.schema)
+        whole = cpu_engine.encode_batch(table_to_batch(tab, sch), "Example")
+        parts = []
+        step = max(1, tab.num_rows // 3)
+        for lo in range(0, tab.num_rows, step):
+            ch = tab.slice(lo, min(step, tab.num_rows - lo))
+            parts.append(cpu_engine.encode_batch(table_to_batch(ch, sch),
+                                                 "Example"))
+        assert b"".join(parts) == bytes(whole)
+
+    def test_numeric_scalar_and_list(self):
+        rng = np.random.default_rng(0)
+        off = np.concatenate([[0], np.cumsum(rng.integers(0, 5, 100))])
+        vals = rng.integers(-9, 9, off[-1]).astype(np.int64)
+        tab = pa.table({
+            "a": np.arange(100, dtype=np.int64),
+            "f": rng.random(100).astype(np.float32),
+            "l": pa.LargeListArray.from_arrays(off.astype(np.int64), vals),
+        })
+        self._roundtrip_eq(tab)
+
+    def test_string_scalar_and_list(self):
+        rng = np.random.default_rng(1)
+        strs = [f"v{i}" * (i % 4) for i in range(100)]
+        off = np.concatenate([[0], np.cumsum(rng.integers(0, 4, 100))])
+        toks = pa.array([f"t{i%13}" for i in range(off[-1])])
+        tab = pa.table({
+            "s": pa.array(strs),
+            "ts": pa.LargeListArray.from_arrays(off.astype(np.int64),
+                                                toks.cast(pa.large_string())),
+        })
+        self._roundtrip_eq(tab)
+
+    def test_nested_2d_sequence(self):
+        from spark_tfrecord_amd.arrow_interop import table_to_batch
+        rng = np.random.default_rng(2)
+        o2 = np.concatenate([[0], np.cumsum(rng.integers(0, 4, 50))])
+        inner = pa.LargeListArray.from_arrays(
+            o2.astype(np.int64), rng.random(o2[-1]).astype(np.float32))
+        o1 = np.concatenate([[0], np.cumsum(rng.integers(0, 3, 20))])
+        assert o1[-1] <= 50
+        outer = pa.LargeListArray.from_arrays(
+            o1.astype(np.int64), inner.slice(0, int(o1[-1])))
+        tab = pa.table({"seq": outer})
+        sch = stf.StructType([stf.StructField(
+            "seq", stf.ArrayType(stf.ArrayType(stf.FloatType())), True)])
+        from spark_tfrecord_amd.engine import cpu as cpu_engine
+        whole = cpu_engine.encode_batch(table_to_batch(tab, sch),
+                                        "SequenceExample")
+        parts = []
+        for lo in range(0, 20, 7):
+            ch = tab.slice(lo, min(7, 20 - lo))
+            parts.append(cpu_engine.encode_batch(table_to_batch(ch, sch),
+                                                 "SequenceExample"))
+        assert b"".join(parts) == bytes(whole)
+
+    def test_sliced_with_nulls(self):
+        tab = pa.table({
+            "x": pa.array([1, None, 3, None, 5, 6, None, 8], type=pa.int64()),
+            "s": pa.array(["a", None, "c", "d", None, "f", "g", "h"]),
+        })
+        self._roundtrip_eq(tab)

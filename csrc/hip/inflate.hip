@@ -61,13 +61,14 @@ __global__ void __launch_bounds__(256) inflate_segments_kernel(
   // ~12 KiB/block, still far below the 13-block LDS ceiling — occupancy
   // stays register-limited at 5 waves/SIMD.
   __shared__ uint16_t T[4][tfrec::inflate::kLitTabSize];
+  __shared__ uint16_t D[4][tfrec::inflate::kDistTabSize];
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   i64 wave = blockIdx.x * 4 + wid;
   i64 nwaves = (i64)gridDim.x * 4;
   for (i64 seg = wave; seg < nseg; seg += nwaves) {
     int rc = inflate_one(comp + in_off[seg], in_len[seg], out + out_off[seg],
-                         out_len[seg], S[wid], T[wid], lane);
+                         out_len[seg], S[wid], T[wid], D[wid], lane);
     if (rc && lane == 0)
       atomicMin(err, ((unsigned long long)(seg + 1) << 8) | (u32)rc);
   }

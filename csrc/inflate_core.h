@@ -201,17 +201,20 @@ TFR_HOSTDEV inline bool build_huff16(const uint16_t* lens, int nsym,
 // compare chain below, writing disjoint LDS slots.
 constexpr int kLitTabBits = 10;
 constexpr int kLitTabSize = 1 << kLitTabBits;
+constexpr int kDistTabBits = 8;
+constexpr int kDistTabSize = 1 << kDistTabBits;
 
-TFR_HOSTDEV inline void fill_lit_table(const u32* bc, const uint16_t* sym,
-                                       uint16_t* tab, int lane) {
+template <int BITS>
+TFR_HOSTDEV inline void fill_huff_table(const u32* bc, const uint16_t* sym,
+                                        uint16_t* tab, int lane) {
   int start = lane < 0 ? 0 : lane;
   int step = lane < 0 ? 1 : 64;
-  for (int idx = start; idx < kLitTabSize; idx += step) {
-    u32 rev = (u32)idx << (15 - kLitTabBits);  // left-justified prefix
+  for (int idx = start; idx < (1 << BITS); idx += step) {
+    u32 rev = (u32)idx << (15 - BITS);  // left-justified prefix
     u32 e = 0;
-    for (int l = 1; l <= kLitTabBits; ++l) {
+    for (int l = 1; l <= BITS; ++l) {
       if (rev < (bc[l] >> 16)) {
-        u32 s = sym[(uint16_t)((u32)(idx >> (kLitTabBits - l)) +
+        u32 s = sym[(uint16_t)((u32)(idx >> (BITS - l)) +
                                (u32)(uint16_t)bc[l])];
         e = (s << 4) | (u32)l;
         break;
@@ -222,15 +225,16 @@ TFR_HOSTDEV inline void fill_lit_table(const u32* bc, const uint16_t* sym,
   TFR_WAVE_FENCE();  // all lanes read every lane's entries
 }
 
-// Table-first decode for the lit/len alphabet; falls back to the compare
-// chain for code lengths 11..15 (rare — the table covers the whole
-// fixed-code alphabet and virtually all dynamic-code literals).
+// Table-first decode; falls back to the compare chain for code lengths
+// > BITS (rare — 10 bits cover the whole fixed-code lit alphabet and
+// virtually all dynamic-code literals; 8 bits cover typical distances).
+template <int BITS>
 TFR_HOSTDEV inline int huff_decode_tab(BitRd& br, const u32* bc,
                                        const uint16_t* sym,
                                        const uint16_t* tab) {
   if (br.n < 15) br_refill(br);  // short tail: zero bits pad the peek
   u32 rev = __builtin_bitreverse32((u32)br.buf) >> 17;
-  u32 e = tab[rev >> (15 - kLitTabBits)];
+  u32 e = tab[rev >> (15 - BITS)];
   if (e) {
     int l = (int)(e & 15u);
     if (br.n < l) {
@@ -242,7 +246,7 @@ TFR_HOSTDEV inline int huff_decode_tab(BitRd& br, const u32* bc,
     return (int)(e >> 4);
   }
   int l = 16;
-  for (int k = kLitTabBits + 1; k <= 15; ++k) {
+  for (int k = BITS + 1; k <= 15; ++k) {
     if (rev < (bc[k] >> 16)) {
       l = k;
       break;
@@ -340,6 +344,7 @@ TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane) {
 TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
                                   u8* __restrict__ dst, i64 expect,
                                   LaneScratch& L, uint16_t* __restrict__ lit_tab,
+                                  uint16_t* __restrict__ dist_tab,
                                   int lane = -1) {
   BitRd br;
   br_init(br, in, ilen);
@@ -425,7 +430,8 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
     if (!build_huff4(L.lens4, dist_off, hdist, L.bc_dist, L.rank_dist,
                      &L.sym[288]))
       return 10;
-    fill_lit_table(L.bc_lit, L.sym, lit_tab, lane);
+    fill_huff_table<kLitTabBits>(L.bc_lit, L.sym, lit_tab, lane);
+    fill_huff_table<kDistTabBits>(L.bc_dist, &L.sym[288], dist_tab, lane);
     // literal accumulation window: byte-per-literal global stores made the
     // literal-heavy path store-bound; 8 literals flush as one u64 store
     // (flushed before matches, which may read the freshly-written bytes)
@@ -445,7 +451,7 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
       ln = 0;
     };
     for (;;) {
-      int s = huff_decode_tab(br, L.bc_lit, L.sym, lit_tab);
+      int s = huff_decode_tab<kLitTabBits>(br, L.bc_lit, L.sym, lit_tab);
       if (s < 0) return 11;
       if (s < 256) {
         if (opos + ln >= expect) return 12;
@@ -459,7 +465,8 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
         s -= 257;
         if (s >= 29) return 13;
         i64 mlen = kLenBase[s] + (i64)br_bits(br, kLenExtra[s]);
-        int d = huff_decode(br, L.bc_dist, L.rank_dist, &L.sym[288]);
+        int d = huff_decode_tab<kDistTabBits>(br, L.bc_dist, &L.sym[288],
+                                              dist_tab);
         if (d < 0 || d >= 30) return 14;
         i64 dist = kDistBase[d] + (i64)br_bits(br, kDistExtra[d]);
         if (br.n < 0) return 15;

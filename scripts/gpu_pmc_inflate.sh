@@ -4,7 +4,7 @@ mkdir -p gpurun_out/pmc_inf2
 export TMPDIR=/tmp
 rocprofv3 --list-avail > gpurun_out/pmc_avail2.txt 2>&1
 PMC=""
-for c in SQ_WAVES SQ_INSTS_VALU SQ_INSTS_SALU SQ_INSTS_SMEM SQ_INSTS_LDS SQ_INSTS_VMEM SQ_BUSY_CYCLES GRBM_GUI_ACTIVE; do
+for c in SQ_WAVES SQ_INSTS_VALU SQ_INSTS_SALU SQ_INSTS_LDS SQ_INSTS_VMEM SQ_WAIT_ANY SQ_ACTIVE_INST_ANY SQ_BUSY_CYCLES; do
   grep -qw "$c" gpurun_out/pmc_avail2.txt && PMC="$PMC $c"
 done
 echo "PMC set:$PMC" | tee gpurun_out/pmc_set2.txt

@@ -1188,12 +1188,30 @@ def batch_to_host(batch: RecordBatch) -> RecordBatch:
         if not t.is_cuda:
             return t.numpy()
         host = torch.empty_like(t, device="cpu", pin_memory=True)
-        st = streams[state["k"] % len(streams)]
-        state["k"] += 1
         if not state["used"]:
             for s in streams:
                 s.wait_stream(main)
             state["used"] = True
+        nbytes = t.numel() * t.element_size()
+        if nbytes >= (8 << 20) and t.is_contiguous() and len(streams) > 1:
+            # a single dominant tensor (ByteArray values, big value columns)
+            # would ride ONE stream under round-robin and cap at the
+            # single-blit rate (~21 GB/s measured); split it across both
+            flat_d = t.view(-1)
+            flat_h = host.view(-1)
+            n = flat_d.numel()
+            step = (n + len(streams) - 1) // len(streams)
+            for i, st in enumerate(streams):
+                lo = i * step
+                hi = min(n, lo + step)
+                if lo >= hi:
+                    break
+                with torch.cuda.stream(st):
+                    flat_h[lo:hi].copy_(flat_d[lo:hi], non_blocking=True)
+                t.record_stream(st)
+            return host.numpy()
+        st = streams[state["k"] % len(streams)]
+        state["k"] += 1
         with torch.cuda.stream(st):
             host.copy_(t, non_blocking=True)
         t.record_stream(st)

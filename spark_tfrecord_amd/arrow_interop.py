@@ -320,11 +320,12 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
         data = np.frombuffer(data_buf, np.uint8, len(data_buf)) if data_buf else np.zeros(0, np.uint8)
         if null_count == 0:
             # all-present: elements are a contiguous byte subrange — no gather
-            # (a sliced chunk has off[0] != 0; rebase instead of copying)
-            base = int(off[0]) if R else 0
+            # (a sliced chunk has off[0] != 0; rebase instead of copying —
+            # including R == 0, where elem_off must still start at 0)
+            base = int(off[0])
             return WireColumn(kind, False, presence,
                               np.arange(R + 1, dtype=np.int64),
-                              data[base:int(off[-1])] if R else data[:0],
+                              data[base:int(off[-1])],
                               off.astype(np.int64) - base)
         lens = (off[1:] - off[:-1]).astype(np.int64)
         lens_present = lens[mask]
@@ -347,8 +348,9 @@ def arrow_to_wire(arr: pa.Array, dt: DataType, nullable: bool, name: str) -> Wir
         if null_count == 0:
             # all-present: the slice's values are the contiguous child range
             # [off[0], off[-1]) — view it and rebase the offsets, no gather
-            base = int(off[0]) if R else 0
-            vals = numeric_cast(child[base:int(off[-1])] if R else child[:0])
+            # (R == 0 included: row_off must still start at 0)
+            base = int(off[0])
+            vals = numeric_cast(child[base:int(off[-1])])
             return WireColumn(kind, False, presence, off - base, vals)
         starts = off[:-1][mask]
         idx = _ragged_gather_idx(starts.astype(np.int64), lens[mask])

@@ -155,3 +155,22 @@ class TestSlicedConversion:
             "s": pa.array(["a", None, "c", "d", None, "f", "g", "h"]),
         })
         self._roundtrip_eq(tab)
+
+    def test_empty_slice_of_offset_column(self):
+        """An empty slice taken at a non-zero offset must still produce
+        wire columns whose offsets start at 0."""
+        from spark_tfrecord_amd.arrow_interop import table_to_batch, schema_from_arrow
+        rng = np.random.default_rng(4)
+        off = np.concatenate([[0], np.cumsum(rng.integers(1, 4, 50))])
+        tab = pa.table({
+            "s": pa.array([f"x{i}" for i in range(50)]),
+            "l": pa.LargeListArray.from_arrays(
+                off.astype(np.int64), rng.integers(0, 9, off[-1]).astype(np.int64)),
+        })
+        empty = tab.slice(30, 0)
+        b = table_to_batch(empty, schema_from_arrow(tab.schema))
+        for c in b.columns:
+            assert c.row_off[0] == 0
+            assert len(c.values) == 0
+            if c.elem_off is not None:
+                assert c.elem_off[0] == 0

@@ -862,28 +862,6 @@ def _device_inflate_group(data: torch.Tensor, gz_items, device) -> bool:
     inflated bytes (a corrupt stream also breaks the frame chain)."""
     import os as _os
 
-    # Hybrid split: the 16-thread host zlib pool is otherwise idle while the
-    # latency-bound inflate kernel runs, so a tail share of FILES (by
-    # uncompressed bytes) inflates on the host concurrently and uploads into
-    # its slices of `data`. Host zlib moves ~1.5 GB/s, the device ~5 GB/s on
-    # entropy-coded data, so ~0.22 of the bytes on the host evens the finish
-    # times. TFREC_GZ_HYBRID=0 disables.
-    host_items = []
-    try:
-        _frac = float(_os.environ.get("TFREC_GZ_HYBRID", "0.22"))
-    except ValueError:
-        _frac = 0.0
-    if _frac > 0 and len(gz_items) >= 4:
-        usizes = [sum(u for _, u in m[1]) for _, m, _ in gz_items]
-        total_u = sum(usizes)
-        acc = 0
-        cut = len(gz_items)
-        while cut > 1 and acc + usizes[cut - 1] <= _frac * total_u:
-            acc += usizes[cut - 1]
-            cut -= 1
-        host_items = gz_items[cut:]
-        gz_items = gz_items[:cut]
-
     comp_sizes = [_os.path.getsize(p) for p, _, _ in gz_items]
     comp_total = sum(comp_sizes)
     comp = torch.empty(max(comp_total, 1), dtype=torch.uint8,
@@ -922,62 +900,10 @@ def _device_inflate_group(data: torch.Tensor, gz_items, device) -> bool:
     meta_np = np.array([in_off, in_len, out_off, out_len], np.int64)
     meta_dev = torch.as_tensor(meta_np).to(device)
     err = torch.full((1,), -1, dtype=torch.int64, device=device)
-    if in_off:
-        _native.gpu_inflate_segments(
-            comp.data_ptr(), meta_dev[0].data_ptr(), meta_dev[1].data_ptr(),
-            meta_dev[2].data_ptr(), meta_dev[3].data_ptr(), len(in_off),
-            data.data_ptr(), err.data_ptr(), _stream())
-    if host_items:
-        # runs WHILE the kernel executes (launch above is async); segments
-        # inflate in parallel on the shared pool into a pinned buffer, then
-        # one async H2D per file lands the bytes in `data`
-        import zlib as _zlib
-        from ..io import paths as P
-
-        pool = P.shared_pool()
-        up_streams = _dma_streams()
-        used_up = []
-        try:
-            for k, (p, meta, out_base) in enumerate(host_items):
-                body_off, segs, _crc, _isize = meta
-                with open(p, "rb") as f:
-                    raw = f.read()
-                n_u = sum(u for _, u in segs)
-                host_buf = torch.empty(max(n_u, 1), dtype=torch.uint8,
-                                       pin_memory=True)[:n_u]
-                view = host_buf.numpy()
-                spans = []
-                so, uo = body_off, 0
-                for c, u in segs:
-                    spans.append((so, c, uo, u))
-                    so += c
-                    uo += u
-
-                def _one(span, raw=raw, view=view):
-                    so, c, uo, u = span
-                    d = _zlib.decompressobj(-15)
-                    out = d.decompress(raw[so:so + c]) + d.flush()
-                    if len(out) != u:
-                        raise ValueError("segment size mismatch")
-                    view[uo:uo + u] = np.frombuffer(out, np.uint8)
-
-                list(pool.map(_one, spans))
-                st = up_streams[k % len(up_streams)]
-                st.wait_stream(main)
-                with torch.cuda.stream(st):
-                    data[out_base:out_base + n_u].copy_(host_buf,
-                                                        non_blocking=True)
-                used_up.append((st, host_buf))
-        except (ValueError, _zlib.error):
-            for st, _ in used_up:
-                st.synchronize()
-            return False
-        # host-sync the uploads: the pinned buffers die here, so the copies
-        # must be complete before their memory recycles
-        for st, _ in used_up:
-            st.synchronize()
-        for st, _ in used_up:
-            main.wait_stream(st)
+    _native.gpu_inflate_segments(
+        comp.data_ptr(), meta_dev[0].data_ptr(), meta_dev[1].data_ptr(),
+        meta_dev[2].data_ptr(), meta_dev[3].data_ptr(), len(in_off),
+        data.data_ptr(), err.data_ptr(), _stream())
     return int(err.item()) == -1
 
 

@@ -830,25 +830,3 @@ class TestPipelinedWriteGPU:
             assert f1.read() == f2.read()
         df = stf.read_tfrecord(a, engine="gpu")
         assert len(df.collect()) == rows
-
-
-class TestHybridInflate:
-    def test_hybrid_split_matches_device_only(self, tmp_sandbox, monkeypatch):
-        """Reads must be identical whether segments inflate on the device
-        only, hybrid host+device, or mostly host."""
-        rng = np.random.default_rng(21)
-        rows = 60_000
-        t = pa.table({"byteArray": pa.array(
-            [rng.bytes(int(rng.integers(50, 400))) for _ in range(rows)],
-            type=pa.large_binary())})
-        out = str(tmp_sandbox / "hyb")
-        stf.write_tfrecord(t, out, record_type="ByteArray", codec="gzip",
-                           num_shards=8, engine="cpu")
-        results = []
-        for frac in ("0", "0.22", "0.8"):
-            monkeypatch.setenv("TFREC_GZ_HYBRID", frac)
-            df = stf.read_tfrecord(out, record_type="ByteArray", engine="gpu")
-            tab = df.to_arrow_table()
-            results.append(sorted(tab.column("byteArray").to_pylist()))
-        assert results[0] == results[1] == results[2]
-        assert len(results[0]) == rows

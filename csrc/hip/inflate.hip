@@ -51,6 +51,9 @@ using tfrec::inflate::inflate_one;
 // of 64 independent streams' branches), and the lanes split the bulk
 // copies. 4 waves per block => 4 KiB LDS, so occupancy is wave-limited,
 // not LDS-limited.
+// NOTE: amdgpu_waves_per_eu(6) was tried (80 VGPRs, 6 waves/SIMD) and
+// measured ~25% SLOWER — the forced VGPR cap introduced vector spills on
+// the serial decode chain, which cost more than the extra wave hid.
 __global__ void __launch_bounds__(256) inflate_segments_kernel(
     const u8* __restrict__ comp, const i64* __restrict__ in_off,
     const i64* __restrict__ in_len, const i64* __restrict__ out_off,

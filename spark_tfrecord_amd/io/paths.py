@@ -100,6 +100,12 @@ def codec_from_path(path: str) -> Optional[str]:
 # ratio cost of the extra dictionary resets is a few % at level 6
 # (TFREC_GZ_SEGMENT overrides for ratio-sensitive datasets).
 _GZ_SEGMENT = int(os.environ.get("TFREC_GZ_SEGMENT", 32 << 10))
+# Segments whose level-6 deflate saves less than this ratio re-emit as
+# STORED blocks (zstd-style incompressibility bailout, still standard
+# gzip): decode streams at copy speed instead of one bit-serial Huffman
+# symbol per byte. <= 0 disables.
+_GZ_STORED_THRESHOLD = float(os.environ.get("TFREC_GZ_STORED_THRESHOLD",
+                                            "0.98"))
 _GZ_MARK = b"\x00\x00\xff\xff"
 _GZ_MAX_SEGS = 8189  # FEXTRA payload cap: 65535 bytes / 8 per segment
 
@@ -126,6 +132,18 @@ def compress_bytes(data: bytes, codec: Optional[str]) -> bytes:
             hi = min(pos + seg, n)
             body = c.compress(data[pos:hi])
             body += (c.flush() if hi == n else c.flush(zlib.Z_FULL_FLUSH))
+            if len(body) >= _GZ_STORED_THRESHOLD * (hi - pos) > 0:
+                # effectively incompressible: entropy-coding buys almost
+                # nothing but costs ~30x on decode (every output byte
+                # becomes a bit-serial Huffman symbol; stored segments
+                # stream at copy speed on host AND device). Re-emit this
+                # segment as stored blocks (level 0 — still standard
+                # deflate; the full flush resets the dictionary, so
+                # segments are independent and the swap is local).
+                c0 = zlib.compressobj(0, zlib.DEFLATED, -15)
+                body = c0.compress(data[pos:hi])
+                body += (c0.flush() if hi == n else c0.flush(zlib.Z_FULL_FLUSH))
+                c = zlib.compressobj(6, zlib.DEFLATED, -15)  # fresh dict
             chunks.append(body)
             seg_lens.append((len(body), hi - pos))
             pos = hi

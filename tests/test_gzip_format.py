@@ -97,3 +97,43 @@ class TestSegmentTableFormat:
         df = stf.read_tfrecord(out, engine="cpu").sort("x")
         rows = df.collect()
         assert len(rows) == 5000 and rows[17]["s"] == "value-17"
+
+
+class TestStoredBailout:
+    """Incompressible segments re-emit as stored blocks (zstd-style bailout,
+    still standard gzip): entropy-coding buys <2% there but costs ~30x on
+    decode. Mixed files keep deflate where it pays, per segment."""
+
+    def test_incompressible_goes_stored(self):
+        rng = np.random.default_rng(3)
+        data = rng.bytes(1 << 20)
+        gz = P.compress_bytes(data, "gzip")
+        assert len(gz) < len(data) * 1.01  # stored overhead only
+        import gzip as _gzip
+        assert _gzip.decompress(gz) == data  # foreign readers unaffected
+
+    def test_mixed_file_per_segment_choice(self):
+        rng = np.random.default_rng(4)
+        data = rng.bytes(200_000) + b"abc" * 100_000 + rng.bytes(200_000)
+        gz = P.compress_bytes(data, "gzip")
+        import gzip as _gzip
+        assert _gzip.decompress(gz) == data
+        # the compressible middle must still shrink the whole file
+        assert len(gz) < len(data) * 0.95
+        # and every segment round-trips through the shared inflate core
+        body_off, segs, _, _ = P.parse_gz_segments(gz)
+        pos, upos = body_off, 0
+        from spark_tfrecord_amd import _native
+        for c, u in segs:
+            assert _native.host_inflate_segment(gz[pos:pos + c], u) == \
+                data[upos:upos + u]
+            pos += c
+            upos += u
+
+    def test_threshold_disable(self, monkeypatch):
+        monkeypatch.setattr(P, "_GZ_STORED_THRESHOLD", 0.0)
+        rng = np.random.default_rng(5)
+        data = rng.bytes(200_000)
+        gz = P.compress_bytes(data, "gzip")
+        import gzip as _gzip
+        assert _gzip.decompress(gz) == data

@@ -139,6 +139,30 @@ def _w_save_mode_ignore(rank, tmp, extra):
     assert sorted(os.listdir(out)) == before
 
 
+def _w_pipelined_write(rank, tmp, extra):
+    """Each rank writes its own part file through the PIPELINED single-shard
+    path (chunk appends overlapped with encode); a rank-sharded read must
+    see every row exactly once."""
+    import spark_tfrecord_amd as stf
+    from spark_tfrecord_amd.io import writer as W
+    from spark_tfrecord_amd.parallel import write_tfrecord_distributed
+
+    W._PIPE_CHUNK_ROWS = 400  # force many chunks per rank
+    out = os.path.join(tmp, "pipew")
+    rows = 3_000
+    data = {"x": np.arange(rank * rows, (rank + 1) * rows, dtype=np.int64)}
+    write_tfrecord_distributed(data, out, mode="overwrite")
+    import torch.distributed as dist
+
+    dist.barrier()
+    if rank == 0:
+        df = stf.read_tfrecord(out, engine="cpu")
+        got = sorted(r["x"] for r in df.collect())
+        import torch.distributed as d2
+        world = d2.get_world_size()
+        assert got == list(range(world * rows))
+
+
 # -- pytest entry points -----------------------------------------------------
 
 @pytest.mark.timeout(240)
@@ -386,3 +410,8 @@ def test_distributed_infer_bytearray(tmp_path):
 @pytest.mark.timeout(420)
 def test_distributed_read_fewer_files_than_ranks(tmp_path):
     _run(_w_read_fewer_files_than_ranks, tmp_path, world=5)
+
+
+@pytest.mark.timeout(240)
+def test_distributed_pipelined_write(tmp_path):
+    _run(_w_pipelined_write, tmp_path)

@@ -25,6 +25,7 @@
 #include <pybind11/pybind11.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 #include "../codec_core.h"
 #include "../inflate_core.h"
@@ -54,15 +55,45 @@ using tfrec::inflate::inflate_one;
 // NOTE: amdgpu_waves_per_eu(6) was tried (80 VGPRs, 6 waves/SIMD) and
 // measured ~25% SLOWER — the forced VGPR cap introduced vector spills on
 // the serial decode chain, which cost more than the extra wave hid.
+//
+// TWO segments per wave (one per 32-lane HALF). PMC showed the "wave-
+// uniform" decode was never actually scalarized: all 64 lanes execute the
+// same vector instructions redundantly on one stream. Giving each half its
+// own stream makes those SAME instructions carry two streams — near-free
+// except (a) EXEC-mask divergence where the halves take different paths
+// (rare on literal-heavy data; amortized on headers), (b) per-half LDS
+// scratch/tables (~28 KiB/block, still under the 5-block register cap),
+// (c) half-width bulk copies for stored blocks/long matches. The launcher
+// pairs adjacent segments, which the callers emit in like-sized order.
 __global__ void __launch_bounds__(256) inflate_segments_kernel(
     const u8* __restrict__ comp, const i64* __restrict__ in_off,
     const i64* __restrict__ in_len, const i64* __restrict__ out_off,
     const i64* __restrict__ out_len, i64 nseg, u8* __restrict__ out,
     unsigned long long* __restrict__ err) {
+  __shared__ LaneScratch S[8];
+  __shared__ uint16_t T[8][tfrec::inflate::kLitTabSize];
+  __shared__ uint16_t D[8][tfrec::inflate::kDistTabSize];
+  const int half = threadIdx.x >> 5;       // 0..7 within the block
+  const int lane = threadIdx.x & 31;
+  i64 stream0 = blockIdx.x * 8 + half;
+  i64 nstreams = (i64)gridDim.x * 8;
+  for (i64 seg = stream0; seg < nseg; seg += nstreams) {
+    int rc = inflate_one<32>(comp + in_off[seg], in_len[seg],
+                             out + out_off[seg], out_len[seg], S[half],
+                             T[half], D[half], lane);
+    if (rc && lane == 0)
+      atomicMin(err, ((unsigned long long)(seg + 1) << 8) | (u32)rc);
+  }
+}
+
+// Single-stream-per-wave variant kept for A/B measurement
+// (TFREC_INFLATE_STREAMS=1).
+__global__ void __launch_bounds__(256) inflate_segments_kernel1(
+    const u8* __restrict__ comp, const i64* __restrict__ in_off,
+    const i64* __restrict__ in_len, const i64* __restrict__ out_off,
+    const i64* __restrict__ out_len, i64 nseg, u8* __restrict__ out,
+    unsigned long long* __restrict__ err) {
   __shared__ LaneScratch S[4];
-  // 10-bit direct lit/len decode tables, one per wave (8 KiB): total LDS
-  // ~12 KiB/block, still far below the 13-block LDS ceiling — occupancy
-  // stays register-limited at 5 waves/SIMD.
   __shared__ uint16_t T[4][tfrec::inflate::kLitTabSize];
   __shared__ uint16_t D[4][tfrec::inflate::kDistTabSize];
   const int wid = threadIdx.x >> 6;
@@ -81,7 +112,19 @@ void gpu_inflate_segments(uintptr_t comp, uintptr_t in_off, uintptr_t in_len,
                           uintptr_t out_off, uintptr_t out_len, i64 nseg,
                           uintptr_t out, uintptr_t err, uintptr_t stream) {
   if (nseg <= 0) return;
-  i64 blocks = (nseg + 3) / 4;  // 4 waves (segments) per block
+  const char* env = getenv("TFREC_INFLATE_STREAMS");
+  if (env && env[0] == '1' && env[1] == '\0') {
+    i64 blocks = (nseg + 3) / 4;  // 4 waves (segments) per block
+    if (blocks > 16384) blocks = 16384;
+    hipLaunchKernelGGL(inflate_segments_kernel1, dim3((uint32_t)blocks),
+                       dim3(256), 0, (hipStream_t)stream, (const u8*)comp,
+                       (const i64*)in_off, (const i64*)in_len,
+                       (const i64*)out_off, (const i64*)out_len, nseg,
+                       (u8*)out, (unsigned long long*)err);
+    HIPI_CHECK(hipGetLastError());
+    return;
+  }
+  i64 blocks = (nseg + 7) / 8;  // 8 half-wave streams per block
   if (blocks > 16384) blocks = 16384;
   hipLaunchKernelGGL(inflate_segments_kernel, dim3((uint32_t)blocks), dim3(256),
                      0, (hipStream_t)stream, (const u8*)comp,

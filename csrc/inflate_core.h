@@ -206,9 +206,10 @@ constexpr int kDistTabSize = 1 << kDistTabBits;
 
 template <int BITS>
 TFR_HOSTDEV inline void fill_huff_table(const u32* bc, const uint16_t* sym,
-                                        uint16_t* tab, int lane) {
+                                        uint16_t* tab, int lane,
+                                        int nlanes = 64) {
   int start = lane < 0 ? 0 : lane;
-  int step = lane < 0 ? 1 : 64;
+  int step = lane < 0 ? 1 : nlanes;
   for (int idx = start; idx < (1 << BITS); idx += step) {
     u32 rev = (u32)idx << (15 - BITS);  // left-justified prefix
     u32 e = 0;
@@ -307,11 +308,12 @@ TFR_HOSTDEV inline int huff_decode(BitRd& br, const u32* bc,
   return sym[(uint16_t)((rev >> (15 - l)) + (u32)(uint16_t)w)];
 }
 
-// Bulk copy: serial on host (lane < 0); lane-strided u64s when a whole
-// wave runs ONE segment uniformly (the GPU decomposition — per-lane
+// Bulk copy: serial on host (lane < 0); lane-strided u64s when `nlanes`
+// lanes run ONE segment in lockstep (the GPU decomposition — per-lane
 // segments paid a ~5x SIMT divergence tax on 64 independent bitstreams).
 // Only used where src/dst cannot overlap within the stride window.
-TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane) {
+TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane,
+                                  int nlanes = 64) {
   if (lane < 0) {
     i64 i = 0;
     for (; i + 8 <= n; i += 8) {
@@ -322,13 +324,13 @@ TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane) {
     for (; i < n; ++i) dst[i] = src[i];
     return;
   }
-  for (i64 i = (i64)lane * 8; i + 8 <= n; i += 64 * 8) {
+  for (i64 i = (i64)lane * 8; i + 8 <= n; i += (i64)nlanes * 8) {
     u64 w;
     __builtin_memcpy(&w, src + i, 8);
     __builtin_memcpy(dst + i, &w, 8);
   }
   i64 tail = n & ~((i64)7);
-  for (i64 b = tail + lane; b < n; b += 64) dst[b] = src[b];
+  for (i64 b = tail + lane; b < n; b += nlanes) dst[b] = src[b];
   TFR_WAVE_FENCE();  // other lanes may read these bytes (later matches)
 }
 
@@ -341,11 +343,13 @@ TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane) {
 // WHOLE WAVE through this function in lockstep on one segment — every
 // value is wave-uniform (the compiler keeps it in scalar registers), and
 // `lane` only splits the big copies.
+template <int NLANES = 64>
 TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
                                   u8* __restrict__ dst, i64 expect,
                                   LaneScratch& L, uint16_t* __restrict__ lit_tab,
                                   uint16_t* __restrict__ dist_tab,
                                   int lane = -1) {
+  constexpr int nlanes = NLANES;
   BitRd br;
   br_init(br, in, ilen);
   i64 opos = 0;
@@ -367,7 +371,7 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
       // rewind both the bit buffer's and the prefetch register's bytes
       const u8* src = br.p - br.pre_n - (br.n >> 3);
       if (src + len > br.end || opos + (i64)len > expect) return 3;
-      bulk_copy(dst + opos, src, (i64)len, lane);
+      bulk_copy(dst + opos, src, (i64)len, lane, nlanes);
       opos += len;
       br.p = src + len;
       br.buf = 0;
@@ -430,8 +434,9 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
     if (!build_huff4(L.lens4, dist_off, hdist, L.bc_dist, L.rank_dist,
                      &L.sym[288]))
       return 10;
-    fill_huff_table<kLitTabBits>(L.bc_lit, L.sym, lit_tab, lane);
-    fill_huff_table<kDistTabBits>(L.bc_dist, &L.sym[288], dist_tab, lane);
+    fill_huff_table<kLitTabBits>(L.bc_lit, L.sym, lit_tab, lane, nlanes);
+    fill_huff_table<kDistTabBits>(L.bc_dist, &L.sym[288], dist_tab, lane,
+                                  nlanes);
     // literal accumulation window: byte-per-literal global stores made the
     // literal-heavy path store-bound; 8 literals flush as one u64 store
     // (flushed before matches, which may read the freshly-written bytes)
@@ -474,7 +479,7 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
         const u8* sp = dst + (opos - dist);
         u8* dp = dst + opos;
         if (lane >= 0 && dist >= mlen) {
-          bulk_copy(dp, sp, mlen, lane);  // non-overlapping: wave-strided
+          bulk_copy(dp, sp, mlen, lane, nlanes);  // non-overlap: strided
         } else if (dist >= 8) {
           i64 i = 0;
           for (; i + 8 <= mlen; i += 8) {

@@ -108,12 +108,16 @@ __global__ void __launch_bounds__(256) inflate_segments_kernel1(
   }
 }
 
+// streams_per_wave: 1 = whole-wave (best for match-heavy segments — wide
+// cooperative copies, zero divergence), 2 = half-wave pairs (best for
+// literal-heavy segments — the same instructions carry two streams). The
+// Python caller routes each segment by its compression ratio.
 void gpu_inflate_segments(uintptr_t comp, uintptr_t in_off, uintptr_t in_len,
                           uintptr_t out_off, uintptr_t out_len, i64 nseg,
-                          uintptr_t out, uintptr_t err, uintptr_t stream) {
+                          uintptr_t out, uintptr_t err, uintptr_t stream,
+                          int streams_per_wave) {
   if (nseg <= 0) return;
-  const char* env = getenv("TFREC_INFLATE_STREAMS");
-  if (env && env[0] == '1' && env[1] == '\0') {
+  if (streams_per_wave <= 1) {
     i64 blocks = (nseg + 3) / 4;  // 4 waves (segments) per block
     if (blocks > 16384) blocks = 16384;
     hipLaunchKernelGGL(inflate_segments_kernel1, dim3((uint32_t)blocks),
@@ -140,7 +144,8 @@ void register_inflate(py::module_& m) {
   m.def("gpu_inflate_segments", &gpu_inflate_segments, py::arg("comp"),
         py::arg("in_off"), py::arg("in_len"), py::arg("out_off"),
         py::arg("out_len"), py::arg("nseg"), py::arg("out"), py::arg("err"),
-        py::arg("stream"),
-        "Inflate full-flush deflate segments, one per lane; err[0] (init "
-        "~0ull) collects ((seg+1)<<8)|cause of the first failure");
+        py::arg("stream"), py::arg("streams_per_wave") = 1,
+        "Inflate full-flush deflate segments (1 = one per wave, 2 = one per "
+        "half-wave); err[0] (init ~0ull) collects ((seg+1)<<8)|cause of the "
+        "first failure");
 }

@@ -897,13 +897,29 @@ def _device_inflate_group(data: torch.Tensor, gz_items, device) -> bool:
         coff += comp_sizes[k]
     for st in set(used):
         main.wait_stream(st)
+    # Route each segment to the kernel shape its data favors (measured,
+    # profiles/RESULTS.md): literal-heavy segments — compression ratio near
+    # 1, every output byte is one Huffman symbol — run ~20% faster TWO per
+    # wave (half-wave streams share the same vector instructions); match-
+    # heavy segments favor the whole-wave kernel (wide cooperative copies,
+    # no divergence between halves). TFREC_INFLATE_STREAMS=1|2 forces one.
     meta_np = np.array([in_off, in_len, out_off, out_len], np.int64)
-    meta_dev = torch.as_tensor(meta_np).to(device)
+    force = _os.environ.get("TFREC_INFLATE_STREAMS", "")
+    if force in ("1", "2"):
+        lit = np.full(meta_np.shape[1], force == "2")
+    else:
+        lit = meta_np[1] >= (0.85 * np.maximum(meta_np[3], 1))
     err = torch.full((1,), -1, dtype=torch.int64, device=device)
-    _native.gpu_inflate_segments(
-        comp.data_ptr(), meta_dev[0].data_ptr(), meta_dev[1].data_ptr(),
-        meta_dev[2].data_ptr(), meta_dev[3].data_ptr(), len(in_off),
-        data.data_ptr(), err.data_ptr(), _stream())
+    for mask, spw in ((lit, 2), (~lit, 1)):
+        k = int(mask.sum())
+        if k == 0:
+            continue
+        sub = torch.as_tensor(
+            np.ascontiguousarray(meta_np[:, mask])).to(device)
+        _native.gpu_inflate_segments(
+            comp.data_ptr(), sub[0].data_ptr(), sub[1].data_ptr(),
+            sub[2].data_ptr(), sub[3].data_ptr(), k,
+            data.data_ptr(), err.data_ptr(), _stream(), spw)
     return int(err.item()) == -1
 
 

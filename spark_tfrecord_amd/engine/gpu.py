@@ -914,8 +914,13 @@ def _device_inflate_group(data: torch.Tensor, gz_items, device) -> bool:
         k = int(mask.sum())
         if k == 0:
             continue
+        idx = np.nonzero(mask)[0]
+        if spw == 2 and k > 2:
+            # pair like-sized segments in one wave: halves reconverge at
+            # segment end, so a short partner idles under a long one
+            idx = idx[np.argsort(-meta_np[3, idx], kind="stable")]
         sub = torch.as_tensor(
-            np.ascontiguousarray(meta_np[:, mask])).to(device)
+            np.ascontiguousarray(meta_np[:, idx])).to(device)
         _native.gpu_inflate_segments(
             comp.data_ptr(), sub[0].data_ptr(), sub[1].data_ptr(),
             sub[2].data_ptr(), sub[3].data_ptr(), k,

@@ -830,3 +830,22 @@ class TestPipelinedWriteGPU:
             assert f1.read() == f2.read()
         df = stf.read_tfrecord(a, engine="gpu")
         assert len(df.collect()) == rows
+
+
+class TestStreamedShardDeviceRead:
+    def test_streamed_gzip_shard_inflates_on_device(self, tmp_sandbox):
+        """A ShardWriter gzip shard carries the FEXTRA segment table, so the
+        GPU read path must take the device-inflate branch (gz_device_meta
+        not None) and decode identically."""
+        from spark_tfrecord_amd.engine import gpu as gpu_engine
+        d = str(tmp_sandbox / "swgz")
+        os.makedirs(d, exist_ok=True)
+        p = os.path.join(d, "part-00000.tfrecord.gz")
+        with stf.ShardWriter(p, record_type="Example", codec="gzip",
+                             engine="cpu") as w:
+            for k in range(4):
+                w.write({"x": np.arange(k * 30_000, (k + 1) * 30_000,
+                                        dtype=np.int64)})
+        assert gpu_engine.gz_device_meta(p) is not None
+        df = stf.read_tfrecord(d, engine="gpu")
+        assert sorted(r["x"] for r in df.collect()) == list(range(120_000))

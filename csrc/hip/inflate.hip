@@ -86,6 +86,32 @@ __global__ void __launch_bounds__(256) inflate_segments_kernel(
   }
 }
 
+// FOUR streams per wave (16 lanes each) with narrower tables (9-bit lit,
+// 7-bit dist) so LDS still fits 4 blocks/CU at occupancy 4 — twice the
+// streams in flight of the half-wave kernel for the latency-bound
+// literal-heavy decode. Copies are quarter-width and 4 streams diverge
+// more often, so this is routed only at very high compression ratios.
+__global__ void __launch_bounds__(256) inflate_segments_kernel4(
+    const u8* __restrict__ comp, const i64* __restrict__ in_off,
+    const i64* __restrict__ in_len, const i64* __restrict__ out_off,
+    const i64* __restrict__ out_len, i64 nseg, u8* __restrict__ out,
+    unsigned long long* __restrict__ err) {
+  __shared__ LaneScratch S[16];
+  __shared__ uint16_t T[16][1 << 9];
+  __shared__ uint16_t D[16][1 << 7];
+  const int q = threadIdx.x >> 4;          // 0..15 within the block
+  const int lane = threadIdx.x & 15;
+  i64 stream0 = blockIdx.x * 16 + q;
+  i64 nstreams = (i64)gridDim.x * 16;
+  for (i64 seg = stream0; seg < nseg; seg += nstreams) {
+    int rc = inflate_one<16, 9, 7>(comp + in_off[seg], in_len[seg],
+                                   out + out_off[seg], out_len[seg], S[q],
+                                   T[q], D[q], lane);
+    if (rc && lane == 0)
+      atomicMin(err, ((unsigned long long)(seg + 1) << 8) | (u32)rc);
+  }
+}
+
 // Single-stream-per-wave variant kept for A/B measurement
 // (TFREC_INFLATE_STREAMS=1).
 __global__ void __launch_bounds__(256) inflate_segments_kernel1(
@@ -117,6 +143,17 @@ void gpu_inflate_segments(uintptr_t comp, uintptr_t in_off, uintptr_t in_len,
                           uintptr_t out, uintptr_t err, uintptr_t stream,
                           int streams_per_wave) {
   if (nseg <= 0) return;
+  if (streams_per_wave >= 4) {
+    i64 blocks = (nseg + 15) / 16;  // 16 quarter-wave streams per block
+    if (blocks > 16384) blocks = 16384;
+    hipLaunchKernelGGL(inflate_segments_kernel4, dim3((uint32_t)blocks),
+                       dim3(256), 0, (hipStream_t)stream, (const u8*)comp,
+                       (const i64*)in_off, (const i64*)in_len,
+                       (const i64*)out_off, (const i64*)out_len, nseg,
+                       (u8*)out, (unsigned long long*)err);
+    HIPI_CHECK(hipGetLastError());
+    return;
+  }
   if (streams_per_wave <= 1) {
     i64 blocks = (nseg + 3) / 4;  // 4 waves (segments) per block
     if (blocks > 16384) blocks = 16384;

@@ -343,7 +343,8 @@ TFR_HOSTDEV inline void bulk_copy(u8* dst, const u8* src, i64 n, int lane,
 // WHOLE WAVE through this function in lockstep on one segment — every
 // value is wave-uniform (the compiler keeps it in scalar registers), and
 // `lane` only splits the big copies.
-template <int NLANES = 64>
+template <int NLANES = 64, int LIT_BITS = kLitTabBits,
+          int DIST_BITS = kDistTabBits>
 TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
                                   u8* __restrict__ dst, i64 expect,
                                   LaneScratch& L, uint16_t* __restrict__ lit_tab,
@@ -434,9 +435,9 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
     if (!build_huff4(L.lens4, dist_off, hdist, L.bc_dist, L.rank_dist,
                      &L.sym[288]))
       return 10;
-    fill_huff_table<kLitTabBits>(L.bc_lit, L.sym, lit_tab, lane, nlanes);
-    fill_huff_table<kDistTabBits>(L.bc_dist, &L.sym[288], dist_tab, lane,
-                                  nlanes);
+    fill_huff_table<LIT_BITS>(L.bc_lit, L.sym, lit_tab, lane, nlanes);
+    fill_huff_table<DIST_BITS>(L.bc_dist, &L.sym[288], dist_tab, lane,
+                               nlanes);
     // literal accumulation window: byte-per-literal global stores made the
     // literal-heavy path store-bound; 8 literals flush as one u64 store
     // (flushed before matches, which may read the freshly-written bytes)
@@ -456,7 +457,7 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
       ln = 0;
     };
     for (;;) {
-      int s = huff_decode_tab<kLitTabBits>(br, L.bc_lit, L.sym, lit_tab);
+      int s = huff_decode_tab<LIT_BITS>(br, L.bc_lit, L.sym, lit_tab);
       if (s < 0) return 11;
       if (s < 256) {
         if (opos + ln >= expect) return 12;
@@ -470,8 +471,8 @@ TFR_HOSTDEV inline int inflate_one(const u8* __restrict__ in, i64 ilen,
         s -= 257;
         if (s >= 29) return 13;
         i64 mlen = kLenBase[s] + (i64)br_bits(br, kLenExtra[s]);
-        int d = huff_decode_tab<kDistTabBits>(br, L.bc_dist, &L.sym[288],
-                                              dist_tab);
+        int d = huff_decode_tab<DIST_BITS>(br, L.bc_dist, &L.sym[288],
+                                           dist_tab);
         if (d < 0 || d >= 30) return 14;
         i64 dist = kDistBase[d] + (i64)br_bits(br, kDistExtra[d]);
         if (br.n < 0) return 15;

@@ -905,12 +905,14 @@ def _device_inflate_group(data: torch.Tensor, gz_items, device) -> bool:
     # no divergence between halves). TFREC_INFLATE_STREAMS=1|2 forces one.
     meta_np = np.array([in_off, in_len, out_off, out_len], np.int64)
     force = _os.environ.get("TFREC_INFLATE_STREAMS", "")
-    if force in ("1", "2"):
-        lit = np.full(meta_np.shape[1], force == "2")
+    spw_lit = 2
+    if force in ("1", "2", "4"):
+        spw_lit = int(force)
+        lit = np.full(meta_np.shape[1], spw_lit > 1)
     else:
         lit = meta_np[1] >= (0.85 * np.maximum(meta_np[3], 1))
     err = torch.full((1,), -1, dtype=torch.int64, device=device)
-    for mask, spw in ((lit, 2), (~lit, 1)):
+    for mask, spw in ((lit, spw_lit), (~lit, 1)):
         k = int(mask.sum())
         if k == 0:
             continue

@@ -88,9 +88,11 @@ __global__ void __launch_bounds__(256) inflate_segments_kernel(
 
 // FOUR streams per wave (16 lanes each) with narrower tables (9-bit lit,
 // 7-bit dist) so LDS still fits 4 blocks/CU at occupancy 4 — twice the
-// streams in flight of the half-wave kernel for the latency-bound
-// literal-heavy decode. Copies are quarter-width and 4 streams diverge
-// more often, so this is routed only at very high compression ratios.
+// streams in flight of the half-wave kernel. MEASURED: no gain on
+// literal-heavy data (21.0 vs 21.7M rows/s on config 5) and 33% slower on
+// match-heavy text — 4-way refill/flush divergence eats the extra latency
+// hiding. Kept behind TFREC_INFLATE_STREAMS=4 for measurement; the auto
+// router never picks it.
 __global__ void __launch_bounds__(256) inflate_segments_kernel4(
     const u8* __restrict__ comp, const i64* __restrict__ in_off,
     const i64* __restrict__ in_len, const i64* __restrict__ out_off,

@@ -33,3 +33,8 @@ for f in $(find gpurun_out/w_prof_inf gpurun_out/w_prof_bench -name "*kernel_sta
   echo "== $f"; head -18 "$f"
 done
 echo DONE
+# PMC counters for the routed inflate kernels (separate pass, counters only)
+cd /tmp
+timeout 200 rocprofv3 --pmc SQ_WAVES SQ_INSTS_VALU SQ_INSTS_SALU SQ_INSTS_LDS SQ_BUSY_CYCLES -d "$GRAFT_REPO_ROOT/gpurun_out/w_pmc_inf" -- python "$GRAFT_REPO_ROOT/exp/exp_inflate_only.py" > "$GRAFT_REPO_ROOT/gpurun_out/w_pmc_inf.log" 2>&1
+cd "$GRAFT_REPO_ROOT"
+echo PMC_DONE

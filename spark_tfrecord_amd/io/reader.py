@@ -236,10 +236,18 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
         return cpu_engine.decode_buffer(data, data_schema, record_type,
                                         verify_crc=verify_crc)
 
+    # tiny jobs (<= 2 small files) skip the pool: submit/wait latency
+    # dominates sub-ms decodes (the 1k-row plumbing config round trip)
+    _inline_small = (len(files) <= 2 and
+                     all(os.path.getsize(f) < (256 << 10) for f in files))
+
     def _blob(i: int):
-        if i not in futures:
-            futures[i] = _pool().submit(_host_task, files[i])
-        return futures.pop(i).result()
+        f = futures.pop(i, None)
+        if f is not None:
+            return f.result()
+        if _inline_small:
+            return _host_task(files[i])
+        return _pool().submit(_host_task, files[i]).result()
 
     # GPU path: consecutive uncompressed files are decoded as ONE pipeline
     # (images concatenated in HBM — frames are concatenable — scanned and
@@ -298,7 +306,8 @@ def read_tfrecord(path: str, schema: Optional[StructType] = None,
                 _append_with_parts(t, group, row_counts)
                 continue
             for j in range(i, min(i + window, len(files))):
-                if _needs_host_bytes(files[j]) and j not in futures:
+                if (not _inline_small and _needs_host_bytes(files[j])
+                        and j not in futures):
                     futures[j] = _pool().submit(_host_task, files[j])
             metrics.add(files=1, nbytes=os.path.getsize(fpath))
             got = _blob(i)

@@ -344,3 +344,61 @@ class TestFullTypeMatrix:
             out = cpu_engine.decode_buffer(np.frombuffer(img, np.uint8), schema, rt)
             np.testing.assert_array_equal(np.asarray(col.values),
                                           np.asarray(out.columns[0].values))
+
+
+class TestEdgeSemanticsParity:
+    """Pin edge-case semantics to the reference's behavior."""
+
+    def test_int64_to_integer_silent_truncation(self):
+        """IntegerType reads of out-of-range Int64List values truncate like
+        the reference's Scala .toInt (TFRecordDeserializer.scala:84-86) —
+        numpy astype(int32) has identical wraparound semantics."""
+        from spark_tfrecord_amd.columnar import column_from_values, column_to_pylist
+        from spark_tfrecord_amd.engine import cpu as cpu_engine
+        big = 2**31 + 5  # wraps to -2**31 + 5 in int32
+        schema = stf.StructType([stf.StructField("f", stf.LongType(), True)])
+        b = RecordBatch(schema, [column_from_values(
+            [big], stf.LongType(), True, "f")], 1)
+        raw = cpu_engine.encode_batch(b, "Example")
+        int_schema = stf.StructType([stf.StructField("f", stf.IntegerType(), True)])
+        out = cpu_engine.decode_buffer(
+            np.frombuffer(raw, np.uint8), int_schema, "Example")
+        assert column_to_pylist(out.columns[0], stf.IntegerType(), True, "f") \
+            == [np.int64(big).astype(np.int32).item()]
+
+    def test_megabyte_single_string(self, tmp_path):
+        s = "x" * (1 << 20) + "end"
+        out = str(tmp_path / "big")
+        stf.write_tfrecord({"s": [s, "tiny"]}, out, engine="cpu")
+        got = stf.read_tfrecord(out, engine="cpu").collect()
+        assert sorted(r["s"] for r in got) == sorted([s, "tiny"])
+
+    def test_unicode_feature_names(self, tmp_path):
+        out = str(tmp_path / "uni")
+        stf.write_tfrecord({"héllo_名前": np.arange(10, dtype=np.int64)},
+                           out, engine="cpu")
+        df = stf.read_tfrecord(out, engine="cpu")
+        assert df.columns == ["héllo_名前"]
+        assert df.count() == 10
+
+    def test_empty_list_vs_missing_distinction(self, tmp_path):
+        """An EMPTY Int64List feature reads as an empty array; a MISSING
+        feature reads as null (TFRecordDeserializer.scala nullable rules +
+        parseInt64List length-0 semantics)."""
+        import pyarrow as pa
+        out = str(tmp_path / "el")
+        t = pa.table({"a": pa.array([[1, 2], [], None],
+                                    type=pa.large_list(pa.int64()))})
+        schema = stf.StructType([
+            stf.StructField("a", stf.ArrayType(stf.LongType()), True)])
+        stf.write_tfrecord(t, out, schema=schema, engine="cpu")
+        got = stf.read_tfrecord(out, schema=schema, engine="cpu") \
+            .to_arrow_table().column("a").to_pylist()
+        # written [] keeps its presence as an empty Int64List feature; a
+        # written None is OMITTED from the record (reference serializer
+        # null rule) and reads back as null
+        assert got[0] == [1, 2]
+        # [] was written as an empty feature -> empty array on read
+        assert got[1] == []
+        # None was omitted -> null on read
+        assert got[2] is None

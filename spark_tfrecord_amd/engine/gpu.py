@@ -57,15 +57,27 @@ def check_native():
 # One reusable pinned buffer per purpose, grown geometrically.
 # ---------------------------------------------------------------------------
 
-_pinned_pool = {}
+import threading as _threading
+
+_pinned_tls = _threading.local()
 
 
 def pinned_buffer(tag: str, nbytes: int) -> torch.Tensor:
-    buf = _pinned_pool.get(tag)
+    """Reusable pinned staging buffer for `tag`, grown geometrically.
+
+    Pools are THREAD-LOCAL: a tag names a pipeline slot (e.g. the writer's
+    rotating D2H buffers), and two user threads writing concurrently must
+    not stage through the same memory — one thread's host-side read of the
+    view races the other's next async copy into it. Thread-local storage
+    also frees a thread's buffers when it exits."""
+    pool = getattr(_pinned_tls, "pool", None)
+    if pool is None:
+        pool = _pinned_tls.pool = {}
+    buf = pool.get(tag)
     if buf is None or buf.numel() < nbytes:
         cap = max(nbytes, int((buf.numel() if buf is not None else 1 << 20) * 1.5))
         buf = torch.empty(cap, dtype=torch.uint8, pin_memory=True)
-        _pinned_pool[tag] = buf
+        pool[tag] = buf
     return buf
 
 

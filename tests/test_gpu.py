@@ -908,3 +908,35 @@ class TestGpuSaveModesAndEscaping:
         df = stf.read_tfrecord(out, record_type="ByteArray", engine="gpu")
         got = sorted(df.to_arrow_table().column("byteArray").to_pylist())
         assert got == sorted(payloads)
+
+
+class TestThreadConcurrency:
+    def test_concurrent_writers_and_readers(self, tmp_sandbox):
+        """Two user threads driving full API write+read round-trips
+        concurrently must not corrupt each other (pinned staging pools are
+        thread-local; kernel launches serialize on the shared stream)."""
+        import threading
+        errs = []
+
+        def work(tid):
+            try:
+                rng = np.random.default_rng(tid)
+                for it in range(6):
+                    rows = 120_000 + 1000 * tid
+                    d = str(tmp_sandbox / f"thr{tid}_{it % 2}")
+                    vals = rng.integers(0, 2**60, rows)
+                    stf.write_tfrecord({"x": vals}, d, engine="gpu",
+                                       mode="overwrite")
+                    got = stf.read_tfrecord(d, engine="gpu") \
+                        .to_arrow_table().column("x").to_numpy()
+                    assert np.array_equal(np.sort(got), np.sort(vals)), \
+                        f"thread {tid} iter {it} corrupted"
+            except BaseException as e:  # noqa: BLE001
+                errs.append(e)
+
+        ts = [threading.Thread(target=work, args=(t,)) for t in (1, 2)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(120)
+        assert not errs, errs

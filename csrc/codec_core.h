@@ -394,6 +394,15 @@ TFR_HOSTDEV inline int32_t scan_features_body(const u8* p, const u8* end, i64 ba
     int f = schema_find(schema, key, static_cast<int>(key_len), want_seq);
     if (f < 0) continue;  // unknown feature: ignored, like the reference
     FieldStat* st = &stats[f];
+    // duplicate map keys are valid protobuf with LAST-entry-wins semantics
+    // (what protobuf-java gives the reference); reset the accumulators so
+    // only the final body's counts survive — otherwise the prefix sums
+    // reserve slots the extract pass never fills (garbage in the column)
+    st->nvals = 0;
+    st->nbytes = 0;
+    st->nlists = 0;
+    st->kind_found = 0;
+    st->err = 0;
     st->pos = base + (val ? (val - body_start) : 0);
     st->len = static_cast<i64>(val_len);
     if (!want_seq) {
@@ -762,6 +771,13 @@ TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_
           continue;
         }
         FieldStat* st = &stats[f];
+        // duplicate map keys: last-entry-wins (see scan_features_body) —
+        // reset the accumulators so only this body's counts survive
+        st->nvals = 0;
+        st->nbytes = 0;
+        st->nlists = 0;
+        st->kind_found = 0;
+        st->err = 0;
         st->pos = data_rel + c.pos;
         st->len = (i64)vlen;
         if (!want_seq) {
@@ -814,10 +830,16 @@ TFR_HOSTDEV inline int32_t cur_scan_features_body(ScanCur& c, i64 end, i64 data_
     if (f >= 0 && !val_seen) {
       // key with no value field: protobuf map semantics give the default
       // (empty) Feature — the feature IS present (two-pass form behavior:
-      // scan_features_body sets pos = features-body start, len 0)
+      // scan_features_body sets pos = features-body start, len 0). Also a
+      // last-wins reset: an earlier duplicate's counts must not survive.
       FieldStat* st = &stats[f];
       st->pos = data_rel + body_start;
       st->len = 0;
+      st->nvals = 0;
+      st->nbytes = 0;
+      st->nlists = 0;
+      st->kind_found = 0;
+      st->err = 0;
     }
   }
   return ERR_OK;

@@ -940,3 +940,19 @@ class TestThreadConcurrency:
         for t in ts:
             t.join(120)
         assert not errs, errs
+
+
+class TestDuplicateMapKeysGpu:
+    def test_last_entry_wins_fused_scan(self, tmp_sandbox):
+        """The fused-CRC scan path must apply the same last-entry-wins reset
+        as the two-pass form for duplicate Features-map keys."""
+        from tests.test_serde import _raw_example_with_dup_keys
+        d = str(tmp_sandbox / "dup")
+        os.makedirs(d, exist_ok=True)
+        with open(os.path.join(d, "part-00000.tfrecord"), "wb") as f:
+            f.write(_raw_example_with_dup_keys())
+        schema = stf.StructType([
+            stf.StructField("x", stf.ArrayType(stf.LongType()), True)])
+        got = stf.read_tfrecord(d, schema=schema, engine="gpu") \
+            .to_arrow_table().column("x").to_pylist()
+        assert got == [[7]]

@@ -402,3 +402,53 @@ class TestEdgeSemanticsParity:
         assert got[1] == []
         # None was omitted -> null on read
         assert got[2] is None
+
+
+def _raw_example_with_dup_keys():
+    """Hand-built Example whose Features map repeats the key 'x' — valid
+    protobuf; map semantics are LAST-entry-wins (what protobuf-java gives
+    the reference). Returns a framed single-record file image."""
+    import struct
+
+    def varint(v):
+        out = b""
+        while v >= 0x80:
+            out += bytes([v & 0x7F | 0x80])
+            v >>= 7
+        return out + bytes([v])
+
+    def feature_int64(vals):
+        packed = b"".join(varint(x) for x in vals)
+        body = bytes([0x0a]) + varint(len(packed)) + packed
+        return bytes([0x1a]) + varint(len(body)) + body
+
+    def map_entry(key, feat):
+        e = (bytes([0x0a]) + varint(len(key)) + key.encode() +
+             bytes([0x12]) + varint(len(feat)) + feat)
+        return bytes([0x0a]) + varint(len(e)) + e
+
+    entries = (map_entry("x", feature_int64([1, 2])) +
+               map_entry("x", feature_int64([7])))
+    example = bytes([0x0a]) + varint(len(entries)) + entries
+
+    def mask(c):
+        return ((c >> 15) | (c << 17)) + 0xa282ead8 & 0xFFFFFFFF
+
+    ln = struct.pack("<Q", len(example))
+    return (ln + struct.pack("<I", mask(_native.crc32c(ln))) + example +
+            struct.pack("<I", mask(_native.crc32c(example))))
+
+
+class TestDuplicateMapKeys:
+    def test_last_entry_wins(self):
+        """Duplicate Features-map keys must decode as the LAST entry only —
+        the scan previously accumulated counts across duplicates while the
+        extract read only the final body, leaving garbage slots."""
+        img = _raw_example_with_dup_keys()
+        schema = stf.StructType([
+            stf.StructField("x", stf.ArrayType(stf.LongType()), True)])
+        out = cpu_engine.decode_buffer(np.frombuffer(bytearray(img), np.uint8),
+                                       schema, "Example")
+        c = out.columns[0]
+        assert list(np.asarray(c.row_off)) == [0, 1]
+        assert list(np.asarray(c.values)) == [7]

@@ -174,3 +174,30 @@ class TestSlicedConversion:
             assert len(c.values) == 0
             if c.elem_off is not None:
                 assert c.elem_off[0] == 0
+
+
+class TestMultiChunkInput:
+    def test_concat_table_multi_chunk_write(self, tmp_path):
+        """pa.concat_tables yields multi-chunk columns; the conversion must
+        concatenate (combine_chunks fallback) and round-trip exactly."""
+        rng = np.random.default_rng(9)
+        parts = []
+        for k in range(4):
+            parts.append(pa.table({
+                "id": np.arange(k * 1000, (k + 1) * 1000, dtype=np.int64),
+                "l": pa.LargeListArray.from_arrays(
+                    np.arange(0, 2002, 2, dtype=np.int64),
+                    rng.integers(0, 99, 2000).astype(np.int64)),
+                "s": pa.array([f"c{k}_{i}" for i in range(1000)]),
+            }))
+        tab = pa.concat_tables(parts)
+        assert tab.column("id").num_chunks == 4
+        out = str(tmp_path / "mc")
+        stf.write_tfrecord(tab, out, engine="cpu")
+        got = stf.read_tfrecord(out, engine="cpu").to_arrow_table()
+        assert sorted(got.column("id").to_pylist()) == list(range(4000))
+        by_id = {r["id"]: r for r in got.to_pylist()}
+        want = tab.to_pylist()
+        for w in want[:50] + want[-50:]:
+            assert by_id[w["id"]]["l"] == w["l"]
+            assert by_id[w["id"]]["s"] == w["s"]
